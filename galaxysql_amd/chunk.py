@@ -1,0 +1,148 @@
+"""Host-side mirror of the reference columnar batch format.
+
+Chunk = list of Blocks (reference: polardbx-executor/.../chunk/Chunk.java:41-66);
+Block physical layouts mirror LongBlock (long[] values + boolean[] valueIsNull,
+chunk/LongBlock.java:41-55), IntegerBlock (chunk/IntegerBlock.java:38-73),
+DoubleBlock, and SliceBlock (int[] end-offsets + byte data,
+chunk/SliceBlock.java:40-57), re-expressed as numpy arrays so they map 1:1
+onto the C-ABI `gx_block` (include/gxop.h).
+
+CHUNK_SIZE default 1000 rows mirrors ConnectionParams.java:1088.
+"""
+from __future__ import annotations
+
+import numpy as np
+
+CHUNK_SIZE = 1000
+
+I64, I32, F64, SLICE = 0, 1, 2, 3
+
+_NP_DTYPES = {I64: np.int64, I32: np.int32, F64: np.float64}
+
+
+class Block:
+    """One column of a batch.
+
+    values: numpy array (i64/i32/f64); None for SLICE.
+    nulls:  uint8 array (1 = NULL) or None.
+    offsets/data: SLICE only — int32 END offsets per value + byte buffer.
+    """
+
+    __slots__ = ("type", "values", "nulls", "offsets", "data")
+
+    def __init__(self, btype, values=None, nulls=None, offsets=None, data=None):
+        self.type = btype
+        self.values = values
+        self.nulls = nulls
+        self.offsets = offsets
+        self.data = data
+
+    @property
+    def n_rows(self):
+        if self.type == SLICE:
+            return len(self.offsets)
+        return len(self.values)
+
+    @staticmethod
+    def of(btype, pyvals):
+        """Build from a python list with None for NULL (mirrors the reference
+        test fixture helpers IntegerBlock.of / LongBlock.of /
+        StringBlock.of)."""
+        n = len(pyvals)
+        nulls = np.array([1 if v is None else 0 for v in pyvals], dtype=np.uint8)
+        has_null = bool(nulls.any())
+        if btype == SLICE:
+            data = bytearray()
+            offsets = np.zeros(n, dtype=np.int32)
+            for i, v in enumerate(pyvals):
+                if v is not None:
+                    data.extend(v.encode() if isinstance(v, str) else bytes(v))
+                offsets[i] = len(data)
+            return Block(SLICE, nulls=nulls if has_null else None,
+                         offsets=offsets, data=np.frombuffer(bytes(data), dtype=np.uint8))
+        vals = np.array([0 if v is None else v for v in pyvals], dtype=_NP_DTYPES[btype])
+        return Block(btype, values=vals, nulls=nulls if has_null else None)
+
+    def get(self, i):
+        if self.nulls is not None and self.nulls[i]:
+            return None
+        if self.type == SLICE:
+            start = int(self.offsets[i - 1]) if i > 0 else 0
+            return bytes(self.data[start:int(self.offsets[i])])
+        v = self.values[i]
+        if self.type == F64:
+            return float(v)
+        return int(v)
+
+
+class Chunk:
+    __slots__ = ("blocks", "n_rows")
+
+    def __init__(self, blocks):
+        self.blocks = list(blocks)
+        self.n_rows = self.blocks[0].n_rows if self.blocks else 0
+        for b in self.blocks:
+            assert b.n_rows == self.n_rows, "ragged chunk"
+
+    @property
+    def types(self):
+        return [b.type for b in self.blocks]
+
+    def rows(self):
+        """Materialize python row tuples (test/parity use only)."""
+        return [tuple(b.get(i) for b in self.blocks) for i in range(self.n_rows)]
+
+
+def rows_of(chunks):
+    out = []
+    for c in chunks:
+        out.extend(c.rows())
+    return out
+
+
+def chunks_from_columns(btypes, columns, chunk_size=CHUNK_SIZE):
+    """Split full-length numpy columns into CHUNK_SIZE-row Chunks.
+
+    columns: list of (values, nulls_or_None) per column, or Block instances
+    covering all rows.
+    """
+    blocks = []
+    for t, col in zip(btypes, columns):
+        if isinstance(col, Block):
+            blocks.append(col)
+        else:
+            vals, nulls = col
+            blocks.append(Block(t, values=np.asarray(vals, dtype=_NP_DTYPES[t]),
+                                nulls=None if nulls is None else np.asarray(nulls, dtype=np.uint8)))
+    n = blocks[0].n_rows
+    out = []
+    for start in range(0, n, chunk_size):
+        end = min(start + chunk_size, n)
+        sub = []
+        for b in blocks:
+            if b.type == SLICE:
+                base = int(b.offsets[start - 1]) if start > 0 else 0
+                off = (b.offsets[start:end] - base).astype(np.int32)
+                dend = int(b.offsets[end - 1]) if end > 0 else 0
+                sub.append(Block(SLICE,
+                                 nulls=None if b.nulls is None else b.nulls[start:end],
+                                 offsets=off, data=b.data[base:dend]))
+            else:
+                sub.append(Block(b.type, values=b.values[start:end],
+                                 nulls=None if b.nulls is None else b.nulls[start:end]))
+        out.append(Chunk(sub))
+    return out
+
+
+def multiset(rows, f64_round=None):
+    """Canonical multiset of rows for order-insensitive parity compares
+    (mirrors BaseExecTest.assertExecResultByRow, BaseExecTest.java:78-103).
+    f64_round: decimal places to round floats (tolerance compare)."""
+    from collections import Counter
+
+    def canon(v):
+        if isinstance(v, float) and f64_round is not None:
+            return round(v, f64_round)
+        return v
+
+    return Counter(tuple(canon(v) for v in r) for r in rows)

@@ -1,0 +1,225 @@
+"""Host-side operators mirroring the reference's Executor / ConsumerExecutor
+lifecycle (operator/Executor.java, operator/ConsumerExecutor.java), driving
+the gxop C-ABI.
+
+Lifecycle exercised by Driver.processInternal (mpp/operator/Driver.java:
+449-525): openConsume -> consumeChunk* -> buildConsume -> (operator flips to
+producer) nextChunk* -> close. Names and argument meaning follow the
+reference factories (mpp/operator/factory/ParallelHashJoinExecutorFactory
+.java:77-117, HashAggExecutorFactory.java:71-104).
+"""
+from __future__ import annotations
+
+import ctypes as C
+
+from . import abi
+from .abi import (GxJoinCfg, GxEquiKey, GxAggCfg, GxAggSpec, GxPartCfg, GxResult)
+from .chunk import Chunk, CHUNK_SIZE
+
+
+class EquiJoinKey:
+    """Mirror of EquiJoinKey (polardbx-optimizer/.../core/join/EquiJoinKey
+    .java:25-43). Key columns arrive unified to unified_type."""
+
+    def __init__(self, outer_index, inner_index, unified_type):
+        self.outer_index = outer_index
+        self.inner_index = inner_index
+        self.unified_type = unified_type
+
+
+class ParallelHashJoinExec:
+    """One operator instance (the C side holds the shared build state).
+
+    join_type: abi.INNER/LEFT/RIGHT/SEMI/ANTI; mirrors
+    ParallelHashJoinExec.java:49 construction.
+    """
+
+    def __init__(self, lib, join_type, join_keys, outer_types, inner_types,
+                 max_one_row=False, build_outer=False, anti_null_col=-1,
+                 device=-1, stream=0, expected_build_rows=0):
+        self._lib = lib
+        self._keep = []
+        keys = (GxEquiKey * len(join_keys))()
+        for i, k in enumerate(join_keys):
+            keys[i] = GxEquiKey(k.outer_index, k.inner_index, k.unified_type, 0)
+        ot = (C.c_int32 * len(outer_types))(*outer_types)
+        it = (C.c_int32 * len(inner_types))(*inner_types)
+        cfg = GxJoinCfg(
+            join_type=join_type, single_join=int(max_one_row),
+            build_outer=int(build_outer), n_keys=len(join_keys), keys=keys,
+            n_outer_cols=len(outer_types), outer_types=ot,
+            n_inner_cols=len(inner_types), inner_types=it,
+            anti_null_col=anti_null_col, device=device, stream=stream,
+            expected_build_rows=expected_build_rows)
+        self._keep += [keys, ot, it, cfg]
+        self._op = lib.lib.gxop_join_create(C.byref(cfg))
+        if not self._op:
+            raise RuntimeError(f"gxop_join_create: {lib.error()}")
+
+    def consume_chunk(self, chunk: Chunk):
+        ka = []
+        gc = self._lib.to_gx_chunk(chunk, ka)
+        self._lib.check(self._lib.lib.gxop_join_consume(self._op, C.byref(gc)),
+                        "join_consume")
+
+    def build_consume(self):
+        self._lib.check(self._lib.lib.gxop_join_build(self._op), "join_build")
+
+    def probe_chunk(self, chunk: Chunk) -> Chunk:
+        ka = []
+        gc = self._lib.to_gx_chunk(chunk, ka)
+        out = C.POINTER(GxResult)()
+        self._lib.check(self._lib.lib.gxop_join_probe(self._op, C.byref(gc),
+                                                      C.byref(out)), "join_probe")
+        if not out:
+            return None
+        return self._lib.result_to_chunk(out)
+
+    def tail_chunks(self):
+        chunks = []
+        while True:
+            out = C.POINTER(GxResult)()
+            self._lib.check(self._lib.lib.gxop_join_tail(self._op, C.byref(out)),
+                            "join_tail")
+            if not out:
+                break
+            chunks.append(self._lib.result_to_chunk(out))
+        return chunks
+
+    def close(self):
+        if self._op:
+            self._lib.lib.gxop_join_close(self._op)
+            self._op = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+class HashAggExec:
+    def __init__(self, lib, group_cols, aggs, input_types,
+                 expected_groups=0, device=-1, stream=0):
+        """aggs: list of (abi.COUNT_ROW.., input_col_or_-1)."""
+        self._lib = lib
+        self._keep = []
+        gc = (C.c_int32 * max(1, len(group_cols)))(*(group_cols or [0]))
+        sp = (GxAggSpec * max(1, len(aggs)))()
+        for i, (f, col) in enumerate(aggs):
+            sp[i] = GxAggSpec(f, col)
+        it = (C.c_int32 * len(input_types))(*input_types)
+        cfg = GxAggCfg(n_group_cols=len(group_cols), group_cols=gc,
+                       n_aggs=len(aggs), aggs=sp,
+                       n_input_cols=len(input_types), input_types=it,
+                       expected_groups=expected_groups, device=device,
+                       stream=stream)
+        self._keep += [gc, sp, it, cfg]
+        self._op = lib.lib.gxop_agg_create(C.byref(cfg))
+        if not self._op:
+            raise RuntimeError(f"gxop_agg_create: {lib.error()}")
+
+    def consume_chunk(self, chunk: Chunk):
+        ka = []
+        gch = self._lib.to_gx_chunk(chunk, ka)
+        self._lib.check(self._lib.lib.gxop_agg_consume(self._op, C.byref(gch)),
+                        "agg_consume")
+
+    def build_consume(self):
+        self._lib.check(self._lib.lib.gxop_agg_build(self._op), "agg_build")
+
+    def result_chunks(self):
+        chunks = []
+        while True:
+            out = C.POINTER(GxResult)()
+            self._lib.check(self._lib.lib.gxop_agg_next(self._op, C.byref(out)),
+                            "agg_next")
+            if not out:
+                break
+            chunks.append(self._lib.result_to_chunk(out))
+        return chunks
+
+    def close(self):
+        if self._op:
+            self._lib.lib.gxop_agg_close(self._op)
+            self._op = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+class PartitioningExchanger:
+    """Mirrors mpp/operator/PartitioningExchanger.java:71-134."""
+
+    def __init__(self, lib, n_parts, key_cols, input_types, device=-1, stream=0):
+        self._lib = lib
+        self._keep = []
+        self.n_parts = n_parts
+        kc = (C.c_int32 * len(key_cols))(*key_cols)
+        it = (C.c_int32 * len(input_types))(*input_types)
+        cfg = GxPartCfg(n_parts=n_parts, n_key_cols=len(key_cols), key_cols=kc,
+                        n_input_cols=len(input_types), input_types=it,
+                        device=device, stream=stream)
+        self._keep += [kc, it, cfg]
+        self._op = lib.lib.gxop_part_create(C.byref(cfg))
+        if not self._op:
+            raise RuntimeError(f"gxop_part_create: {lib.error()}")
+
+    def consume_chunk(self, chunk: Chunk):
+        """Returns list of n_parts chunks (None where empty)."""
+        ka = []
+        gc = self._lib.to_gx_chunk(chunk, ka)
+        outs = (C.POINTER(GxResult) * self.n_parts)()
+        self._lib.check(self._lib.lib.gxop_part_consume(self._op, C.byref(gc), outs),
+                        "part_consume")
+        res = []
+        for p in range(self.n_parts):
+            res.append(self._lib.result_to_chunk(outs[p]) if outs[p] else None)
+        return res
+
+    def close(self):
+        if self._op:
+            self._lib.lib.gxop_part_close(self._op)
+            self._op = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+def run_join(lib, join_type, join_keys, build_chunks, probe_chunks,
+             outer_types, inner_types, **kw):
+    """Convenience: full consume -> build -> probe -> tail pass; returns the
+    list of output Chunks (the shape SingleExecTest drives,
+    polardbx-executor/src/test/.../SingleExecTest.java:72-110)."""
+    op = ParallelHashJoinExec(lib, join_type, join_keys, outer_types,
+                              inner_types, **kw)
+    try:
+        for ch in build_chunks:
+            op.consume_chunk(ch)
+        op.build_consume()
+        out = []
+        for ch in probe_chunks:
+            r = op.probe_chunk(ch)
+            if r is not None and r.n_rows > 0:
+                out.append(r)
+        out.extend(op.tail_chunks())
+        return out
+    finally:
+        op.close()
+
+
+def run_agg(lib, group_cols, aggs, input_types, input_chunks, **kw):
+    op = HashAggExec(lib, group_cols, aggs, input_types, **kw)
+    try:
+        for ch in input_chunks:
+            op.consume_chunk(ch)
+        op.build_consume()
+        return op.result_chunks()
+    finally:
+        op.close()

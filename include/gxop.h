@@ -1,0 +1,217 @@
+/*
+ * gxop.h — C-ABI drop-in boundary for the PolarDB-X CN MPP hot path
+ * (ParallelHashJoinExec → HashAggExec → LocalExchanger PARTITION), built
+ * MI355X-native (HIP/gfx950 kernels behind this ABI).
+ *
+ * Each entry point replaces one piece of the reference operator API
+ * (paths under /root/reference/polardbx-executor/src/main/java/com/alibaba/
+ * polardbx/executor/ unless noted):
+ *
+ *  - gx_chunk / gx_block  mirror  Chunk = Block[] (chunk/Chunk.java:41-66)
+ *    with the physical layouts of LongBlock (chunk/LongBlock.java:41-55),
+ *    IntegerBlock (chunk/IntegerBlock.java:38-73), DoubleBlock, and
+ *    SliceBlock int[] offsets + byte data (chunk/SliceBlock.java:40-57).
+ *  - gxop_join_*  replace the ConsumerExecutor/Executor lifecycle of
+ *    ParallelHashJoinExec (operator/ParallelHashJoinExec.java:49):
+ *    consume = consumeChunk(:157-166), build = buildConsume(:107-128),
+ *    probe  = the doNextChunk probe loop (AbstractBufferedJoinExec.java:116+),
+ *    tail   = nextJoinNullRows (:372+, buildOuter unmatched emission).
+ *  - gx_join_cfg mirrors the factory args of
+ *    mpp/operator/factory/ParallelHashJoinExecutorFactory.java:77-117 and
+ *    EquiJoinKey (polardbx-optimizer/.../core/join/EquiJoinKey.java:25-43).
+ *  - gxop_agg_*   replace HashAggExec (operator/HashAggExec.java:133-145) +
+ *    AggOpenHashMap.putChunk/buildChunks (operator/util/AggOpenHashMap.java:
+ *    100-139,190-194).
+ *  - gxop_part_*  replace PartitioningExchanger.consumeChunk
+ *    (mpp/operator/PartitioningExchanger.java:71-134) with partition id =
+ *    ExecUtils.partition (utils/ExecUtils.java:1023-1033).
+ *
+ * Thread contract: one stream per operator instance;
+ * consume is callable from one thread at a time per instance; build is a
+ * barrier. Errors: negative int codes + gx_last_error() string (the Java
+ * shim maps them to TddlRuntimeException — see INTEGRATION.md).
+ */
+#ifndef GXOP_H
+#define GXOP_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- data model ------------------------------------------------------- */
+
+typedef enum gx_type {
+    GX_I64 = 0,   /* LongBlock:    int64 values[] + nulls[]                */
+    GX_I32 = 1,   /* IntegerBlock: int32 values[] + nulls[]                */
+    GX_F64 = 2,   /* DoubleBlock:  double values[] + nulls[]               */
+    GX_SLICE = 3  /* SliceBlock:   int32 end-offsets[] + byte data         */
+} gx_type;
+
+typedef enum gx_mem {
+    GX_MEM_HOST = 0,
+    GX_MEM_DEVICE = 1
+} gx_mem;
+
+/* One column of a batch. For GX_SLICE, offsets[i] is the END offset of
+ * value i in data (offset of value 0 starts at 0), as in SliceBlock. */
+typedef struct gx_block {
+    int32_t        type;      /* gx_type */
+    int32_t        mem;       /* gx_mem — where values/nulls/offsets/data live */
+    const void    *values;    /* i64[n] / i32[n] / f64[n]; NULL for GX_SLICE  */
+    const uint8_t *nulls;     /* u8[n], 1 = NULL; may be NULL (no nulls)      */
+    const int32_t *offsets;   /* GX_SLICE only */
+    const uint8_t *data;      /* GX_SLICE only */
+    int64_t        data_len;  /* GX_SLICE only: bytes in data */
+} gx_block;
+
+typedef struct gx_chunk {
+    int32_t         n_rows;
+    int32_t         n_blocks;
+    const gx_block *blocks;
+} gx_chunk;
+
+/* An output batch owned by the library; free with gxop_result_release.
+ * Blocks' mem tells where the data lives; gxop_result_to_host() copies a
+ * device-resident result into host memory in place. */
+typedef struct gx_result {
+    gx_chunk chunk;
+    void    *opaque;
+} gx_result;
+
+/* ---- join ------------------------------------------------------------- */
+
+/* JoinRelType subset honoured by ParallelHashJoinExec
+ * (operator/AbstractJoinExec.java:54-86). */
+typedef enum gx_join_type {
+    GX_JOIN_INNER = 0,
+    GX_JOIN_LEFT  = 1,
+    GX_JOIN_RIGHT = 2,
+    GX_JOIN_SEMI  = 3,
+    GX_JOIN_ANTI  = 4
+} gx_join_type;
+
+/* EquiJoinKey (EquiJoinKey.java:25-43). Key columns arrive already unified
+ * to unified_type by the factory's ChunkConverter (chunk/Converters.java:
+ * 35-80) — the library sees homogeneous key types. */
+typedef struct gx_equi_key {
+    int32_t outer_index;
+    int32_t inner_index;
+    int32_t unified_type;     /* gx_type */
+    int32_t null_safe_equal;  /* present in the reference but never consulted
+                                 on this path — must be 0 */
+} gx_equi_key;
+
+typedef struct gx_join_cfg {
+    int32_t join_type;        /* gx_join_type */
+    int32_t single_join;      /* maxOneRow: output = outer + first inner col;
+                                 >1 match per probe row is an error
+                                 (AbstractBufferedJoinExec.java:218-221) */
+    int32_t build_outer;      /* buildOuterInput (ParallelHashJoinExec.java:56):
+                                 the build side is the preserved/outer side;
+                                 unmatched build rows drain via gxop_join_tail */
+    int32_t n_keys;
+    const gx_equi_key *keys;
+    int32_t n_outer_cols;     /* outer = probe input unless build_outer */
+    const int32_t *outer_types;
+    int32_t n_inner_cols;
+    const int32_t *inner_types;
+    /* ANTI "x NOT IN (...)": probe column whose NULL suppresses emission
+     * (antiJoinOperands InputRef — AbstractBufferedJoinExec.java:247-252);
+     * -1 = none (NOT EXISTS shape). */
+    int32_t anti_null_col;
+    int32_t device;           /* HIP device ordinal; -1 only in the CPU oracle */
+    uint64_t stream;          /* hipStream_t, 0 = default stream */
+    int64_t expected_build_rows; /* size hint; 0 = unknown */
+} gx_join_cfg;
+
+typedef struct gx_op gx_op;  /* opaque operator instance */
+
+gx_op *gxop_join_create(const gx_join_cfg *cfg);
+/* consume one BUILD-side chunk (inner input unless build_outer). */
+int gxop_join_consume(gx_op *op, const gx_chunk *build_chunk);
+/* barrier: build the hash table (Synchronizer.buildHashTable:406-426). */
+int gxop_join_build(gx_op *op);
+/* probe with one chunk; *out receives the joined rows for this chunk
+ * (may be empty; may exceed 1000 rows — the shim re-chunks). */
+int gxop_join_probe(gx_op *op, const gx_chunk *probe_chunk, gx_result **out);
+/* drain pass-through / outer-null tail rows after the last probe chunk;
+ * returns 0 and *out=NULL when exhausted. */
+int gxop_join_tail(gx_op *op, gx_result **out);
+int gxop_join_close(gx_op *op);
+
+/* ---- hash aggregation ------------------------------------------------- */
+
+/* Aggregator subset (calc/aggfunctions/*, selection mirrors
+ * AggregateUtils.convertAggregators, operator/util/AggregateUtils.java:
+ * 147-220). Null/init semantics follow the named reference class. */
+typedef enum gx_agg_func {
+    GX_AGG_COUNT_ROW = 0,  /* CountRow: count(*)                  */
+    GX_AGG_COUNT_COL = 1,  /* Count: count non-null               */
+    GX_AGG_SUM_I64   = 2,  /* Long2LongSum0: init 0, add non-null */
+    GX_AGG_SUM_F64   = 3,  /* Double2DoubleSum: init NULL         */
+    GX_AGG_MIN_I64   = 4,  /* Long2LongMin: init NULL             */
+    GX_AGG_MAX_I64   = 5,  /* Long2LongMax: init NULL             */
+    GX_AGG_MIN_F64   = 6,  /* Double2DoubleMin: init NULL         */
+    GX_AGG_MAX_F64   = 7   /* Double2DoubleMax: init NULL         */
+} gx_agg_func;
+
+typedef struct gx_agg_spec {
+    int32_t func;        /* gx_agg_func */
+    int32_t input_col;   /* column in the input chunk; -1 for COUNT_ROW */
+} gx_agg_spec;
+
+typedef struct gx_agg_cfg {
+    int32_t n_group_cols;         /* 0 = global aggregate (one group) */
+    const int32_t *group_cols;    /* indices into the input chunk */
+    int32_t n_aggs;
+    const gx_agg_spec *aggs;
+    int32_t n_input_cols;
+    const int32_t *input_types;   /* gx_type per input column */
+    int64_t expected_groups;      /* size hint; 0 = unknown */
+    int32_t device;
+    uint64_t stream;
+} gx_agg_cfg;
+
+gx_op *gxop_agg_create(const gx_agg_cfg *cfg);
+int gxop_agg_consume(gx_op *op, const gx_chunk *input_chunk);
+int gxop_agg_build(gx_op *op);
+/* emit result batches: group-key columns then one column per aggregator
+ * (HashAggResultIterator). Returns 0 and *out=NULL when exhausted. */
+int gxop_agg_next(gx_op *op, gx_result **out);
+int gxop_agg_close(gx_op *op);
+
+/* ---- partition exchange ------------------------------------------------ */
+
+typedef struct gx_part_cfg {
+    int32_t n_parts;
+    int32_t n_key_cols;
+    const int32_t *key_cols;      /* partitionChannels */
+    int32_t n_input_cols;
+    const int32_t *input_types;
+    int32_t device;
+    uint64_t stream;
+} gx_part_cfg;
+
+gx_op *gxop_part_create(const gx_part_cfg *cfg);
+/* route one chunk: outs[p] receives partition p's rows (NULL if empty),
+ * partition id = ExecUtils.partition(rowHash) (ExecUtils.java:1023-1033).
+ * outs must have room for cfg->n_parts pointers. */
+int gxop_part_consume(gx_op *op, const gx_chunk *chunk, gx_result **outs);
+int gxop_part_close(gx_op *op);
+
+/* ---- results / errors -------------------------------------------------- */
+
+int gxop_result_to_host(gx_result *res);
+void gxop_result_release(gx_result *res);
+const char *gx_last_error(void);
+
+/* Library/ABI version + device sanity. Returns gfx arch or 0 for CPU lib. */
+int gxop_abi_version(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* GXOP_H */

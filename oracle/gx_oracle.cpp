@@ -1,0 +1,960 @@
+/*
+ * gx_oracle.cpp — CPU oracle: a C++ restatement of the PolarDB-X CN hot-path
+ * operator semantics (hash join / hash agg / partition exchange), used ONLY
+ * as the parity checker and the reported CPU baseline.
+ *
+ * TEST INFRASTRUCTURE — NOT THE PRODUCT PATH. Only tests/, __graft_entry__'s
+ * smoke() and bench.py's cpu_baseline leg may load this library. The product
+ * path is the HIP library (galaxysql_amd/csrc); it must fail loudly when its
+ * extension is missing, never fall back here.
+ *
+ * Every function cites the reference file:line it restates (paths under
+ * /root/reference/polardbx-executor/src/main/java/com/alibaba/polardbx/
+ * executor/ unless noted). Parity is pinned by golden vectors transcribed
+ * from the reference's own tests (tests/golden/*, from HashJoinTest.java and
+ * HashAggExecTest.java) — see tests/test_oracle_golden.py.
+ *
+ * Third-party arithmetic: fastutil HashCommon.mix / murmurHash3 / arraySize
+ * and airlift Slice.hashCode (XxHash64) are NOT vendored in the reference
+ * (transitive deps, version unpinned). They are restated here from their
+ * published algorithms; they affect bucket/partition PLACEMENT only, never
+ * result rows (SURVEY.md §8c "hash-placement parity: unpinned").
+ */
+#include "../include/gxop.h"
+
+#include <cstring>
+#include <cstdlib>
+#include <cstdio>
+#include <cmath>
+#include <string>
+#include <vector>
+#include <memory>
+#include <algorithm>
+
+namespace {
+
+thread_local std::string g_err;
+void set_err(const char *msg) { g_err = msg ? msg : ""; }
+
+/* ---- Java 32-bit arithmetic + hash functions --------------------------- */
+
+static inline int32_t jmul(int32_t a, int32_t b) {
+    return (int32_t)((uint32_t)a * (uint32_t)b);
+}
+static inline int32_t jadd(int32_t a, int32_t b) {
+    return (int32_t)((uint32_t)a + (uint32_t)b);
+}
+
+/* fastutil HashCommon.mix (int phi 0x9E3779B9, h ^= h >>> 16) —
+ * ConcurrentRawHashTable.java:93,114 uses it for bucket placement. */
+static inline int32_t hc_mix(int32_t x) {
+    int32_t h = jmul(x, (int32_t)0x9E3779B9);
+    return h ^ (int32_t)((uint32_t)h >> 16);
+}
+
+/* fastutil HashCommon.murmurHash3(int) — ExecUtils.partition
+ * (utils/ExecUtils.java:1023-1033). */
+static inline int32_t hc_murmur3(int32_t x) {
+    uint32_t h = (uint32_t)x;
+    h ^= h >> 16; h *= 0x85ebca6bu; h ^= h >> 13; h *= 0xc2b2ae35u; h ^= h >> 16;
+    return (int32_t)h;
+}
+
+/* fastutil HashCommon.arraySize(expected, f) = max(2, nextPow2(ceil(n/f))),
+ * used by ConcurrentRawHashTable.java:67-75 and GroupOpenHashMap.java. */
+static inline int64_t next_pow2(int64_t x) {
+    if (x <= 1) return 1;
+    x--;
+    x |= x >> 1; x |= x >> 2; x |= x >> 4; x |= x >> 8; x |= x >> 16; x |= x >> 32;
+    return x + 1;
+}
+static inline int64_t hc_array_size(int64_t expected, double f) {
+    int64_t s = next_pow2((int64_t)std::ceil((double)expected / f));
+    if (s < 2) s = 2;
+    return s;
+}
+/* fastutil HashCommon.maxFill(n, f) = min(ceil(n*f), n-1). */
+static inline int64_t hc_max_fill(int64_t n, double f) {
+    int64_t m = (int64_t)std::ceil((double)n * f);
+    return m < n - 1 ? m : n - 1;
+}
+
+/* ConcurrentRawHashTable.selectLoadFactor (ConcurrentRawHashTable.java:67-75):
+ * >=100M -> 0.75 (DEFAULT), >=10M -> 0.5 (FAST), else 0.25 (VERY_FAST). */
+static inline double join_load_factor(int64_t size) {
+    if (size >= 100000000) return 0.75;
+    if (size >= 10000000) return 0.5;
+    return 0.25;
+}
+
+/* Long.hashCode(v) = (int)(v ^ (v >>> 32)) — LongBlock.java:110-127. */
+static inline int32_t hash_i64(int64_t v) {
+    return (int32_t)((uint64_t)v ^ ((uint64_t)v >> 32));
+}
+/* IntegerBlock.hashCode = the value itself (IntegerBlock.java:112-117). */
+static inline int32_t hash_i32(int32_t v) { return v; }
+/* Double.hashCode = Long.hashCode(doubleToLongBits(v)) — DoubleBlock.java;
+ * doubleToLongBits canonicalizes NaN to 0x7ff8000000000000. */
+static inline int32_t hash_f64(double v) {
+    uint64_t bits;
+    if (std::isnan(v)) bits = 0x7ff8000000000000ull;
+    else std::memcpy(&bits, &v, 8);
+    return (int32_t)(bits ^ (bits >> 32));
+}
+
+/* airlift Slice.hashCode(offset,len) = (int) XxHash64.hash(seed=0) over the
+ * raw bytes (SliceBlock.java:183-195, non-"compatible" path). XxHash64
+ * restated from its published spec; placement-only, unpinned (see header). */
+static const uint64_t XXP1 = 0x9E3779B185EBCA87ull, XXP2 = 0xC2B2AE3D27D4EB4Full,
+                      XXP3 = 0x165667B19E3779F9ull, XXP4 = 0x85EBCA77C2B2AE63ull,
+                      XXP5 = 0x27D4EB2F165667C5ull;
+static inline uint64_t rotl64(uint64_t x, int r) { return (x << r) | (x >> (64 - r)); }
+static inline uint64_t xx_read64(const uint8_t *p) { uint64_t v; std::memcpy(&v, p, 8); return v; }
+static inline uint32_t xx_read32(const uint8_t *p) { uint32_t v; std::memcpy(&v, p, 4); return v; }
+static uint64_t xxhash64(const uint8_t *data, size_t len) {
+    const uint8_t *p = data, *end = data + len;
+    uint64_t h;
+    if (len >= 32) {
+        uint64_t v1 = XXP1 + XXP2, v2 = XXP2, v3 = 0, v4 = (uint64_t)0 - XXP1;
+        const uint8_t *limit = end - 32;
+        do {
+            v1 = rotl64(v1 + xx_read64(p) * XXP2, 31) * XXP1; p += 8;
+            v2 = rotl64(v2 + xx_read64(p) * XXP2, 31) * XXP1; p += 8;
+            v3 = rotl64(v3 + xx_read64(p) * XXP2, 31) * XXP1; p += 8;
+            v4 = rotl64(v4 + xx_read64(p) * XXP2, 31) * XXP1; p += 8;
+        } while (p <= limit);
+        h = rotl64(v1, 1) + rotl64(v2, 7) + rotl64(v3, 12) + rotl64(v4, 18);
+        auto merge = [&](uint64_t v) {
+            h ^= rotl64(v * XXP2, 31) * XXP1; h = h * XXP1 + XXP4;
+        };
+        merge(v1); merge(v2); merge(v3); merge(v4);
+    } else {
+        h = XXP5;
+    }
+    h += (uint64_t)len;
+    while (p + 8 <= end) {
+        h ^= rotl64(xx_read64(p) * XXP2, 31) * XXP1;
+        h = rotl64(h, 27) * XXP1 + XXP4; p += 8;
+    }
+    if (p + 4 <= end) {
+        h ^= (uint64_t)xx_read32(p) * XXP1;
+        h = rotl64(h, 23) * XXP2 + XXP3; p += 4;
+    }
+    while (p < end) {
+        h ^= (*p) * XXP5;
+        h = rotl64(h, 11) * XXP1; p++;
+    }
+    h ^= h >> 33; h *= XXP2; h ^= h >> 29; h *= XXP3; h ^= h >> 32;
+    return h;
+}
+
+/* ---- column store ------------------------------------------------------ */
+
+/* Host-side accumulation of consumed chunks, SoA per column — the oracle's
+ * ChunksIndex (operator/util/ChunksIndex.java:38-60) flattened to global
+ * positions (position = running row index across consumed chunks). */
+struct Column {
+    int32_t type = GX_I64;
+    std::vector<int64_t> i64v;
+    std::vector<int32_t> i32v;
+    std::vector<double>  f64v;
+    std::vector<int32_t> off;   /* slice end-offsets (global) */
+    std::vector<uint8_t> bytes; /* slice data */
+    std::vector<uint8_t> null_; /* 1 = NULL */
+
+    size_t size() const { return null_.size(); }
+
+    void append(const gx_block *b, int32_t n) {
+        size_t old = null_.size();
+        null_.resize(old + n, 0);
+        if (b->nulls) std::memcpy(null_.data() + old, b->nulls, n);
+        switch (b->type) {
+        case GX_I64: {
+            i64v.resize(old + n);
+            std::memcpy(i64v.data() + old, b->values, (size_t)n * 8);
+            break; }
+        case GX_I32: {
+            i32v.resize(old + n);
+            std::memcpy(i32v.data() + old, b->values, (size_t)n * 4);
+            break; }
+        case GX_F64: {
+            f64v.resize(old + n);
+            std::memcpy(f64v.data() + old, b->values, (size_t)n * 8);
+            break; }
+        case GX_SLICE: {
+            int32_t base = (int32_t)bytes.size();
+            int64_t blen = (n > 0) ? b->offsets[n - 1] : 0;
+            bytes.insert(bytes.end(), b->data, b->data + blen);
+            off.reserve(off.size() + n);
+            for (int32_t i = 0; i < n; i++) off.push_back(base + b->offsets[i]);
+            break; }
+        }
+    }
+
+    bool is_null(size_t i) const { return null_[i] != 0; }
+    int32_t begin_off(size_t i) const {
+        /* global begin offset of slice value i */
+        if (i == 0) return 0;
+        /* find the previous row's end; rows are global so previous end works
+         * only within the flattened store where offsets are cumulative */
+        return off[i - 1];
+    }
+
+    /* Block.hashCode(position): null -> 0 (LongBlock.java:110-127 et al). */
+    int32_t hash_at(size_t i) const {
+        if (is_null(i)) return 0;
+        switch (type) {
+        case GX_I64: return hash_i64(i64v[i]);
+        case GX_I32: return hash_i32(i32v[i]);
+        case GX_F64: return hash_f64(f64v[i]);
+        case GX_SLICE: {
+            int32_t b = begin_off(i), e = off[i];
+            return (int32_t)xxhash64(bytes.data() + b, (size_t)(e - b));
+        }
+        }
+        return 0;
+    }
+
+    /* Block.equals(position, other, otherPosition): null==null true, else
+     * value equality (LongBlock.java:69-127, IntegerBlock.java:102-166,
+     * DoubleBlock, SliceBlock byte compare). */
+    bool equal_at(size_t i, const Column &o, size_t j) const {
+        bool n1 = is_null(i), n2 = o.is_null(j);
+        if (n1 && n2) return true;
+        if (n1 != n2) return false;
+        switch (type) {
+        case GX_I64: return i64v[i] == o.i64v[j];
+        case GX_I32: return i32v[i] == o.i32v[j];
+        case GX_F64: return f64v[i] == o.f64v[j]; /* Java '==' on double */
+        case GX_SLICE: {
+            int32_t b1 = begin_off(i), e1 = off[i];
+            int32_t b2 = o.begin_off(j), e2 = o.off[j];
+            if (e1 - b1 != e2 - b2) return false;
+            return std::memcmp(bytes.data() + b1, o.bytes.data() + b2,
+                               (size_t)(e1 - b1)) == 0;
+        }
+        }
+        return false;
+    }
+};
+
+struct Store {
+    std::vector<Column> cols;
+    size_t n_rows = 0;
+    void init(int32_t n_cols, const int32_t *types) {
+        cols.resize(n_cols);
+        for (int32_t c = 0; c < n_cols; c++) cols[c].type = types[c];
+    }
+    int append(const gx_chunk *ch) {
+        if ((size_t)ch->n_blocks != cols.size()) { set_err("column count mismatch"); return -1; }
+        for (int32_t c = 0; c < ch->n_blocks; c++) {
+            if (ch->blocks[c].mem != GX_MEM_HOST) { set_err("oracle accepts host memory only"); return -1; }
+            if (ch->blocks[c].type != cols[c].type) { set_err("column type mismatch"); return -1; }
+            cols[c].append(&ch->blocks[c], ch->n_rows);
+        }
+        n_rows += ch->n_rows;
+        return 0;
+    }
+    /* Chunk.hashCode(position): h = 31*h + block.hashCode(position)
+     * (chunk/Chunk.java:116-129), over the given columns in order. */
+    int32_t row_hash(size_t row, const std::vector<int> &key_cols) const {
+        int32_t h = 0;
+        for (int c : key_cols) h = jadd(jmul(h, 31), cols[c].hash_at(row));
+        return h;
+    }
+    bool row_has_null(size_t row, const std::vector<int> &key_cols) const {
+        for (int c : key_cols) if (cols[c].is_null(row)) return true;
+        return false;
+    }
+    bool keys_equal(size_t row, const Store &o, size_t orow,
+                    const std::vector<int> &my_cols, const std::vector<int> &o_cols) const {
+        for (size_t k = 0; k < my_cols.size(); k++)
+            if (!cols[my_cols[k]].equal_at(row, o.cols[o_cols[k]], orow)) return false;
+        return true;
+    }
+};
+
+/* ---- output builder ---------------------------------------------------- */
+
+struct OutCol {
+    int32_t type;
+    std::vector<int64_t> i64v;
+    std::vector<int32_t> i32v;
+    std::vector<double>  f64v;
+    std::vector<int32_t> off;
+    std::vector<uint8_t> bytes;
+    std::vector<uint8_t> null_;
+
+    void append_null() {
+        null_.push_back(1);
+        switch (type) {
+        case GX_I64: i64v.push_back(0); break;
+        case GX_I32: i32v.push_back(0); break;
+        case GX_F64: f64v.push_back(0); break;
+        case GX_SLICE: off.push_back((int32_t)bytes.size()); break;
+        }
+    }
+    void append_from(const Column &c, size_t i) {
+        if (c.is_null(i)) { append_null(); return; }
+        null_.push_back(0);
+        switch (type) {
+        case GX_I64: i64v.push_back(c.i64v[i]); break;
+        case GX_I32: i32v.push_back(c.i32v[i]); break;
+        case GX_F64: f64v.push_back(c.f64v[i]); break;
+        case GX_SLICE: {
+            int32_t b = c.begin_off(i), e = c.off[i];
+            bytes.insert(bytes.end(), c.bytes.data() + b, c.bytes.data() + e);
+            off.push_back((int32_t)bytes.size());
+            break; }
+        }
+    }
+    void append_i64(int64_t v) { null_.push_back(0); i64v.push_back(v); }
+    void append_f64(double v)  { null_.push_back(0); f64v.push_back(v); }
+};
+
+struct ResultHolder {
+    std::vector<OutCol> cols;
+    std::vector<gx_block> blocks;
+    gx_result res;
+};
+
+static gx_result *make_result(std::vector<OutCol> &&cols, int32_t n_rows) {
+    auto *h = new ResultHolder();
+    h->cols = std::move(cols);
+    h->blocks.resize(h->cols.size());
+    for (size_t c = 0; c < h->cols.size(); c++) {
+        OutCol &oc = h->cols[c];
+        gx_block &b = h->blocks[c];
+        std::memset(&b, 0, sizeof(b));
+        b.type = oc.type;
+        b.mem = GX_MEM_HOST;
+        b.nulls = oc.null_.data();
+        switch (oc.type) {
+        case GX_I64: b.values = oc.i64v.data(); break;
+        case GX_I32: b.values = oc.i32v.data(); break;
+        case GX_F64: b.values = oc.f64v.data(); break;
+        case GX_SLICE:
+            b.offsets = oc.off.data();
+            b.data = oc.bytes.data();
+            b.data_len = (int64_t)oc.bytes.size();
+            break;
+        }
+    }
+    h->res.chunk.n_rows = n_rows;
+    h->res.chunk.n_blocks = (int32_t)h->blocks.size();
+    h->res.chunk.blocks = h->blocks.data();
+    h->res.opaque = h;
+    return &h->res;
+}
+
+/* ---- operator base ----------------------------------------------------- */
+
+enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3 };
+
+} // anonymous namespace
+
+struct gx_op {
+    int kind;
+    virtual ~gx_op() = default;
+protected:
+    explicit gx_op(int k) : kind(k) {}
+};
+
+namespace {
+
+/* ---- join operator ------------------------------------------------------
+ * Restates ParallelHashJoinExec + AbstractBufferedJoinExec.nextRows
+ * (AbstractBufferedJoinExec.java:185-266) + AbstractHashJoinExec.matchInit/
+ * matchNext (AbstractHashJoinExec.java:80-106) + Synchronizer.buildHashTable
+ * (ParallelHashJoinExec.java:406-426) with numPartitions=1 (single-threaded:
+ * insertion order = consume order, as in the reference's SingleExecTest). */
+struct JoinOp : gx_op {
+    gx_join_cfg cfg;
+    std::vector<gx_equi_key> keys;
+    std::vector<int32_t> outer_types, inner_types;
+
+    Store build;                 /* build-side input columns */
+    std::vector<int> build_key_cols, probe_key_cols;
+    std::vector<int32_t> key_types;
+
+    /* hash table: bucket = mix(hash) & mask holds LATEST inserted position
+     * (ConcurrentRawHashTable.put = getAndSet, java:100-121); chains via
+     * position_links (head insertion, ExecUtils.buildOneChunk:914-944). */
+    std::vector<int32_t> table;
+    std::vector<int32_t> links;
+    int32_t mask = 0;
+    bool built = false;
+
+    bool pass_nothing = false, pass_through = false;
+    /* buildOuter unmatched tracking (Synchronizer.joinNullRowBitSet,
+     * ParallelHashJoinExec.java:437-461). */
+    std::vector<uint8_t> matched_build;
+    size_t tail_cursor = 0;
+    bool tail_done = false;
+
+    JoinOp(const gx_join_cfg *c) : gx_op(OP_JOIN), cfg(*c) {
+        keys.assign(c->keys, c->keys + c->n_keys);
+        outer_types.assign(c->outer_types, c->outer_types + c->n_outer_cols);
+        inner_types.assign(c->inner_types, c->inner_types + c->n_inner_cols);
+        for (auto &k : keys) {
+            /* build side = inner unless build_outer (AbstractBufferedJoinExec
+             * getBuildInput/getProbeInput; buildOuter flips them) */
+            build_key_cols.push_back(cfg.build_outer ? k.outer_index : k.inner_index);
+            probe_key_cols.push_back(cfg.build_outer ? k.inner_index : k.outer_index);
+            key_types.push_back(k.unified_type);
+        }
+        const auto &bt = cfg.build_outer ? outer_types : inner_types;
+        build.init((int32_t)bt.size(), bt.data());
+    }
+
+    int consume(const gx_chunk *ch) { return build.append(ch); }
+
+    int do_build() {
+        const int64_t size = (int64_t)build.n_rows;
+        int64_t n = hc_array_size(size, join_load_factor(size));
+        mask = (int32_t)(n - 1);
+        table.assign((size_t)n, -1);
+        links.assign((size_t)size, -1);
+        /* ExecUtils.buildOneChunk (utils/ExecUtils.java:914-944): rows with a
+         * NULL in ANY key column are skipped unconditionally (:933-941). */
+        for (size_t pos = 0; pos < build.n_rows; pos++) {
+            if (build.row_has_null(pos, build_key_cols)) continue;
+            int32_t h = build.row_hash(pos, build_key_cols);
+            int32_t slot = hc_mix(h) & mask;
+            links[pos] = table[slot];
+            table[slot] = (int32_t)pos;
+        }
+        built = true;
+
+        /* special modes (ParallelHashJoinExec.buildConsume:107-128 +
+         * AbstractBufferedJoinExec.doSpecialCheckForSemiJoin:293-316) */
+        bool semi_join = (cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI)
+                         && !cfg.single_join;
+        if (build.n_rows == 0 && cfg.join_type == GX_JOIN_INNER) pass_nothing = true;
+        if (semi_join && build.n_rows == 0) {
+            if (cfg.join_type == GX_JOIN_SEMI) pass_nothing = true;
+            else pass_through = true; /* ANTI with empty build passes all */
+        } else if (cfg.join_type == GX_JOIN_ANTI && cfg.anti_null_col >= 0
+                   && build.cols.size() == 1) {
+            /* x NOT IN (...NULL...) -> empty result
+             * (doSpecialCheckForSemiJoin:305-312) */
+            for (size_t i = 0; i < build.n_rows && !pass_nothing; i++)
+                if (build.cols[0].is_null(i)) pass_nothing = true;
+        }
+        if (cfg.build_outer) matched_build.assign(build.n_rows, 0);
+        return 0;
+    }
+
+    int32_t match_init(int32_t h) const {
+        int32_t m = table[hc_mix(h) & mask];
+        return m;
+    }
+
+    /* emit one joined row into out columns. Column order per
+     * AbstractJoinExec.buildJoinRow/buildRightJoinRow (java:175-227) and the
+     * buildOuter overrides (ParallelHashJoinExec.java:234-287). */
+    void emit_match(std::vector<OutCol> &out, const Store &probe_store,
+                    size_t prow, int32_t bpos) {
+        size_t col = 0;
+        const size_t n_outer = outer_types.size();
+        const size_t n_inner_out = cfg.single_join ? 1 : inner_types.size();
+        if (cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI) {
+            for (size_t i = 0; i < n_outer; i++)
+                out[col++].append_from(probe_store.cols[i], prow);
+            return;
+        }
+        if (cfg.join_type != GX_JOIN_RIGHT) {
+            /* outer cols then inner cols */
+            if (!cfg.build_outer) {
+                for (size_t i = 0; i < n_outer; i++)
+                    out[col++].append_from(probe_store.cols[i], prow);
+                for (size_t i = 0; i < n_inner_out; i++)
+                    out[col++].append_from(build.cols[i], (size_t)bpos);
+            } else {
+                for (size_t i = 0; i < n_outer; i++)
+                    out[col++].append_from(build.cols[i], (size_t)bpos);
+                for (size_t i = 0; i < n_inner_out; i++)
+                    out[col++].append_from(probe_store.cols[i], prow);
+            }
+        } else {
+            /* RIGHT: inner cols then outer cols */
+            if (!cfg.build_outer) {
+                for (size_t i = 0; i < inner_types.size(); i++)
+                    out[col++].append_from(build.cols[i], (size_t)bpos);
+                for (size_t i = 0; i < n_outer; i++)
+                    out[col++].append_from(probe_store.cols[i], prow);
+            } else {
+                for (size_t i = 0; i < inner_types.size(); i++)
+                    out[col++].append_from(probe_store.cols[i], prow);
+                for (size_t i = 0; i < n_outer; i++)
+                    out[col++].append_from(build.cols[i], (size_t)bpos);
+            }
+        }
+    }
+
+    /* LEFT/RIGHT unmatched probe row (AbstractJoinExec.buildLeftNullRow /
+     * buildRightNullRow, java:203-227). Only reachable when !build_outer
+     * (outputNullRowInTime). */
+    void emit_null_row(std::vector<OutCol> &out, const Store &probe_store, size_t prow) {
+        size_t col = 0;
+        const size_t n_inner_out = cfg.single_join ? 1 : inner_types.size();
+        if (cfg.join_type != GX_JOIN_RIGHT) {
+            for (size_t i = 0; i < outer_types.size(); i++)
+                out[col++].append_from(probe_store.cols[i], prow);
+            for (size_t i = 0; i < n_inner_out; i++)
+                out[col++].append_null();
+        } else {
+            for (size_t i = 0; i < inner_types.size(); i++)
+                out[col++].append_null();
+            for (size_t i = 0; i < outer_types.size(); i++)
+                out[col++].append_from(probe_store.cols[i], prow);
+        }
+    }
+
+    std::vector<int32_t> output_types() const {
+        std::vector<int32_t> t;
+        if (cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI)
+            return outer_types;
+        const std::vector<int32_t> inner_out = cfg.single_join
+            ? std::vector<int32_t>(inner_types.begin(), inner_types.begin() + 1)
+            : inner_types;
+        if (cfg.join_type != GX_JOIN_RIGHT) {
+            t = outer_types;
+            t.insert(t.end(), inner_out.begin(), inner_out.end());
+        } else {
+            t = inner_types;
+            t.insert(t.end(), outer_types.begin(), outer_types.end());
+        }
+        return t;
+    }
+
+    int probe(const gx_chunk *ch, gx_result **out) {
+        *out = nullptr;
+        if (!built) { set_err("probe before build"); return -1; }
+
+        std::vector<int32_t> otypes = output_types();
+        /* probe side column types: outer unless build_outer */
+        const auto &pt = cfg.build_outer ? inner_types : outer_types;
+        Store probe_store;
+        probe_store.init((int32_t)pt.size(), pt.data());
+        if (probe_store.append(ch) != 0) return -1;
+
+        std::vector<OutCol> cols(otypes.size());
+        for (size_t c = 0; c < otypes.size(); c++) cols[c].type = otypes[c];
+
+        if (pass_nothing) { *out = make_result(std::move(cols), 0); return 0; }
+        if (pass_through) {
+            /* ANTI with empty build: every probe row passes, no operand check
+             * (AbstractBufferedJoinExec.doSpecialCheckForSemiJoin:296-301) */
+            for (size_t r = 0; r < probe_store.n_rows; r++)
+                emit_match(cols, probe_store, r, -1);
+            *out = make_result(std::move(cols), (int32_t)probe_store.n_rows);
+            return 0;
+        }
+
+        bool semi_join = (cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI)
+                         && !cfg.single_join;
+        bool outer_join = (cfg.join_type == GX_JOIN_LEFT || cfg.join_type == GX_JOIN_RIGHT);
+        int32_t n_rows = 0;
+
+        for (size_t r = 0; r < probe_store.n_rows; r++) {
+            int32_t h = probe_store.row_hash(r, probe_key_cols);
+            bool matched = false;
+            int32_t m = match_init(h);
+            for (; m != -1; m = links[m]) {
+                if (!build.keys_equal((size_t)m, probe_store, r,
+                                      build_key_cols, probe_key_cols))
+                    continue;
+                /* NOTE: probe rows with a NULL key can never reach here —
+                 * build NULL keys were never inserted and Block.equals(null,
+                 * value) is false; null==null would match but null build keys
+                 * are skipped (ExecUtils.buildOneChunk:933-941). */
+                if (cfg.join_type == GX_JOIN_INNER || cfg.join_type == GX_JOIN_LEFT) {
+                    if (cfg.single_join && matched) {
+                        set_err("ERR_SCALAR_SUBQUERY_RETURN_MORE_THAN_ONE_ROW");
+                        return -2;
+                    }
+                    emit_match(cols, probe_store, r, m);
+                    n_rows++;
+                } else if (cfg.join_type == GX_JOIN_RIGHT) {
+                    emit_match(cols, probe_store, r, m);
+                    n_rows++;
+                }
+                if (cfg.build_outer) matched_build[(size_t)m] = 1;
+                matched = true;
+                if (semi_join) break;
+            }
+            if (outer_join && !cfg.build_outer && !matched) {
+                emit_null_row(cols, probe_store, r);
+                n_rows++;
+            }
+            if (cfg.single_join && !semi_join && !matched
+                && cfg.join_type == GX_JOIN_LEFT) {
+                /* covered by outer_join branch above */
+            }
+            if (semi_join) {
+                if (cfg.join_type == GX_JOIN_SEMI && matched) {
+                    emit_match(cols, probe_store, r, -1);
+                    n_rows++;
+                } else if (cfg.join_type == GX_JOIN_ANTI && !matched) {
+                    /* checkAntiJoinOperands: operand NULL suppresses the row
+                     * (AbstractBufferedJoinExec.java:247-252,268-271) */
+                    bool ok = true;
+                    if (cfg.anti_null_col >= 0)
+                        ok = !probe_store.cols[cfg.anti_null_col].is_null(r);
+                    if (ok) {
+                        emit_match(cols, probe_store, r, -1);
+                        n_rows++;
+                    }
+                }
+            }
+        }
+        *out = make_result(std::move(cols), n_rows);
+        return 0;
+    }
+
+    /* drain buildOuter unmatched rows (ParallelHashJoinExec.nextJoinNullRows
+     * :372-401): JoinRelType != RIGHT -> build cols + nulls; RIGHT -> nulls +
+     * build cols. */
+    int tail(gx_result **out) {
+        *out = nullptr;
+        if (!cfg.build_outer || tail_done) { tail_done = true; return 0; }
+        std::vector<int32_t> otypes = output_types();
+        std::vector<OutCol> cols(otypes.size());
+        for (size_t c = 0; c < otypes.size(); c++) cols[c].type = otypes[c];
+        int32_t n_rows = 0;
+        const size_t n_inner_out = cfg.single_join ? 1 : inner_types.size();
+        for (size_t pos = tail_cursor; pos < build.n_rows; pos++) {
+            if (matched_build[pos]) continue;
+            size_t col = 0;
+            if (cfg.join_type != GX_JOIN_RIGHT) {
+                for (size_t i = 0; i < outer_types.size(); i++)
+                    cols[col++].append_from(build.cols[i], pos);
+                for (size_t i = 0; i < n_inner_out; i++)
+                    cols[col++].append_null();
+            } else {
+                for (size_t i = 0; i < inner_types.size(); i++)
+                    cols[col++].append_null();
+                for (size_t i = 0; i < outer_types.size(); i++)
+                    cols[col++].append_from(build.cols[i], pos);
+            }
+            n_rows++;
+        }
+        tail_cursor = build.n_rows;
+        tail_done = true;
+        if (n_rows > 0) *out = make_result(std::move(cols), n_rows);
+        return 0;
+    }
+};
+
+/* ---- agg operator -------------------------------------------------------
+ * Restates HashAggExec.consumeChunk (operator/HashAggExec.java:133-145) →
+ * AggOpenHashMap.putChunk (operator/util/AggOpenHashMap.java:100-139) with
+ * the GroupOpenHashMap open-addressing array path (GroupOpenHashMap.java:
+ * 142-193: linear probe on mix(hash)&mask, x2 rehash at maxFill). */
+struct AggOp : gx_op {
+    gx_agg_cfg cfg;
+    std::vector<int32_t> group_cols_, input_types;
+    std::vector<gx_agg_spec> aggs;
+
+    Store input;                     /* group-key columns only, appended per group */
+    Store group_keys;                /* TypedBuffer: one row per group */
+    std::vector<int> key_col_idx;    /* 0..n_group_cols-1 into group_keys */
+
+    std::vector<int32_t> table;      /* open addressing: groupId or -1 */
+    int64_t n_slots = 0, mask = 0, max_fill = 0, fill = 0;
+    double load_f = 0.75;            /* GroupOpenHashMap DEFAULT_LOAD_FACTOR */
+
+    /* aggregator states, one entry per group per agg */
+    struct AggState { std::vector<int64_t> i64; std::vector<double> f64; std::vector<uint8_t> isnull; };
+    std::vector<AggState> states;
+
+    bool built = false;
+    size_t emit_cursor = 0;
+    static constexpr int32_t CHUNK_SIZE = 1000; /* ConnectionParams.java:1088 */
+
+    AggOp(const gx_agg_cfg *c) : gx_op(OP_AGG), cfg(*c) {
+        group_cols_.assign(c->group_cols, c->group_cols + c->n_group_cols);
+        input_types.assign(c->input_types, c->input_types + c->n_input_cols);
+        aggs.assign(c->aggs, c->aggs + c->n_aggs);
+
+        std::vector<int32_t> key_types;
+        for (int32_t gc : group_cols_) key_types.push_back(input_types[gc]);
+        group_keys.init((int32_t)key_types.size(), key_types.data());
+        for (size_t k = 0; k < key_types.size(); k++) key_col_idx.push_back((int)k);
+
+        int64_t expected = cfg.expected_groups > 0 ? cfg.expected_groups : 1024;
+        n_slots = hc_array_size(expected, load_f);
+        mask = n_slots - 1;
+        max_fill = hc_max_fill(n_slots, load_f);
+        table.assign((size_t)n_slots, -1);
+
+        states.resize(aggs.size());
+        if (cfg.n_group_cols == 0) append_group_direct(); /* noGroupBy: one group */
+    }
+
+    int32_t n_groups() const { return (int32_t)(states.empty()
+        ? group_keys.n_rows : states[0].isnull.size()); }
+
+    void append_group_direct() {
+        for (size_t a = 0; a < aggs.size(); a++) {
+            AggState &s = states[a];
+            switch (aggs[a].func) {
+            case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
+                s.i64.push_back(0); s.isnull.push_back(0); break;
+            case GX_AGG_SUM_I64:
+                s.i64.push_back(0); s.isnull.push_back(0); break; /* Long2LongSum0 init 0 */
+            case GX_AGG_SUM_F64: case GX_AGG_MIN_F64: case GX_AGG_MAX_F64:
+                s.f64.push_back(0); s.isnull.push_back(1); break; /* init NULL */
+            case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+                s.i64.push_back(0); s.isnull.push_back(1); break; /* init NULL */
+            }
+        }
+    }
+
+    /* GroupOpenHashMap.doInnerPutArray (java:142-169) + rehash (:171-187).
+     * 'from' = store+row providing the key (input chunk or group_keys on
+     * rehash); group id -1 = allocate. */
+    int32_t inner_put(const Store &from, size_t row, const std::vector<int> &from_cols,
+                      int32_t group_id) {
+        int32_t h = (int32_t)(hc_mix(from.row_hash(row, from_cols)) & (int32_t)mask);
+        while (true) {
+            int32_t k = table[(size_t)h];
+            if (k == -1) break;
+            if (group_keys.keys_equal((size_t)k, from, row, key_col_idx, from_cols))
+                return k;
+            h = (int32_t)((h + 1) & mask);
+        }
+        if (group_id == -1) {
+            /* appendGroup: copy key into TypedBuffer, assign next id */
+            for (size_t kc = 0; kc < from_cols.size(); kc++) {
+                const Column &src = from.cols[from_cols[kc]];
+                Column &dst = group_keys.cols[kc];
+                dst.null_.push_back(src.is_null(row) ? 1 : 0);
+                switch (src.type) {
+                case GX_I64: dst.i64v.push_back(src.is_null(row) ? 0 : src.i64v[row]); break;
+                case GX_I32: dst.i32v.push_back(src.is_null(row) ? 0 : src.i32v[row]); break;
+                case GX_F64: dst.f64v.push_back(src.is_null(row) ? 0 : src.f64v[row]); break;
+                case GX_SLICE: {
+                    if (!src.is_null(row)) {
+                        int32_t b = src.begin_off(row), e = src.off[row];
+                        dst.bytes.insert(dst.bytes.end(), src.bytes.data() + b, src.bytes.data() + e);
+                    }
+                    dst.off.push_back((int32_t)dst.bytes.size());
+                    break; }
+                }
+            }
+            group_keys.n_rows++;
+            group_id = (int32_t)group_keys.n_rows - 1;
+            append_group_direct();
+        }
+        table[(size_t)h] = group_id;
+        if (fill++ >= max_fill) rehash();
+        return group_id;
+    }
+
+    void rehash() {
+        n_slots *= 2; mask = n_slots - 1;
+        max_fill = hc_max_fill(n_slots, load_f);
+        fill = 0;
+        table.assign((size_t)n_slots, -1);
+        std::vector<int> kcols(key_col_idx.begin(), key_col_idx.end());
+        for (size_t g = 0; g < group_keys.n_rows; g++)
+            inner_put(group_keys, g, kcols, (int32_t)g);
+    }
+
+    void accumulate(size_t a, int32_t gid, const Store &in, size_t row) {
+        const gx_agg_spec &sp = aggs[a];
+        AggState &s = states[a];
+        const Column *c = sp.input_col >= 0 ? &in.cols[sp.input_col] : nullptr;
+        switch (sp.func) {
+        case GX_AGG_COUNT_ROW: s.i64[gid]++; break;
+        case GX_AGG_COUNT_COL: if (!c->is_null(row)) s.i64[gid]++; break;
+        case GX_AGG_SUM_I64:
+            if (!c->is_null(row))
+                s.i64[gid] = (int64_t)((uint64_t)s.i64[gid] +
+                    (uint64_t)(c->type == GX_I32 ? (int64_t)c->i32v[row] : c->i64v[row]));
+            break;
+        case GX_AGG_SUM_F64:
+            if (!c->is_null(row)) {
+                double v = c->type == GX_F64 ? c->f64v[row]
+                         : c->type == GX_I32 ? (double)c->i32v[row] : (double)c->i64v[row];
+                if (s.isnull[gid]) { s.f64[gid] = v; s.isnull[gid] = 0; }
+                else s.f64[gid] += v;
+            }
+            break;
+        case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            if (!c->is_null(row)) {
+                int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[row] : c->i64v[row];
+                if (s.isnull[gid]) { s.i64[gid] = v; s.isnull[gid] = 0; }
+                else s.i64[gid] = sp.func == GX_AGG_MIN_I64 ? std::min(s.i64[gid], v)
+                                                            : std::max(s.i64[gid], v);
+            }
+            break;
+        case GX_AGG_MIN_F64: case GX_AGG_MAX_F64:
+            if (!c->is_null(row)) {
+                double v = c->f64v[row];
+                if (s.isnull[gid]) { s.f64[gid] = v; s.isnull[gid] = 0; }
+                else s.f64[gid] = sp.func == GX_AGG_MIN_F64 ? std::min(s.f64[gid], v)
+                                                            : std::max(s.f64[gid], v);
+            }
+            break;
+        }
+    }
+
+    int consume(const gx_chunk *ch) {
+        Store in;
+        in.init((int32_t)input_types.size(), input_types.data());
+        if (in.append(ch) != 0) return -1;
+        std::vector<int> gcols(group_cols_.begin(), group_cols_.end());
+        for (size_t r = 0; r < in.n_rows; r++) {
+            int32_t gid = cfg.n_group_cols == 0 ? 0 : inner_put(in, r, gcols, -1);
+            for (size_t a = 0; a < aggs.size(); a++) accumulate(a, gid, in, r);
+        }
+        return 0;
+    }
+
+    int next(gx_result **out) {
+        *out = nullptr;
+        int32_t total = n_groups();
+        if ((int32_t)emit_cursor >= total) return 0;
+        int32_t n = std::min<int32_t>(CHUNK_SIZE, total - (int32_t)emit_cursor);
+        std::vector<int32_t> otypes;
+        for (int32_t gc : group_cols_) otypes.push_back(input_types[gc]);
+        for (auto &sp : aggs) {
+            switch (sp.func) {
+            case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
+            case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+                otypes.push_back(GX_I64); break;
+            default: otypes.push_back(GX_F64); break;
+            }
+        }
+        std::vector<OutCol> cols(otypes.size());
+        for (size_t c = 0; c < otypes.size(); c++) cols[c].type = otypes[c];
+        for (int32_t i = 0; i < n; i++) {
+            size_t g = emit_cursor + (size_t)i;
+            size_t col = 0;
+            for (size_t kc = 0; kc < group_cols_.size(); kc++)
+                cols[col++].append_from(group_keys.cols[kc], g);
+            for (size_t a = 0; a < aggs.size(); a++) {
+                AggState &s = states[a];
+                bool is_i64 = otypes[col] == GX_I64;
+                if (s.isnull[g] &&
+                    !(aggs[a].func == GX_AGG_COUNT_ROW || aggs[a].func == GX_AGG_COUNT_COL
+                      || aggs[a].func == GX_AGG_SUM_I64))
+                    cols[col++].append_null();
+                else if (is_i64) cols[col++].append_i64(s.i64[g]);
+                else cols[col++].append_f64(s.f64[g]);
+            }
+        }
+        emit_cursor += (size_t)n;
+        *out = make_result(std::move(cols), n);
+        return 0;
+    }
+};
+
+/* ---- partition operator -------------------------------------------------
+ * Restates PartitioningExchanger.consumeChunk (mpp/operator/
+ * PartitioningExchanger.java:71-134): row hash over key cols (HashBucketFunction
+ * shape, PartitionedOutputCollector.java:270-302) -> ExecUtils.partition. */
+struct PartOp : gx_op {
+    gx_part_cfg cfg;
+    std::vector<int32_t> key_cols_, input_types;
+    bool pow2;
+    PartOp(const gx_part_cfg *c) : gx_op(OP_PART), cfg(*c) {
+        key_cols_.assign(c->key_cols, c->key_cols + c->n_key_cols);
+        input_types.assign(c->input_types, c->input_types + c->n_input_cols);
+        pow2 = (cfg.n_parts & (-cfg.n_parts)) == cfg.n_parts;
+    }
+    int consume(const gx_chunk *ch, gx_result **outs) {
+        Store in;
+        in.init((int32_t)input_types.size(), input_types.data());
+        if (in.append(ch) != 0) return -1;
+        std::vector<int> kc(key_cols_.begin(), key_cols_.end());
+        std::vector<std::vector<size_t>> parts((size_t)cfg.n_parts);
+        for (size_t r = 0; r < in.n_rows; r++) {
+            int32_t h = in.row_hash(r, kc);
+            int32_t p;
+            if (pow2) p = hc_murmur3(h) & (cfg.n_parts - 1);
+            else p = (int32_t)(((uint32_t)hc_murmur3(h) & 0x7fffffffu) % (uint32_t)cfg.n_parts);
+            parts[(size_t)p].push_back(r);
+        }
+        for (int32_t p = 0; p < cfg.n_parts; p++) {
+            outs[p] = nullptr;
+            if (parts[(size_t)p].empty()) continue;
+            std::vector<OutCol> cols(input_types.size());
+            for (size_t c = 0; c < input_types.size(); c++) cols[c].type = input_types[c];
+            for (size_t r : parts[(size_t)p])
+                for (size_t c = 0; c < input_types.size(); c++)
+                    cols[c].append_from(in.cols[c], r);
+            outs[p] = make_result(std::move(cols), (int32_t)parts[(size_t)p].size());
+        }
+        return 0;
+    }
+};
+
+} // anonymous namespace
+
+/* ---- C ABI ------------------------------------------------------------- */
+
+extern "C" {
+
+gx_op *gxop_join_create(const gx_join_cfg *cfg) {
+    if (!cfg || cfg->n_keys <= 0) { set_err("bad join cfg"); return nullptr; }
+    return new JoinOp(cfg);
+}
+int gxop_join_consume(gx_op *op, const gx_chunk *c) {
+    if (!op || op->kind != OP_JOIN) { set_err("not a join op"); return -1; }
+    return static_cast<JoinOp *>(op)->consume(c);
+}
+int gxop_join_build(gx_op *op) {
+    if (!op || op->kind != OP_JOIN) { set_err("not a join op"); return -1; }
+    return static_cast<JoinOp *>(op)->do_build();
+}
+int gxop_join_probe(gx_op *op, const gx_chunk *c, gx_result **out) {
+    if (!op || op->kind != OP_JOIN) { set_err("not a join op"); return -1; }
+    return static_cast<JoinOp *>(op)->probe(c, out);
+}
+int gxop_join_tail(gx_op *op, gx_result **out) {
+    if (!op || op->kind != OP_JOIN) { set_err("not a join op"); return -1; }
+    return static_cast<JoinOp *>(op)->tail(out);
+}
+int gxop_join_close(gx_op *op) { delete op; return 0; }
+
+gx_op *gxop_agg_create(const gx_agg_cfg *cfg) {
+    if (!cfg) { set_err("bad agg cfg"); return nullptr; }
+    return new AggOp(cfg);
+}
+int gxop_agg_consume(gx_op *op, const gx_chunk *c) {
+    if (!op || op->kind != OP_AGG) { set_err("not an agg op"); return -1; }
+    return static_cast<AggOp *>(op)->consume(c);
+}
+int gxop_agg_build(gx_op *op) {
+    if (!op || op->kind != OP_AGG) { set_err("not an agg op"); return -1; }
+    static_cast<AggOp *>(op)->built = true;
+    return 0;
+}
+int gxop_agg_next(gx_op *op, gx_result **out) {
+    if (!op || op->kind != OP_AGG) { set_err("not an agg op"); return -1; }
+    return static_cast<AggOp *>(op)->next(out);
+}
+int gxop_agg_close(gx_op *op) { delete op; return 0; }
+
+gx_op *gxop_part_create(const gx_part_cfg *cfg) {
+    if (!cfg || cfg->n_parts <= 0) { set_err("bad part cfg"); return nullptr; }
+    return new PartOp(cfg);
+}
+int gxop_part_consume(gx_op *op, const gx_chunk *c, gx_result **outs) {
+    if (!op || op->kind != OP_PART) { set_err("not a part op"); return -1; }
+    return static_cast<PartOp *>(op)->consume(c, outs);
+}
+int gxop_part_close(gx_op *op) { delete op; return 0; }
+
+int gxop_result_to_host(gx_result *) { return 0; /* oracle results are host */ }
+void gxop_result_release(gx_result *res) {
+    if (res) delete static_cast<ResultHolder *>(res->opaque);
+}
+const char *gx_last_error(void) { return g_err.c_str(); }
+int gxop_abi_version(void) { return 0; /* CPU oracle */ }
+
+} /* extern "C" */
