@@ -517,6 +517,14 @@ struct JoinOp : gx_op {
             else pass_through = true;
         }
 
+        if (n == 0) {
+            /* LEFT/RIGHT over an empty build still probes (emitting null
+             * rows) — give the probe kernel a valid all-empty table. */
+            n_buckets = 2;
+            mask = 1;
+            if (d_starts.grow((size_t)(n_buckets + 1) * 4, stream)) return -1;
+            HIP_OK(hipMemsetAsync(d_starts.p, 0, (size_t)(n_buckets + 1) * 4, stream));
+        }
         if (n > 0) {
             if (d_hashes.grow((size_t)n * 4, stream) ||
                 d_keynull.grow((size_t)n, stream)) return -1;
