@@ -88,6 +88,15 @@ class GxAggCfg(C.Structure):
     ]
 
 
+class GxJoinStats(C.Structure):
+    _fields_ = [
+        ("probe_kernel_ms", C.c_double),
+        ("probe_launches", C.c_int64),
+        ("probe_rows", C.c_int64),
+        ("matches", C.c_int64),
+    ]
+
+
 class GxPartCfg(C.Structure):
     _fields_ = [
         ("n_parts", C.c_int32),
@@ -140,7 +149,13 @@ class GxLib:
         L.gxop_part_create.argtypes = [C.POINTER(GxPartCfg)]
         L.gxop_part_consume.argtypes = [C.c_void_p, C.POINTER(GxChunk),
                                         C.POINTER(C.POINTER(GxResult))]
+        L.gxop_part_consume_concat.argtypes = [C.c_void_p, C.POINTER(GxChunk),
+                                               C.POINTER(C.POINTER(GxResult)),
+                                               C.POINTER(C.c_int64)]
         L.gxop_part_close.argtypes = [C.c_void_p]
+        L.gxop_result_copy_col.argtypes = [C.POINTER(GxResult), C.c_int32,
+                                           C.c_void_p, C.c_void_p]
+        L.gxop_join_get_stats.argtypes = [C.c_void_p, C.POINTER(GxJoinStats)]
         L.gxop_result_to_host.argtypes = [C.POINTER(GxResult)]
         L.gxop_result_release.argtypes = [C.POINTER(GxResult)]
         L.gx_last_error.restype = C.c_char_p

@@ -467,6 +467,11 @@ struct JoinOp : gx_op {
     bool pass_nothing = false, pass_through = false;
     bool tail_done = false;
 
+    /* live probe-kernel stats for bench.py's roofline leg */
+    hipEvent_t ev0 = nullptr, ev1 = nullptr;
+    double probe_kernel_ms = 0.0;
+    int64_t probe_launches = 0, probe_rows_total = 0, matches_total = 0;
+
     JoinOp(const gx_join_cfg *c) : gx_op(OP_JOIN, c->device, c->stream), cfg(*c) {
         keys.assign(c->keys, c->keys + c->n_keys);
         outer_types.assign(c->outer_types, c->outer_types + c->n_outer_cols);
@@ -484,6 +489,8 @@ struct JoinOp : gx_op {
         d_hashes.release(); d_keynull.release(); d_counts.release();
         d_starts.release(); d_entries.release(); d_bitmap.release();
         d_scan_tmp.release();
+        if (ev0) (void)hipEventDestroy(ev0);
+        if (ev1) (void)hipEventDestroy(ev1);
     }
 
     KeyViews key_views(const DevStore &s, const std::vector<int> &colidx) const {
@@ -723,7 +730,13 @@ struct JoinOp : gx_op {
                 P.counter = (uint32_t *)d_meta.p;
                 P.err = (uint32_t *)d_meta.p + 1;
                 P.build_matched = cfg.build_outer ? (uint32_t *)d_bitmap.p : nullptr;
+                if (!ev0) {
+                    HIP_OK(hipEventCreate(&ev0));
+                    HIP_OK(hipEventCreate(&ev1));
+                }
+                HIP_OK(hipEventRecord(ev0, stream));
                 hipLaunchKernelGGL(k_probe, dim3(gx_grid(n)), dim3(256), 0, stream, P);
+                HIP_OK(hipEventRecord(ev1, stream));
                 uint32_t meta[2];
                 HIP_OK(hipMemcpyAsync(meta, d_meta.p, 8, hipMemcpyDeviceToHost, stream));
                 HIP_OK(hipStreamSynchronize(stream));
@@ -732,7 +745,15 @@ struct JoinOp : gx_op {
                     probe_st.release();
                     return -2;
                 }
+                {
+                    float ms = 0;
+                    HIP_OK(hipEventElapsedTime(&ms, ev0, ev1));
+                    probe_kernel_ms += ms;
+                    probe_launches++;
+                }
                 if (meta[0] <= cap) {
+                    probe_rows_total += n;
+                    matches_total += meta[0];
                     HipResult *h = materialize(probe_st, (uint32_t *)d_pidx.p,
                                                (uint32_t *)d_bpos.p, meta[0]);
                     if (!h) { attempt = 99; break; }
@@ -864,6 +885,33 @@ int gxop_result_to_host(gx_result *res) {
 
 void gxop_result_release(gx_result *res) {
     if (res && res->opaque) free_result(static_cast<HipResult *>(res->opaque));
+}
+
+int gxop_result_copy_col(const gx_result *res, int32_t col, void *dst_values,
+                         void *dst_nulls) {
+    if (!res || col < 0 || col >= res->chunk.n_blocks) {
+        gx_set_err("bad result column");
+        return -1;
+    }
+    const gx_block *b = &res->chunk.blocks[col];
+    int64_t n = res->chunk.n_rows;
+    if (n == 0) return 0;
+    size_t es = (b->type == GX_I32) ? 4 : 8;
+    HIP_OK(hipMemcpy(dst_values, b->values, (size_t)n * es, hipMemcpyDefault));
+    if (dst_nulls && b->nulls)
+        HIP_OK(hipMemcpy(dst_nulls, (const void *)b->nulls, (size_t)n,
+                         hipMemcpyDefault));
+    return 0;
+}
+
+int gxop_join_get_stats(gx_op *op, gx_join_stats *out) {
+    if (!op || op->kind != OP_JOIN || !out) { gx_set_err("not a join op"); return -1; }
+    JoinOp *j = static_cast<JoinOp *>(op);
+    out->probe_kernel_ms = j->probe_kernel_ms;
+    out->probe_launches = j->probe_launches;
+    out->probe_rows = j->probe_rows_total;
+    out->matches = j->matches_total;
+    return 0;
 }
 const char *gx_last_error(void) { return gx_err.c_str(); }
 int gxop_abi_version(void) { return 950; }

@@ -75,6 +75,16 @@ class ParallelHashJoinExec:
             return None
         return self._lib.result_to_chunk(out)
 
+    def stats(self):
+        """Cumulative probe-kernel stats (bench roofline leg)."""
+        from .abi import GxJoinStats
+        s = GxJoinStats()
+        self._lib.check(self._lib.lib.gxop_join_get_stats(self._op, C.byref(s)),
+                        "join_get_stats")
+        return {"probe_kernel_ms": s.probe_kernel_ms,
+                "probe_launches": s.probe_launches,
+                "probe_rows": s.probe_rows, "matches": s.matches}
+
     def tail_chunks(self):
         chunks = []
         while True:

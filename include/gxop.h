@@ -200,13 +200,35 @@ gx_op *gxop_part_create(const gx_part_cfg *cfg);
  * partition id = ExecUtils.partition(rowHash) (ExecUtils.java:1023-1033).
  * outs must have room for cfg->n_parts pointers. */
 int gxop_part_consume(gx_op *op, const gx_chunk *chunk, gx_result **outs);
+/* same routing, but emit ONE result whose rows are grouped by partition id
+ * (partition p occupies rows [sum(counts[<p]), sum(counts[<=p])) ) and fill
+ * counts[n_parts] — the exact layout an RCCL all-to-allv send buffer needs
+ * (the cross-node PartitionedOutputCollector shape). */
+int gxop_part_consume_concat(gx_op *op, const gx_chunk *chunk,
+                             gx_result **out, int64_t *counts);
 int gxop_part_close(gx_op *op);
 
 /* ---- results / errors -------------------------------------------------- */
 
 int gxop_result_to_host(gx_result *res);
 void gxop_result_release(gx_result *res);
+/* copy one result column into caller-provided buffers (e.g. torch tensors
+ * for the RCCL exchange); dst_nulls may be NULL to skip. Device/host
+ * direction is auto-detected. */
+int gxop_result_copy_col(const gx_result *res, int32_t col,
+                         void *dst_values, void *dst_nulls);
 const char *gx_last_error(void);
+
+/* perf introspection: cumulative probe-kernel time (HIP events around the
+ * k_probe launches on the op's stream) + rows/matches, for bench.py's live
+ * roofline leg. CPU oracle returns zeros. */
+typedef struct gx_join_stats {
+    double probe_kernel_ms;
+    int64_t probe_launches;
+    int64_t probe_rows;
+    int64_t matches;
+} gx_join_stats;
+int gxop_join_get_stats(gx_op *op, gx_join_stats *out);
 
 /* Library/ABI version + device sanity. Returns gfx arch or 0 for CPU lib. */
 int gxop_abi_version(void);

@@ -866,6 +866,32 @@ struct PartOp : gx_op {
         input_types.assign(c->input_types, c->input_types + c->n_input_cols);
         pow2 = (cfg.n_parts & (-cfg.n_parts)) == cfg.n_parts;
     }
+    int consume_concat(const gx_chunk *ch, gx_result **out, int64_t *counts) {
+        *out = nullptr;
+        Store in;
+        in.init((int32_t)input_types.size(), input_types.data());
+        if (in.append(ch) != 0) return -1;
+        std::vector<int> kc(key_cols_.begin(), key_cols_.end());
+        std::vector<std::vector<size_t>> parts((size_t)cfg.n_parts);
+        for (size_t r = 0; r < in.n_rows; r++) {
+            int32_t h = in.row_hash(r, kc);
+            int32_t p;
+            if (pow2) p = hc_murmur3(h) & (cfg.n_parts - 1);
+            else p = (int32_t)(((uint32_t)hc_murmur3(h) & 0x7fffffffu) % (uint32_t)cfg.n_parts);
+            parts[(size_t)p].push_back(r);
+        }
+        std::vector<OutCol> cols(input_types.size());
+        for (size_t c = 0; c < input_types.size(); c++) cols[c].type = input_types[c];
+        for (int32_t p = 0; p < cfg.n_parts; p++) {
+            counts[p] = (int64_t)parts[(size_t)p].size();
+            for (size_t r : parts[(size_t)p])
+                for (size_t c = 0; c < input_types.size(); c++)
+                    cols[c].append_from(in.cols[c], r);
+        }
+        *out = make_result(std::move(cols), (int32_t)in.n_rows);
+        return 0;
+    }
+
     int consume(const gx_chunk *ch, gx_result **outs) {
         Store in;
         in.init((int32_t)input_types.size(), input_types.data());
@@ -948,7 +974,33 @@ int gxop_part_consume(gx_op *op, const gx_chunk *c, gx_result **outs) {
     if (!op || op->kind != OP_PART) { set_err("not a part op"); return -1; }
     return static_cast<PartOp *>(op)->consume(c, outs);
 }
+int gxop_part_consume_concat(gx_op *op, const gx_chunk *c, gx_result **out,
+                             int64_t *counts) {
+    if (!op || op->kind != OP_PART) { set_err("not a part op"); return -1; }
+    return static_cast<PartOp *>(op)->consume_concat(c, out, counts);
+}
 int gxop_part_close(gx_op *op) { delete op; return 0; }
+
+int gxop_result_copy_col(const gx_result *res, int32_t col, void *dst_values,
+                         void *dst_nulls) {
+    if (!res || col < 0 || col >= res->chunk.n_blocks) { set_err("bad column"); return -1; }
+    const gx_block *b = &res->chunk.blocks[col];
+    int64_t n = res->chunk.n_rows;
+    if (n == 0) return 0;
+    size_t es = (b->type == GX_I32) ? 4 : 8;
+    std::memcpy(dst_values, b->values, (size_t)n * es);
+    if (dst_nulls) {
+        if (b->nulls) std::memcpy(dst_nulls, b->nulls, (size_t)n);
+        else std::memset(dst_nulls, 0, (size_t)n);
+    }
+    return 0;
+}
+
+int gxop_join_get_stats(gx_op *op, gx_join_stats *out) {
+    if (!op || op->kind != OP_JOIN || !out) { set_err("not a join op"); return -1; }
+    std::memset(out, 0, sizeof(*out));
+    return 0;
+}
 
 int gxop_result_to_host(gx_result *) { return 0; /* oracle results are host */ }
 void gxop_result_release(gx_result *res) {

@@ -1,0 +1,44 @@
+"""Multi-process exchange path on CPU (gloo, world_size 2): the same
+shuffle_columns code the RCCL/xGMI path uses at N>1, with the oracle as the
+partition backend. Verifies completeness (no row lost/duplicated) and
+Java-exact routing (every received key belongs on this rank)."""
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                      "_exchange_worker.py")
+
+
+def test_shuffle_world2(tmp_path):
+    subprocess.run(["make", "-C", os.path.join(REPO, "oracle")], check=True,
+                   capture_output=True)
+    env = dict(os.environ)
+    env.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29612", WORLD_SIZE="2",
+               PYTHONPATH=REPO)
+    procs = []
+    for rank in range(2):
+        e = dict(env, RANK=str(rank))
+        procs.append(subprocess.Popen(
+            [sys.executable, WORKER, str(tmp_path)], env=e,
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=120)[0].decode() for p in procs]
+    for rank, p in enumerate(procs):
+        assert p.returncode == 0, f"rank {rank} failed:\n{outs[rank]}"
+    # workers wrote per-rank received rows; verify completeness here
+    recv = []
+    for rank in range(2):
+        recv.append(np.load(tmp_path / f"recv_{rank}.npz"))
+    all_keys = np.concatenate([r["keys"] for r in recv])
+    all_pay = np.concatenate([r["pay"] for r in recv])
+    sent_keys = np.concatenate([np.load(tmp_path / f"sent_{r}.npz")["keys"]
+                                for r in range(2)])
+    sent_pay = np.concatenate([np.load(tmp_path / f"sent_{r}.npz")["pay"]
+                               for r in range(2)])
+    assert sorted(zip(all_keys.tolist(), all_pay.tolist())) == \
+           sorted(zip(sent_keys.tolist(), sent_pay.tolist()))
