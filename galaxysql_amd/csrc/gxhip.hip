@@ -184,26 +184,59 @@ struct ProbeParams {
     uint32_t *build_matched;    /* bitmap (u32 words), build_outer only */
 };
 
-__device__ static inline void emit_pair(const ProbeParams &P, bool want,
-                                        uint32_t pidx, uint32_t bpos) {
-    unsigned long long m = __ballot(want);
-    if (!m) return;
-    int lane = threadIdx.x & 63;
-    int leader = __ffsll((unsigned long long)m) - 1;
-    uint32_t cnt = (uint32_t)__popcll(m);
+/* Per-wave LDS pair staging: batch STAGE pairs per global atomicAdd.
+ * A bare per-iteration atomic on one counter serializes the whole grid
+ * (~88 adds/us on one word — MI355X_MICROARCH.md row "dequeue"); staging
+ * cuts the atomic count by ~STAGE/2x. Wave-synchronous: no __syncthreads,
+ * counts are wave-uniform via ballot. */
+#define GX_EMIT_STAGE 512
+
+struct EmitStage {
+    uint32_t *s_p;   /* this wave's LDS slice */
+    uint32_t *s_b;
+    uint32_t cnt;    /* wave-uniform */
+    int lane;
+};
+
+__device__ static inline void emit_flush(const ProbeParams &P, EmitStage &E) {
+    if (E.cnt == 0) return;
     uint32_t base = 0;
-    if (lane == leader) base = atomicAdd(P.counter, cnt);
-    base = (uint32_t)__shfl((int)base, leader, 64);
-    if (want) {
-        uint32_t at = base + (uint32_t)__popcll(m & ((1ull << lane) - 1ull));
+    if (E.lane == 0) base = atomicAdd(P.counter, E.cnt);
+    base = (uint32_t)__shfl((int)base, 0, 64);
+    for (uint32_t i = (uint32_t)E.lane; i < E.cnt; i += 64) {
+        uint32_t at = base + i;
         if (at < P.cap) {
-            P.out_probe[at] = pidx;
-            P.out_build[at] = bpos;
+            P.out_probe[at] = E.s_p[i];
+            P.out_build[at] = E.s_b[i];
         }
     }
+    E.cnt = 0;
+}
+
+__device__ static inline void emit_pair(const ProbeParams &P, EmitStage &E,
+                                        bool want, uint32_t pidx, uint32_t bpos) {
+    unsigned long long m = __ballot(want);
+    if (!m) return;
+    uint32_t n = (uint32_t)__popcll(m);
+    if (E.cnt + n > GX_EMIT_STAGE) emit_flush(P, E);
+    if (want) {
+        uint32_t at = E.cnt + (uint32_t)__popcll(m & ((1ull << E.lane) - 1ull));
+        E.s_p[at] = pidx;
+        E.s_b[at] = bpos;
+    }
+    E.cnt += n;
 }
 
 __global__ void k_probe(ProbeParams P) {
+    __shared__ uint32_t s_pairs[2][4][GX_EMIT_STAGE]; /* [p/b][wave][slot] */
+    EmitStage E;
+    {
+        int wid = threadIdx.x >> 6;
+        E.s_p = s_pairs[0][wid];
+        E.s_b = s_pairs[1][wid];
+        E.cnt = 0;
+        E.lane = threadIdx.x & 63;
+    }
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t base_i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
          base_i += stride) {
@@ -247,12 +280,12 @@ __global__ void k_probe(ProbeParams P) {
                 matched = true;
             }
             bool emit_now = is_match && !P.semi_join;
-            emit_pair(P, emit_now, (uint32_t)i, bpos);
+            emit_pair(P, E, emit_now, (uint32_t)i, bpos);
         }
 
         /* post-row emissions (LEFT/RIGHT null rows, SEMI/ANTI rows) */
         bool want_null_row = active && P.outer_join && !P.build_outer && !matched;
-        emit_pair(P, want_null_row, (uint32_t)i, 0xFFFFFFFFu);
+        emit_pair(P, E, want_null_row, (uint32_t)i, 0xFFFFFFFFu);
         if (P.semi_join) {
             bool want;
             if (P.join_type == GX_JOIN_SEMI) want = active && matched;
@@ -261,9 +294,10 @@ __global__ void k_probe(ProbeParams P) {
                 if (want && P.anti_null_col >= 0)
                     want = !col_is_null(P.anti_col, i);
             }
-            emit_pair(P, want, (uint32_t)i, 0xFFFFFFFFu);
+            emit_pair(P, E, want, (uint32_t)i, 0xFFFFFFFFu);
         }
     }
+    emit_flush(P, E);
 }
 
 /* pass-through (ANTI over empty build): emit every probe row */
@@ -460,6 +494,7 @@ struct JoinOp : gx_op {
     DevBuf d_counts, d_starts, d_entries;
     DevBuf d_bitmap;         /* build_outer matched bitmap */
     DevBuf d_scan_tmp;
+    DevBuf d_pidx, d_bpos, d_meta, d_ph, d_pn; /* reused across probe calls */
     uint32_t mask = 0;
     int64_t n_buckets = 0;
     bool fast_i64 = false;
@@ -489,6 +524,8 @@ struct JoinOp : gx_op {
         d_hashes.release(); d_keynull.release(); d_counts.release();
         d_starts.release(); d_entries.release(); d_bitmap.release();
         d_scan_tmp.release();
+        d_pidx.release(); d_bpos.release(); d_meta.release();
+        d_ph.release(); d_pn.release();
         if (ev0) (void)hipEventDestroy(ev0);
         if (ev1) (void)hipEventDestroy(ev1);
     }
@@ -672,14 +709,13 @@ struct JoinOp : gx_op {
         }
 
         /* probe hashes */
-        DevBuf ph, pn;
-        if (ph.grow((size_t)n * 4, stream) || pn.grow((size_t)n, stream)) return -1;
+        if (d_ph.grow((size_t)n * 4, stream) || d_pn.grow((size_t)n, stream))
+            return -1;
         KeyViews pk = key_views_staged(probe_st, probe_key_cols);
         hipLaunchKernelGGL(k_hash_rows, dim3(gx_grid(n)), dim3(256), 0, stream,
-                           pk, n, (int32_t *)ph.p, (uint8_t *)pn.p);
+                           pk, n, (int32_t *)d_ph.p, (uint8_t *)d_pn.p);
 
         int rc = -1;
-        DevBuf d_pidx, d_bpos, d_meta;
         do {
             if (pass_through) {
                 /* ANTI over empty build: all probe rows pass */
@@ -707,8 +743,8 @@ struct JoinOp : gx_op {
                 P.entries = (const JoinEntry *)d_entries.p;
                 P.mask = mask;
                 P.n_probe = n;
-                P.hashes = (const int32_t *)ph.p;
-                P.keynull = (const uint8_t *)pn.p;
+                P.hashes = (const int32_t *)d_ph.p;
+                P.keynull = (const uint8_t *)d_pn.p;
                 P.fast_i64 = (int)fast_i64;
                 P.build_keys = key_views(build, build_key_cols);
                 P.probe_keys = pk;
