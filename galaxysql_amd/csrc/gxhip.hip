@@ -333,6 +333,49 @@ __global__ void k_unmatched(const uint32_t *bitmap, int64_t n_build,
 
 /* ===================== gather (output materialization) ================= */
 
+/* fused multi-column gather: read the pair indices ONCE per output row and
+ * materialize every output column (coalesced writes; the random reads are
+ * the algorithmic payload-gather cost). side: 0 = probe idx, 1 = build idx,
+ * 2 = null fill. */
+struct GatherParams {
+    const uint32_t *pidx;
+    const uint32_t *bidx;
+    int64_t n;
+    int32_t n_cols;
+    int32_t side[GX_MAX_COLS];
+    DevColView src[GX_MAX_COLS];
+    void *out_vals[GX_MAX_COLS];
+    uint8_t *out_nulls[GX_MAX_COLS];
+};
+
+__global__ void k_gather_multi(GatherParams G) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < G.n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t pi = G.pidx ? G.pidx[i] : 0xFFFFFFFFu;
+        uint32_t bi = G.bidx ? G.bidx[i] : 0xFFFFFFFFu;
+        for (int c = 0; c < G.n_cols; c++) {
+            uint32_t s = G.side[c] == 0 ? pi : (G.side[c] == 1 ? bi : 0xFFFFFFFFu);
+            const DevColView &src = G.src[c];
+            if (s == 0xFFFFFFFFu) {
+                G.out_nulls[c][i] = 1;
+                switch (src.type) {
+                case GX_I64: ((int64_t *)G.out_vals[c])[i] = 0; break;
+                case GX_I32: ((int32_t *)G.out_vals[c])[i] = 0; break;
+                case GX_F64: ((double *)G.out_vals[c])[i] = 0; break;
+                }
+                continue;
+            }
+            bool nn = col_is_null(src, s);
+            G.out_nulls[c][i] = nn;
+            switch (src.type) {
+            case GX_I64: ((int64_t *)G.out_vals[c])[i] = nn ? 0 : ((const int64_t *)src.values)[s]; break;
+            case GX_I32: ((int32_t *)G.out_vals[c])[i] = nn ? 0 : ((const int32_t *)src.values)[s]; break;
+            case GX_F64: ((double *)G.out_vals[c])[i] = nn ? 0 : ((const double *)src.values)[s]; break;
+            }
+        }
+    }
+}
+
 /* out[i] = src[idx[i]]; idx 0xFFFFFFFF -> NULL. Coalesced writes, the random
  * reads are the algorithmic cost of the join payload gather. */
 __global__ void k_gather(DevColView src, const uint32_t *idx, int64_t n,
@@ -654,40 +697,44 @@ struct JoinOp : gx_op {
     HipResult *materialize(const StagedChunk &probe, const uint32_t *d_pidx,
                            const uint32_t *d_bpos, int64_t n_out) {
         std::vector<int32_t> otypes = output_types();
+        if (otypes.size() > GX_MAX_COLS) { gx_set_err("too many output cols"); return nullptr; }
         HipResult *h = alloc_result_cols(otypes, n_out, device, stream);
         if (!h) return nullptr;
         if (n_out == 0) return h;
-        /* which source backs each output column + which index array */
-        size_t col = 0;
-        auto emit_col = [&](bool from_build, int src_col, const uint32_t *idx) -> int {
-            DevColView src = from_build ? build.view(src_col) : probe.views[src_col];
-            hipLaunchKernelGGL(k_gather, dim3(gx_grid(n_out)), dim3(256), 0, stream,
-                               src, idx, n_out, h->bufs[col * 2].p,
-                               (uint8_t *)h->bufs[col * 2 + 1].p, idx == nullptr);
+
+        GatherParams G;
+        std::memset(&G, 0, sizeof(G));
+        /* "probe"/"build" sides: d_pidx always indexes the probe chunk and
+         * d_bpos the build store; which output column comes from which side
+         * honors build_outer. side 2 (null fill) when that index ptr is null
+         * (pass-through). */
+        G.pidx = d_pidx;
+        G.bidx = d_bpos;
+        G.n = n_out;
+        int col = 0;
+        auto add_col = [&](bool from_build, int src_col) {
+            G.side[col] = from_build ? (d_bpos ? 1 : 2) : (d_pidx ? 0 : 2);
+            G.src[col] = from_build ? build.view(src_col) : probe.views[src_col];
+            G.out_vals[col] = h->bufs[col * 2].p;
+            G.out_nulls[col] = (uint8_t *)h->bufs[col * 2 + 1].p;
             col++;
-            return 0;
         };
         bool semi = cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI;
         size_t n_outer = outer_types.size();
         size_t n_inner_out = cfg.single_join ? 1 : inner_types.size();
-        /* map "outer"/"inner" to probe/build honoring build_outer */
-        const uint32_t *outer_idx = cfg.build_outer ? d_bpos : d_pidx;
-        const uint32_t *inner_idx = cfg.build_outer ? d_pidx : d_bpos;
         bool outer_from_build = cfg.build_outer;
         if (semi) {
-            for (size_t i = 0; i < n_outer; i++)
-                emit_col(outer_from_build, (int)i, outer_idx);
+            for (size_t i = 0; i < n_outer; i++) add_col(outer_from_build, (int)i);
         } else if (cfg.join_type != GX_JOIN_RIGHT) {
-            for (size_t i = 0; i < n_outer; i++)
-                emit_col(outer_from_build, (int)i, outer_idx);
-            for (size_t i = 0; i < n_inner_out; i++)
-                emit_col(!outer_from_build, (int)i, inner_idx);
+            for (size_t i = 0; i < n_outer; i++) add_col(outer_from_build, (int)i);
+            for (size_t i = 0; i < n_inner_out; i++) add_col(!outer_from_build, (int)i);
         } else {
-            for (size_t i = 0; i < inner_types.size(); i++)
-                emit_col(!outer_from_build, (int)i, inner_idx);
-            for (size_t i = 0; i < n_outer; i++)
-                emit_col(outer_from_build, (int)i, outer_idx);
+            for (size_t i = 0; i < inner_types.size(); i++) add_col(!outer_from_build, (int)i);
+            for (size_t i = 0; i < n_outer; i++) add_col(outer_from_build, (int)i);
         }
+        G.n_cols = col;
+        hipLaunchKernelGGL(k_gather_multi, dim3(gx_grid(n_out)), dim3(256), 0,
+                           stream, G);
         return h;
     }
 
