@@ -1,23 +1,29 @@
 #!/usr/bin/env python3
 """bench.py — headline benchmark for the MI355X operator pack.
 
-Workload (BASELINE.json configs[1], the largest single-GPU judged config):
-C2 = TPC-H SF10 join: orders 15M (build, o_orderkey = 4*i permuted) JOIN
-lineitem ~60M (probe, 1-7 lines per order) on orderkey, full output
-materialization. One "step" = one complete operator pass over the batch:
-hash-table build + probe + payload gather, through the gxop C-ABI
-(HIP/gfx950 kernels), inputs already resident in HBM.
+Metric (BASELINE.json): probe rows/s (+ agg groups/s) on TPC-H-shaped
+synthetic workloads, whole-job over all ranks, %HBM roofline for the
+dominant kernel.
 
-Metric: probe rows/s, whole-job over all ranks (BASELINE.json: "probe
-rows/s ... TPC-H ... 1/2/4/8 MI355X; %HBM roofline").
+Workloads:
+  c3 (default) — SF100 Q3: customer SEMI orders INNER lineitem + 3-key
+       GROUP BY SUM(revenue) as DOUBLE and as scaled-int64 cents
+       (DECIMAL(15,2)-sum semantics). The config the metric is quoted on
+       ("TPC-H SF100 Q3 ... 1xMI355X"); fits one GPU. Inputs pre-filtered /
+       revenue pre-projected (vectorized filter stage is SURVEY §8f row 1 —
+       DESIGN.md states the caveat).
+  c2 — SF10 orders⋈lineitem full materialization (configs[1]).
 
-N>1: weak scaling with a REAL exchange step per the reference's hash
-shuffle — each rank holds an SF10-sized shard, both sides are hash-
-repartitioned (Java-exact murmur routing) and exchanged via RCCL
-all-to-allv over xGMI, then joined locally.
+One "step" = one complete operator pass over the batch through the gxop
+C-ABI (HIP/gfx950), inputs already resident in HBM. N>1: weak scaling with
+the REAL hash-shuffle exchange (Java-exact murmur routing + RCCL
+all-to-allv over xGMI) per the reference's FIXED shuffle; for Q3 the group
+keys are orderkey-led, so rows group-colocate after the orderkey shuffle
+and the local aggregate is final (the reference optimizer's own colocation
+reasoning).
 
-CPU baseline (rank 0, N=1 only): the C++ oracle (oracle/, "port") timed
-probe-only on a bounded sample on this box's host cores.
+CPU baseline (rank 0, N=1 only): the C++ oracle ("port"), bounded sample,
+single thread.
 """
 import argparse
 import json
@@ -39,60 +45,111 @@ def next_pow2(x):
     return n
 
 
-def gen_c2_device(device, seed, sf=10):
-    """Synthetic TPC-H-shaped SF{sf} join inputs, generated directly in HBM."""
-    import torch
-    g = torch.Generator(device=device)
-    g.manual_seed(seed)
-    n_orders = sf * 1_500_000
-    okeys = 4 * torch.randperm(n_orders, generator=g, device=device,
-                               dtype=torch.int64)
-    opay = torch.randint(0, 1 << 30, (n_orders,), generator=g, device=device,
-                         dtype=torch.int64)
-    # lineitem: 1-7 lines per order (TPC-H shape, mean 4 -> ~sf*6M rows)
-    lines = torch.randint(1, 8, (n_orders,), generator=g, device=device,
-                          dtype=torch.int64)
-    lkeys = torch.repeat_interleave(okeys, lines)
-    n_li = lkeys.numel()
-    perm = torch.randperm(n_li, generator=g, device=device)
-    lkeys = lkeys[perm].contiguous()
-    lpay = torch.randint(0, 1 << 30, (n_li,), generator=g, device=device,
-                         dtype=torch.int64)
-    return (okeys, opay), (lkeys, lpay)
+def probe_bytes_per_row(n_build, matches, probe_rows):
+    """Algorithmic bytes per probe row (DESIGN.md 'Roofline accounting'):
+    4 hash + 1 nullflag + 8 starts + 8 key + 16*s entries + 8*m pair,
+    s = m + table load (own match + expected colliders)."""
+    n_buckets = next_pow2(max(2, n_build * 2))
+    m = matches / probe_rows if probe_rows else 0.0
+    s = m + n_build / n_buckets
+    return 4 + 1 + 8 + 8 + 16 * s + 8 * m
 
 
-def run_join_step(lib, build_cols, probe_cols, device):
-    """One full operator pass; returns (out_rows, stats dict)."""
-    import ctypes as C
-    from galaxysql_amd import abi
-    from galaxysql_amd.abi import GxResult
-    from galaxysql_amd.chunk import I64
-    from galaxysql_amd.exchange import chunk_from_torch
-    from galaxysql_amd.operators import ParallelHashJoinExec, EquiJoinKey
-
-    op = ParallelHashJoinExec(
-        lib, abi.INNER, [EquiJoinKey(0, 0, I64)], [I64, I64], [I64, I64],
-        device=device, expected_build_rows=build_cols[0].numel())
+def load_traffic(workload):
+    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "profiles", "pmc_traffic.json")
+    if not os.path.exists(path):
+        return None
     try:
-        ka = []
-        bc = chunk_from_torch(lib, list(build_cols), [I64, I64], ka)
-        lib.check(lib.lib.gxop_join_consume(op._op, C.byref(bc)), "consume")
-        op.build_consume()
-        pc = chunk_from_torch(lib, list(probe_cols), [I64, I64], ka)
-        out = C.POINTER(GxResult)()
-        lib.check(lib.lib.gxop_join_probe(op._op, C.byref(pc), C.byref(out)),
-                  "probe")
-        n_out = out.contents.chunk.n_rows if out else 0
-        if out:
-            lib.lib.gxop_result_release(out)
-        return n_out, op.stats()
-    finally:
-        op.close()
+        return json.load(open(path)).get(workload)
+    except Exception:
+        return None
 
 
-def cpu_baseline_leg(n_orders=15_000_000, n_probe_sample=8_000_000, seed=99):
-    """Oracle ('port') timed on this box's host cores: full build table,
-    probe-only timing on a bounded lineitem sample (~10-30 s of CPU work)."""
+# --------------------------------------------------------------------------
+# C2: SF10 orders ⋈ lineitem
+# --------------------------------------------------------------------------
+
+class C2:
+    name = "c2"
+
+    def __init__(self, args, device, rank):
+        import torch
+        self.sf = args.sf or 10
+        g = torch.Generator(device=device)
+        g.manual_seed(1234 + rank)
+        n_orders = self.sf * 1_500_000
+        okeys = 4 * torch.randperm(n_orders, generator=g, device=device,
+                                   dtype=torch.int64)
+        opay = torch.randint(0, 1 << 30, (n_orders,), generator=g,
+                             device=device, dtype=torch.int64)
+        lines = torch.randint(1, 8, (n_orders,), generator=g, device=device,
+                              dtype=torch.int64)
+        lkeys = torch.repeat_interleave(okeys, lines)
+        perm = torch.randperm(lkeys.numel(), generator=g, device=device)
+        lkeys = lkeys[perm].contiguous()
+        lpay = torch.randint(0, 1 << 30, (lkeys.numel(),), generator=g,
+                             device=device, dtype=torch.int64)
+        self.build = [okeys, opay]
+        self.probe = [lkeys, lpay]
+        self.n_probe = lkeys.numel()
+        self.n_build = okeys.numel()
+
+    def step(self, lib, local_rank, distributed):
+        import ctypes as C
+        from galaxysql_amd import abi
+        from galaxysql_amd.abi import GxResult
+        from galaxysql_amd.chunk import I64
+        from galaxysql_amd.exchange import chunk_from_torch, shuffle_columns
+        from galaxysql_amd.operators import ParallelHashJoinExec, EquiJoinKey
+
+        b, p = self.build, self.probe
+        if distributed:
+            b = shuffle_columns(lib, b, [I64, I64], [0], device=local_rank)
+            p = shuffle_columns(lib, p, [I64, I64], [0], device=local_rank)
+        op = ParallelHashJoinExec(
+            lib, abi.INNER, [EquiJoinKey(0, 0, I64)], [I64, I64], [I64, I64],
+            device=local_rank, expected_build_rows=b[0].numel())
+        try:
+            ka = []
+            bc = chunk_from_torch(lib, b, [I64, I64], ka)
+            lib.check(lib.lib.gxop_join_consume(op._op, C.byref(bc)), "consume")
+            op.build_consume()
+            pc = chunk_from_torch(lib, p, [I64, I64], ka)
+            out = C.POINTER(GxResult)()
+            lib.check(lib.lib.gxop_join_probe(op._op, C.byref(pc), C.byref(out)),
+                      "probe")
+            n_out = out.contents.chunk.n_rows if out else 0
+            if out:
+                lib.lib.gxop_result_release(out)
+            st = op.stats()
+            st["n_build"] = b[0].numel()
+            st["groups"] = 0
+            if not distributed:
+                assert n_out == self.n_probe, \
+                    f"C2 self-check: {n_out} != {self.n_probe}"
+            return st
+        finally:
+            op.close()
+
+    def config(self, world):
+        return {"workload": f"C2_tpch_sf{self.sf}_orders_join_lineitem",
+                "orders_rows_per_gpu": self.n_build,
+                "lineitem_rows_per_gpu": self.n_probe,
+                "output": "full 4-column materialization",
+                "exchange": "rccl_all_to_allv" if world > 1 else "none",
+                "parallelism": f"hash_shuffle_dp{world}"}
+
+    def cpu_baseline(self, sample_rows):
+        return c2_cpu_baseline(self.n_build, sample_rows)
+
+    traffic_key_fmt = "c2_sf%d"
+
+    def traffic_key(self):
+        return self.traffic_key_fmt % self.sf
+
+
+def c2_cpu_baseline(n_orders, n_probe_sample, seed=99):
     import ctypes as C
     from galaxysql_amd import abi
     from galaxysql_amd.abi import GxResult
@@ -103,13 +160,12 @@ def cpu_baseline_leg(n_orders=15_000_000, n_probe_sample=8_000_000, seed=99):
     rng = np.random.default_rng(seed)
     okeys = 4 * rng.permutation(n_orders).astype(np.int64)
     opay = rng.integers(0, 1 << 30, n_orders, dtype=np.int64)
-    pidx = rng.integers(0, n_orders, n_probe_sample)
-    lkeys = okeys[pidx]
+    lkeys = okeys[rng.integers(0, n_orders, n_probe_sample)]
     lpay = rng.integers(0, 1 << 30, n_probe_sample, dtype=np.int64)
 
-    op = ParallelHashJoinExec(lib, abi.INNER,
-                              [EquiJoinKey(0, 0, I64)], [I64, I64], [I64, I64],
-                              device=-1, expected_build_rows=n_orders)
+    op = ParallelHashJoinExec(lib, abi.INNER, [EquiJoinKey(0, 0, I64)],
+                              [I64, I64], [I64, I64], device=-1,
+                              expected_build_rows=n_orders)
     try:
         def mk_chunk(cols, ka):
             blocks = (abi.GxBlock * len(cols))()
@@ -120,7 +176,6 @@ def cpu_baseline_leg(n_orders=15_000_000, n_probe_sample=8_000_000, seed=99):
             ka.append(blocks)
             return abi.GxChunk(n_rows=len(cols[0]), n_blocks=len(cols),
                                blocks=blocks)
-
         ka = []
         bc = mk_chunk([okeys, opay], ka)
         lib.check(lib.lib.gxop_join_consume(op._op, C.byref(bc)), "consume")
@@ -128,36 +183,142 @@ def cpu_baseline_leg(n_orders=15_000_000, n_probe_sample=8_000_000, seed=99):
         pc = mk_chunk([lkeys, lpay], ka)
         t0 = time.perf_counter()
         out = C.POINTER(GxResult)()
-        lib.check(lib.lib.gxop_join_probe(op._op, C.byref(pc), C.byref(out)), "probe")
+        lib.check(lib.lib.gxop_join_probe(op._op, C.byref(pc), C.byref(out)),
+                  "probe")
         t1 = time.perf_counter()
         n_out = out.contents.chunk.n_rows if out else 0
         if out:
             lib.lib.gxop_result_release(out)
-        assert n_out == n_probe_sample, f"cpu baseline self-check: {n_out}"
-        return {
-            "value": n_probe_sample / (t1 - t0),
-            "unit": "rows/s",
-            "cores": 1,
-            "kind": "port",
-            "sample": f"build {n_orders} orders (untimed) + {n_probe_sample} "
-                      f"lineitem probe rows timed, single thread, -O3 -march=native",
-        }
+        assert n_out == n_probe_sample
+        return {"value": n_probe_sample / (t1 - t0), "unit": "rows/s",
+                "cores": 1, "kind": "port",
+                "sample": f"build {n_orders} orders (untimed) + "
+                          f"{n_probe_sample} probe rows timed, 1 thread, "
+                          "-O3 -march=native"}
     finally:
         op.close()
 
 
-def load_traffic(workload):
-    """Per-launch HBM traffic measured by a separate rocprofv3 --pmc pass
-    (profiles/); null when absent."""
-    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                        "profiles", "pmc_traffic.json")
-    if not os.path.exists(path):
-        return None
-    try:
-        d = json.load(open(path))
-        return d.get(workload)
-    except Exception:
-        return None
+# --------------------------------------------------------------------------
+# C3: SF100 Q3 chain
+# --------------------------------------------------------------------------
+
+class C3:
+    name = "c3"
+    # SF100 pre-filtered cardinalities:
+    # customer 15M * 1/5 = 3M build; orders 150M * 48.6%; lineitem 600M * 54%
+    CUST_TOTAL = 15_000_000
+    ORDERS_TOTAL = 150_000_000
+    CUST_SEL = 0.2
+    ORDERS_SEL = 0.486
+    LINEITEM = 324_000_000
+
+    def __init__(self, args, device, rank):
+        import torch
+        scale = args.c3_scale
+        g = torch.Generator(device=device)
+        g.manual_seed(4321 + rank)
+        ct = int(self.CUST_TOTAL * scale)
+        ot = int(self.ORDERS_TOTAL * scale)
+        li = int(self.LINEITEM * scale)
+        n_cust = int(ct * self.CUST_SEL)
+        n_orders = int(ot * self.ORDERS_SEL)
+        self.cust = [torch.randperm(ct, generator=g, device=device,
+                                    dtype=torch.int64)[:n_cust].contiguous()]
+        okeys_all = 4 * torch.randperm(ot, generator=g, device=device,
+                                       dtype=torch.int64)
+        self.orders = [
+            torch.randint(0, ct, (n_orders,), generator=g, device=device,
+                          dtype=torch.int64),
+            okeys_all[:n_orders].contiguous(),
+            torch.randint(8000, 9500, (n_orders,), generator=g, device=device,
+                          dtype=torch.int32),
+            torch.zeros(n_orders, dtype=torch.int32, device=device),
+        ]
+        lkeys = okeys_all[torch.randint(0, ot, (li,), generator=g,
+                                        device=device)]
+        cents = torch.randint(100, 10_000_000, (li,), generator=g,
+                              device=device, dtype=torch.int64)
+        self.lineitem = [lkeys.contiguous(), cents.double() / 100.0, cents]
+        self.n_probe = li
+        self.n_build = n_orders  # join2 build upper bound (pre-SEMI)
+        self.expected_groups = int(n_orders * self.CUST_SEL) + 1024
+        self.last_info = None
+
+    def step(self, lib, local_rank, distributed):
+        from galaxysql_amd.chunk import I64, I32, F64
+        from galaxysql_amd.exchange import shuffle_columns
+        from galaxysql_amd.queries import run_q3, CUST_TYPES, ORDERS_TYPES, \
+            LINEITEM_TYPES
+
+        cust, orders, lineitem = self.cust, self.orders, self.lineitem
+        if distributed:
+            cust = shuffle_columns(lib, cust, CUST_TYPES, [0], device=local_rank)
+            orders = shuffle_columns(lib, orders, ORDERS_TYPES, [0],
+                                     device=local_rank)
+            # join2 side shuffles by orderkey: orders result rows move with
+            # their orderkey hash; Q3's group key is orderkey-led, so the
+            # local aggregate after this shuffle is final.
+            lineitem = shuffle_columns(lib, lineitem, LINEITEM_TYPES, [0],
+                                       device=local_rank)
+            # join1 runs on custkey-sharded data; run_q3 re-shuffles its
+            # RESULT by o_orderkey (col 1) before join2.
+        _, info = run_q3(lib, local_rank, cust, orders, lineitem,
+                         expected_groups=self.expected_groups, to_host=False,
+                         reshuffle_by_orderkey=distributed,
+                         local_rank=local_rank)
+        self.last_info = info
+        st = dict(info["join2_stats"])
+        st["n_build"] = info["orders_kept"]
+        st["groups"] = info["groups"]
+        return st
+
+    def config(self, world):
+        return {"workload": "C3_tpch_sf100_q3_join_chain_groupby",
+                "customer_build_rows_per_gpu": self.cust[0].numel(),
+                "orders_rows_per_gpu": self.orders[0].numel(),
+                "lineitem_rows_per_gpu": self.n_probe,
+                "groups_last_step": self.last_info["groups"] if self.last_info else None,
+                "aggregates": "SUM(revenue) f64 + SUM(cents) i64 + COUNT(*)",
+                "inputs": "pre-filtered, revenue pre-projected (DESIGN.md)",
+                "exchange": "rccl_all_to_allv" if world > 1 else "none",
+                "parallelism": f"hash_shuffle_dp{world}"}
+
+    def cpu_baseline(self, sample_rows):
+        return c3_cpu_baseline(sample_scale=1.0 / 64)
+
+    def traffic_key(self):
+        return "c3_sf100"
+
+
+def c3_cpu_baseline(sample_scale, seed=101):
+    """Oracle Q3 chain on a bounded sample (~1/64 of SF100), whole chain
+    timed (both builds + probes + aggregate), single thread."""
+    import torch
+    from galaxysql_amd import abi
+    from galaxysql_amd.queries import run_q3, gen_q3_numpy
+
+    lib = abi.load_oracle()
+    rng = np.random.default_rng(seed)
+    data = gen_q3_numpy(rng,
+                        n_cust_total=int(C3.CUST_TOTAL * sample_scale),
+                        n_orders_total=int(C3.ORDERS_TOTAL * sample_scale),
+                        n_lineitem=int(C3.LINEITEM * sample_scale))
+    t = [[torch.from_numpy(a) for a in cols] for cols in data]
+    t0 = time.perf_counter()
+    _, info = run_q3(lib, -1, t[0], t[1], t[2], to_host=False)
+    t1 = time.perf_counter()
+    n_li = data[2][0].shape[0]
+    return {"value": n_li / (t1 - t0), "unit": "rows/s", "cores": 1,
+            "kind": "port",
+            "sample": f"full Q3 chain on a 1/{int(1/sample_scale)} SF100 "
+                      f"sample ({n_li} lineitem rows), builds+probes+agg all "
+                      "timed, 1 thread, -O3 -march=native"}
+
+
+# --------------------------------------------------------------------------
+
+WORKLOADS = {"c2": C2, "c3": C3}
 
 
 def main():
@@ -165,7 +326,10 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
-    ap.add_argument("--sf", type=int, default=10)
+    ap.add_argument("--workload", choices=list(WORKLOADS), default="c3")
+    ap.add_argument("--sf", type=int, default=None, help="C2 scale factor")
+    ap.add_argument("--c3-scale", type=float, default=1.0,
+                    help="C3 size fraction of SF100 (1.0 = full)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--cpu-probe-rows", type=int, default=8_000_000)
     args = ap.parse_args()
@@ -187,13 +351,9 @@ def main():
     torch.cuda.set_device(local_rank)
 
     from galaxysql_amd import abi
-    from galaxysql_amd.chunk import I64
-    from galaxysql_amd.exchange import shuffle_columns
     lib = abi.load_hip()  # the HIP extension — no CPU fallback exists
 
-    (okeys, opay), (lkeys, lpay) = gen_c2_device(device, seed=1234 + rank,
-                                                 sf=args.sf)
-    n_build_local, n_probe_local = okeys.numel(), lkeys.numel()
+    wl = WORKLOADS[args.workload](args, device, rank)
 
     def barrier_sync():
         if distributed:
@@ -201,36 +361,18 @@ def main():
             dist.barrier()
         torch.cuda.synchronize()
 
-    stats_acc = {"probe_kernel_ms": 0.0, "probe_launches": 0,
-                 "probe_rows": 0, "matches": 0}
-    out_rows_last = 0
-
-    def step(accumulate):
-        nonlocal out_rows_last
-        if distributed:
-            b = shuffle_columns(lib, [okeys, opay], [I64, I64], [0],
-                                device=local_rank)
-            p = shuffle_columns(lib, [lkeys, lpay], [I64, I64], [0],
-                                device=local_rank)
-        else:
-            b, p = [okeys, opay], [lkeys, lpay]
-        n_out, st = run_join_step(lib, b, p, local_rank)
-        out_rows_last = n_out
-        if accumulate:
-            for k in stats_acc:
-                stats_acc[k] += st[k]
+    acc = {"probe_kernel_ms": 0.0, "probe_launches": 0, "probe_rows": 0,
+           "matches": 0, "n_build": 0, "groups": 0}
 
     for _ in range(args.warmup):
-        step(False)
-    # self-check: every lineitem key exists in orders exactly once
-    if not distributed:
-        assert out_rows_last == n_probe_local, \
-            f"join self-check failed: {out_rows_last} != {n_probe_local}"
+        wl.step(lib, local_rank, distributed)
 
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step(True)
+        st = wl.step(lib, local_rank, distributed)
+        for k in acc:
+            acc[k] += st.get(k, 0)
     barrier_sync()
     t1 = time.perf_counter()
 
@@ -240,37 +382,24 @@ def main():
         t = torch.tensor([elapsed], device=device)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
-        tot = torch.tensor([float(n_probe_local)], device=device)
+        tot = torch.tensor([float(wl.n_probe)], device=device)
         dist.all_reduce(tot)
         total_probe_rows_per_step = float(tot.item())
     else:
-        total_probe_rows_per_step = float(n_probe_local)
+        total_probe_rows_per_step = float(wl.n_probe)
 
     if rank != 0:
         return
 
     value = total_probe_rows_per_step * args.steps / elapsed
-
-    # roofline for the dominant kernel (k_probe), HIP-event timed live.
-    # Algorithmic bytes per probe row (DESIGN.md "Roofline accounting"):
-    #   4 (hash) + 1 (nullflag) + 8 (bucket starts) + 8 (probe key)
-    #   + 16*s (CSR entries scanned) + 8*m (emitted pair), with
-    #   s = m + n_build/n_buckets (own match + expected colliders).
-    n_build = n_build_local  # per-rank build size (weak scaling)
-    n_buckets = next_pow2(max(2, n_build * 2))
-    m_bar = (stats_acc["matches"] / stats_acc["probe_rows"]
-             if stats_acc["probe_rows"] else 0.0)
-    s_bar = m_bar + n_build / n_buckets
-    bytes_per_row = 4 + 1 + 8 + 8 + 16 * s_bar + 8 * m_bar
-    probe_ms = stats_acc["probe_kernel_ms"]
-    achieved = (bytes_per_row * stats_acc["probe_rows"] / (probe_ms / 1e3)
-                if probe_ms > 0 else 0.0)
-    traffic = load_traffic("c2_sf%d" % args.sf)
+    n_build_avg = acc["n_build"] / args.steps
+    bpr = probe_bytes_per_row(int(n_build_avg), acc["matches"], acc["probe_rows"])
+    probe_ms = acc["probe_kernel_ms"]
+    achieved = (bpr * acc["probe_rows"] / (probe_ms / 1e3)) if probe_ms else 0.0
 
     cpu_baseline = None
     if not args.no_cpu_baseline and world == 1:
-        cpu_baseline = cpu_baseline_leg(n_orders=n_build_local,
-                                        n_probe_sample=args.cpu_probe_rows)
+        cpu_baseline = wl.cpu_baseline(args.cpu_probe_rows)
 
     result = {
         "metric": "probe_rows_per_s",
@@ -285,21 +414,16 @@ def main():
         "vs_baseline": None,
         "dtype": "int64",
         "data": "synthetic",
-        "config": {
-            "workload": f"C2_tpch_sf{args.sf}_orders_join_lineitem",
-            "orders_rows_per_gpu": n_build_local,
-            "lineitem_rows_per_gpu": n_probe_local,
-            "output": "full 4-column materialization",
-            "exchange": "rccl_all_to_allv" if distributed else "none",
-            "parallelism": f"hash_shuffle_dp{world}",
-        },
+        "config": dict(wl.config(world),
+                       agg_groups_per_s=(acc["groups"] / elapsed)
+                       if acc["groups"] else None),
         "roofline": {
             "bound": "hbm",
             "achieved": achieved / 1e9,
             "peak": HBM_PEAK_BYTES_PER_S / 1e9,
             "unit": "GB/s",
             "frac": achieved / HBM_PEAK_BYTES_PER_S,
-            "traffic": traffic,
+            "traffic": load_traffic(wl.traffic_key()),
         },
         "cpu_baseline": cpu_baseline,
     }

@@ -1,0 +1,195 @@
+"""Operator-chain pipelines for the judged TPC-H-shaped configs
+(BASELINE.md C1-C5), wired through the gxop C-ABI exactly as the CN's
+physical plan would chain the operators: results stay
+device-resident between operators (a gx_result's chunk is consumed
+directly by the next op — no host round trip).
+
+Q3: customer SEMI-filters orders (customer contributes
+no output columns in Q3), the filtered orders build the second join,
+lineitem probes it, and a 3-key hash aggregate sums revenue per
+(l_orderkey, o_orderdate, o_shippriority).
+
+Inputs arrive PRE-FILTERED and revenue PRE-PROJECTED (extendedprice *
+(1-discount)) — the vectorized filter/project stage is §8(f) row 1, not yet
+built; DESIGN.md states this caveat. Revenue is carried twice: DOUBLE
+(rel-tol 1e-9) and scaled-int64 cents (DECIMAL(15,2)-sum semantics,
+bit-exact).
+"""
+from __future__ import annotations
+
+import ctypes as C
+
+from . import abi
+from .abi import GxResult
+from .chunk import I64, I32, F64
+from .exchange import chunk_from_torch
+from .operators import ParallelHashJoinExec, HashAggExec, EquiJoinKey
+
+CUST_TYPES = [I64]                       # c_custkey
+ORDERS_TYPES = [I64, I64, I32, I32]      # o_custkey, o_orderkey, o_orderdate, o_shippriority
+LINEITEM_TYPES = [I64, F64, I64]         # l_orderkey, revenue, revenue_cents
+
+
+def _consume_tensors(lib, op, tensors, types):
+    ka = []
+    gc = chunk_from_torch(lib, tensors, types, ka)
+    lib.check(lib.lib.gxop_join_consume(op._op, C.byref(gc)), "join_consume")
+
+
+def _probe_tensors(lib, op, tensors, types):
+    ka = []
+    gc = chunk_from_torch(lib, tensors, types, ka)
+    out = C.POINTER(GxResult)()
+    lib.check(lib.lib.gxop_join_probe(op._op, C.byref(gc), C.byref(out)),
+              "join_probe")
+    return out
+
+
+def _probe_result(lib, op, res):
+    out = C.POINTER(GxResult)()
+    lib.check(lib.lib.gxop_join_probe(op._op, C.byref(res.contents.chunk),
+                                      C.byref(out)), "join_probe")
+    return out
+
+
+_TORCH_DT = None
+
+
+def _torch_dtypes():
+    global _TORCH_DT
+    if _TORCH_DT is None:
+        import torch
+        _TORCH_DT = {I64: torch.int64, I32: torch.int32, F64: torch.float64}
+    return _TORCH_DT
+
+
+def result_to_tensors(lib, res, types, device_t):
+    """Copy a gx_result's columns into fresh torch tensors (null-free
+    columns only — the exchange path's constraint this round)."""
+    import torch
+    n = res.contents.chunk.n_rows
+    dt = _torch_dtypes()
+    cols = []
+    for ci, ty in enumerate(types):
+        t = torch.empty(n, dtype=dt[ty], device=device_t)
+        if n:
+            lib.check(lib.lib.gxop_result_copy_col(
+                res, ci, C.c_void_p(t.data_ptr()), None), "result_copy_col")
+        cols.append(t)
+    return cols
+
+
+def run_q3(lib, device, cust, orders, lineitem, expected_groups=0,
+           to_host=True, reshuffle_by_orderkey=False, local_rank=0):
+    """cust/orders/lineitem: lists of torch tensors (cpu for the oracle,
+    cuda for the HIP path) with the column types above.
+
+    Returns (result_rows_chunk, info) where info carries join2 kernel stats
+    + intermediate cardinalities. The caller owns nothing to free."""
+    # join1: SEMI — build customer, probe orders on o_custkey
+    j1 = ParallelHashJoinExec(
+        lib, abi.SEMI, [EquiJoinKey(0, 0, I64)],
+        outer_types=ORDERS_TYPES, inner_types=CUST_TYPES,
+        device=device, expected_build_rows=cust[0].numel())
+    try:
+        _consume_tensors(lib, j1, list(cust), CUST_TYPES)
+        j1.build_consume()
+        r1 = _probe_tensors(lib, j1, list(orders), ORDERS_TYPES)
+        n_orders_kept = r1.contents.chunk.n_rows if r1 else 0
+
+        # join2: INNER — build the surviving orders (key col 1 = o_orderkey),
+        # probe lineitem on l_orderkey
+        j2 = ParallelHashJoinExec(
+            lib, abi.INNER, [EquiJoinKey(0, 1, I64)],
+            outer_types=LINEITEM_TYPES, inner_types=ORDERS_TYPES,
+            device=device, expected_build_rows=n_orders_kept)
+        try:
+            if reshuffle_by_orderkey:
+                # N>1: join1 ran custkey-sharded; its result re-shards by
+                # o_orderkey (col 1) so join2 is orderkey-colocated with the
+                # lineitem shuffle (the reference's FIXED shuffle between
+                # plan fragments).
+                from .exchange import shuffle_columns
+                cols = result_to_tensors(lib, r1, ORDERS_TYPES,
+                                         cust[0].device) if r1 else None
+                if r1:
+                    lib.lib.gxop_result_release(r1)
+                    r1 = None
+                cols = shuffle_columns(lib, cols, ORDERS_TYPES, [1],
+                                       device=device)
+                n_orders_kept = cols[0].numel()
+                _consume_tensors(lib, j2, cols, ORDERS_TYPES)
+            elif r1:
+                lib.check(lib.lib.gxop_join_consume(
+                    j2._op, C.byref(r1.contents.chunk)), "join_consume")
+            j2.build_consume()
+            if r1:
+                lib.lib.gxop_result_release(r1)
+                r1 = None
+            r2 = _probe_tensors(lib, j2, list(lineitem), LINEITEM_TYPES)
+            n_joined = r2.contents.chunk.n_rows if r2 else 0
+            j2_stats = j2.stats()
+
+            # agg: GROUP BY (l_orderkey, o_orderdate, o_shippriority)
+            #      SUM(revenue) f64, SUM(revenue_cents) i64, COUNT(*)
+            # join2 output columns: [l_orderkey, revenue, revenue_cents,
+            #                        o_custkey, o_orderkey, o_orderdate, o_prio]
+            agg = HashAggExec(
+                lib, group_cols=[0, 5, 6],
+                aggs=[(abi.SUM_F64, 1), (abi.SUM_I64, 2), (abi.COUNT_ROW, -1)],
+                input_types=LINEITEM_TYPES + ORDERS_TYPES,
+                expected_groups=expected_groups or max(1024, n_orders_kept),
+                device=device)
+            try:
+                if r2:
+                    lib.check(lib.lib.gxop_agg_consume(
+                        agg._op, C.byref(r2.contents.chunk)), "agg_consume")
+                agg.build_consume()
+                if r2:
+                    lib.lib.gxop_result_release(r2)
+                    r2 = None
+                if to_host:
+                    chunks = agg.result_chunks()
+                    n_groups = sum(c.n_rows for c in chunks)
+                else:
+                    # device-resident result (the real plan feeds TopN next;
+                    # final client rows are ~10 after LIMIT) — emit kernels
+                    # still run, the host copy does not.
+                    chunks = []
+                    n_groups = 0
+                    while True:
+                        out = C.POINTER(GxResult)()
+                        lib.check(lib.lib.gxop_agg_next(agg._op, C.byref(out)),
+                                  "agg_next")
+                        if not out:
+                            break
+                        n_groups += out.contents.chunk.n_rows
+                        lib.lib.gxop_result_release(out)
+                info = {"orders_kept": n_orders_kept, "joined_rows": n_joined,
+                        "groups": n_groups, "join2_stats": j2_stats}
+                return chunks, info
+            finally:
+                agg.close()
+        finally:
+            j2.close()
+    finally:
+        j1.close()
+
+
+def gen_q3_numpy(rng, n_cust_total, n_orders_total, n_lineitem, cust_sel=0.2,
+                 orders_sel=0.486):
+    """Small-scale synthetic Q3-shaped inputs for parity tests (numpy)."""
+    import numpy as np
+    n_cust = int(n_cust_total * cust_sel)
+    cust_keys = rng.choice(n_cust_total, size=n_cust, replace=False).astype(np.int64)
+    n_orders = int(n_orders_total * orders_sel)
+    okeys_all = 4 * rng.permutation(n_orders_total).astype(np.int64)
+    o_orderkey = okeys_all[:n_orders]
+    o_custkey = rng.integers(0, n_cust_total, n_orders).astype(np.int64)
+    o_date = rng.integers(8000, 9500, n_orders).astype(np.int32)
+    o_prio = np.zeros(n_orders, dtype=np.int32)
+    l_orderkey = okeys_all[rng.integers(0, n_orders_total, n_lineitem)]
+    cents = rng.integers(100, 10_000_000, n_lineitem).astype(np.int64)
+    revenue = cents.astype(np.float64) / 100.0
+    return ([cust_keys], [o_custkey, o_orderkey, o_date, o_prio],
+            [l_orderkey, revenue, cents])
