@@ -97,6 +97,15 @@ class GxJoinStats(C.Structure):
     ]
 
 
+class GxAggStats(C.Structure):
+    _fields_ = [
+        ("kernel_ms", C.c_double),
+        ("consumes", C.c_int64),
+        ("rows", C.c_int64),
+        ("groups", C.c_int64),
+    ]
+
+
 class GxPartCfg(C.Structure):
     _fields_ = [
         ("n_parts", C.c_int32),
@@ -156,6 +165,7 @@ class GxLib:
         L.gxop_result_copy_col.argtypes = [C.POINTER(GxResult), C.c_int32,
                                            C.c_void_p, C.c_void_p]
         L.gxop_join_get_stats.argtypes = [C.c_void_p, C.POINTER(GxJoinStats)]
+        L.gxop_agg_get_stats.argtypes = [C.c_void_p, C.POINTER(GxAggStats)]
         L.gxop_result_to_host.argtypes = [C.POINTER(GxResult)]
         L.gxop_result_release.argtypes = [C.POINTER(GxResult)]
         L.gx_last_error.restype = C.c_char_p

@@ -135,6 +135,14 @@ class HashAggExec:
         self._lib.check(self._lib.lib.gxop_agg_consume(self._op, C.byref(gch)),
                         "agg_consume")
 
+    def stats(self):
+        from .abi import GxAggStats
+        s = GxAggStats()
+        self._lib.check(self._lib.lib.gxop_agg_get_stats(self._op, C.byref(s)),
+                        "agg_get_stats")
+        return {"kernel_ms": s.kernel_ms, "consumes": s.consumes,
+                "rows": s.rows, "groups": s.groups}
+
     def build_consume(self):
         self._lib.check(self._lib.lib.gxop_agg_build(self._op), "agg_build")
 

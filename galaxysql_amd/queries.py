@@ -176,6 +176,114 @@ def run_q3(lib, device, cust, orders, lineitem, expected_groups=0,
         j1.close()
 
 
+Q18_LINEITEM_TYPES = [I64, I64]          # l_orderkey, l_quantity
+Q18_ORDERS_TYPES = [I64, I64, I64]       # o_orderkey, o_custkey, o_payload
+Q18_CUST_TYPES = [I64, I64]              # c_custkey, c_payload
+
+
+def run_q18(lib, device, cust, orders, lineitem, having=300,
+            expected_groups=0):
+    """Q18: lineitem GROUP BY l_orderkey SUM(l_quantity)
+    (the dominant operator: ~150M groups at SF100), HAVING sum > 300
+    (~hundreds of survivors), then the tiny survivors join orders and
+    customer. The HAVING filter runs in torch on the device result — the
+    reference's FilterExec is host glue around the operators (SURVEY.md §2
+    out-of-scope rows), the operator work is the aggregate + joins.
+
+    Returns (n_final_rows, info)."""
+    import torch
+    agg = HashAggExec(lib, group_cols=[0], aggs=[(abi.SUM_I64, 1)],
+                      input_types=Q18_LINEITEM_TYPES,
+                      expected_groups=expected_groups or lineitem[0].numel() // 4,
+                      device=device)
+    try:
+        ka = []
+        gc = chunk_from_torch(lib, list(lineitem), Q18_LINEITEM_TYPES, ka)
+        lib.check(lib.lib.gxop_agg_consume(agg._op, C.byref(gc)), "agg_consume")
+        agg.build_consume()
+        agg_stats = agg.stats()
+        parts = []
+        n_groups = 0
+        while True:
+            out = C.POINTER(GxResult)()
+            lib.check(lib.lib.gxop_agg_next(agg._op, C.byref(out)), "agg_next")
+            if not out:
+                break
+            n_groups += out.contents.chunk.n_rows
+            parts.append(result_to_tensors(lib, out, [I64, I64],
+                                           lineitem[0].device))
+            lib.lib.gxop_result_release(out)
+    finally:
+        agg.close()
+
+    keys = torch.cat([p[0] for p in parts]) if parts else \
+        torch.empty(0, dtype=torch.int64, device=lineitem[0].device)
+    sums = torch.cat([p[1] for p in parts]) if parts else \
+        torch.empty(0, dtype=torch.int64, device=lineitem[0].device)
+    mask = sums > having
+    skeys = keys[mask].contiguous()
+    ssums = sums[mask].contiguous()
+    n_surv = skeys.numel()
+
+    # survivors ⋈ orders on orderkey (build = tiny survivors, probe = orders)
+    ja = ParallelHashJoinExec(lib, abi.INNER, [EquiJoinKey(0, 0, I64)],
+                              outer_types=Q18_ORDERS_TYPES,
+                              inner_types=[I64, I64], device=device,
+                              expected_build_rows=max(1, n_surv))
+    try:
+        _consume_tensors(lib, ja, [skeys, ssums], [I64, I64])
+        ja.build_consume()
+        r1 = _probe_tensors(lib, ja, list(orders), Q18_ORDERS_TYPES)
+        n_r1 = r1.contents.chunk.n_rows if r1 else 0
+        # output: [o_orderkey, o_custkey, o_payload, s_key, s_sum]
+        cols1 = result_to_tensors(lib, r1, Q18_ORDERS_TYPES + [I64, I64],
+                                  lineitem[0].device) if r1 else None
+        if r1:
+            lib.lib.gxop_result_release(r1)
+    finally:
+        ja.close()
+
+    # ⋈ customer on custkey (build = the tiny result, probe = customer)
+    jb = ParallelHashJoinExec(lib, abi.INNER, [EquiJoinKey(0, 1, I64)],
+                              outer_types=Q18_CUST_TYPES,
+                              inner_types=Q18_ORDERS_TYPES + [I64, I64],
+                              device=device, expected_build_rows=max(1, n_r1))
+    try:
+        if cols1 is not None and n_r1 > 0:
+            _consume_tensors(lib, jb, cols1, Q18_ORDERS_TYPES + [I64, I64])
+        jb.build_consume()
+        r2 = _probe_tensors(lib, jb, list(cust), Q18_CUST_TYPES)
+        n_final = r2.contents.chunk.n_rows if r2 else 0
+        if r2:
+            lib.lib.gxop_result_release(r2)
+    finally:
+        jb.close()
+
+    info = {"groups": n_groups, "survivors": n_surv, "after_orders": n_r1,
+            "final_rows": n_final, "agg_stats": agg_stats}
+    return n_final, info
+
+
+def gen_q18_numpy(rng, n_cust, n_orders, having_frac=0.0001):
+    """Q18-shaped inputs (numpy, for parity tests): 1-7 lines per order,
+    quantities 1..50, a few orders inflated past the HAVING threshold."""
+    import numpy as np
+    okeys = 4 * rng.permutation(n_orders).astype(np.int64)
+    o_custkey = rng.integers(0, n_cust, n_orders).astype(np.int64)
+    o_pay = rng.integers(0, 1 << 30, n_orders, dtype=np.int64)
+    lines = rng.integers(1, 8, n_orders)
+    l_orderkey = np.repeat(okeys, lines)
+    lqty = rng.integers(1, 51, l_orderkey.shape[0]).astype(np.int64)
+    # inflate a fraction of orders so HAVING>300 has survivors
+    hot = rng.random(l_orderkey.shape[0]) < having_frac
+    lqty[hot] += 400
+    perm = rng.permutation(l_orderkey.shape[0])
+    l_orderkey, lqty = l_orderkey[perm], lqty[perm]
+    c_custkey = np.arange(n_cust, dtype=np.int64)
+    c_pay = rng.integers(0, 1 << 30, n_cust, dtype=np.int64)
+    return ([c_custkey, c_pay], [okeys, o_custkey, o_pay], [l_orderkey, lqty])
+
+
 def gen_q3_numpy(rng, n_cust_total, n_orders_total, n_lineitem, cust_sel=0.2,
                  orders_sel=0.486):
     """Small-scale synthetic Q3-shaped inputs for parity tests (numpy)."""

@@ -230,6 +230,14 @@ typedef struct gx_join_stats {
 } gx_join_stats;
 int gxop_join_get_stats(gx_op *op, gx_join_stats *out);
 
+typedef struct gx_agg_stats {
+    double kernel_ms;     /* insert+gid+accumulate kernel time (HIP events) */
+    int64_t consumes;
+    int64_t rows;
+    int64_t groups;
+} gx_agg_stats;
+int gxop_agg_get_stats(gx_op *op, gx_agg_stats *out);
+
 /* Library/ABI version + device sanity. Returns gfx arch or 0 for CPU lib. */
 int gxop_abi_version(void);
 

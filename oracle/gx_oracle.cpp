@@ -1002,6 +1002,13 @@ int gxop_join_get_stats(gx_op *op, gx_join_stats *out) {
     return 0;
 }
 
+int gxop_agg_get_stats(gx_op *op, gx_agg_stats *out) {
+    if (!op || op->kind != OP_AGG || !out) { set_err("not an agg op"); return -1; }
+    std::memset(out, 0, sizeof(*out));
+    out->groups = static_cast<AggOp *>(op)->n_groups();
+    return 0;
+}
+
 int gxop_result_to_host(gx_result *) { return 0; /* oracle results are host */ }
 void gxop_result_release(gx_result *res) {
     if (res) delete static_cast<ResultHolder *>(res->opaque);
