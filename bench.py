@@ -317,8 +317,118 @@ def c3_cpu_baseline(sample_scale, seed=101):
 
 
 # --------------------------------------------------------------------------
+# C4: SF100 Q18 (agg-dominant: ~150M groups)
+# --------------------------------------------------------------------------
 
-WORKLOADS = {"c2": C2, "c3": C3}
+class C4:
+    name = "c4"
+    CUST = 15_000_000
+    ORDERS = 150_000_000
+
+    def __init__(self, args, device, rank):
+        import torch
+        scale = args.c4_scale
+        g = torch.Generator(device=device)
+        g.manual_seed(5678 + rank)
+        nc = int(self.CUST * scale)
+        no = int(self.ORDERS * scale)
+        okeys = 4 * torch.randperm(no, generator=g, device=device,
+                                   dtype=torch.int64)
+        self.orders = [okeys,
+                       torch.randint(0, nc, (no,), generator=g, device=device,
+                                     dtype=torch.int64),
+                       torch.randint(0, 1 << 30, (no,), generator=g,
+                                     device=device, dtype=torch.int64)]
+        lines = torch.randint(1, 8, (no,), generator=g, device=device,
+                              dtype=torch.int64)
+        lkeys = torch.repeat_interleave(okeys, lines)
+        nl = lkeys.numel()
+        perm = torch.randperm(nl, generator=g, device=device)
+        lkeys = lkeys[perm].contiguous()
+        lqty = torch.randint(1, 51, (nl,), generator=g, device=device,
+                             dtype=torch.int64)
+        hot = torch.rand(nl, generator=g, device=device) < 1e-5
+        lqty = torch.where(hot, lqty + 400, lqty)
+        self.lineitem = [lkeys, lqty]
+        self.cust = [torch.arange(nc, dtype=torch.int64, device=device),
+                     torch.randint(0, 1 << 30, (nc,), generator=g,
+                                   device=device, dtype=torch.int64)]
+        self.n_probe = nl  # metric unit: lineitem rows aggregated
+        self.n_build = no
+        self.expected_groups = no
+        self.last_info = None
+
+    def step(self, lib, local_rank, distributed):
+        from galaxysql_amd.exchange import shuffle_columns
+        from galaxysql_amd.queries import (run_q18, Q18_LINEITEM_TYPES,
+                                           Q18_ORDERS_TYPES, Q18_CUST_TYPES)
+        cust, orders, lineitem = self.cust, self.orders, self.lineitem
+        if distributed:
+            # Q18's group key IS the shuffle key: after the orderkey
+            # all-to-allv the local aggregate is final (two-phase agg
+            # degenerates to one local phase — SURVEY.md §8e).
+            lineitem = shuffle_columns(lib, lineitem, Q18_LINEITEM_TYPES, [0],
+                                       device=local_rank)
+            orders = shuffle_columns(lib, orders, Q18_ORDERS_TYPES, [0],
+                                     device=local_rank)
+            cust = shuffle_columns(lib, cust, Q18_CUST_TYPES, [0],
+                                   device=local_rank)
+        _, info = run_q18(lib, local_rank, cust, orders, lineitem,
+                          expected_groups=self.expected_groups)
+        self.last_info = info
+        ast = info["agg_stats"]
+        return {"probe_kernel_ms": ast["kernel_ms"], "probe_launches":
+                ast["consumes"], "probe_rows": ast["rows"],
+                "matches": ast["groups"], "n_build": ast["groups"],
+                "groups": info["groups"]}
+
+    def config(self, world):
+        return {"workload": "C4_tpch_sf100_q18_groupby_having_join",
+                "lineitem_rows_per_gpu": self.n_probe,
+                "orders_rows_per_gpu": self.n_build,
+                "groups_last_step": self.last_info["groups"] if self.last_info else None,
+                "survivors_last_step": self.last_info["survivors"] if self.last_info else None,
+                "aggregates": "SUM(l_quantity) i64, HAVING > 300",
+                "exchange": "rccl_all_to_allv" if world > 1 else "none",
+                "parallelism": f"hash_shuffle_dp{world}",
+                "roofline_kernel": "agg insert+gid+accumulate chain"}
+
+    def roofline_bytes_per_row(self, acc):
+        # agg chain algorithmic bytes per input row: pack 16 wr
+        # + 4 hash + insert slot 16 (+compare 16 for dup rows ~75%) +
+        # accumulate slot 16 + compare 16 + record RMW 16
+        return 16 + 4 + 16 + 0.75 * 16 + 16 + 16 + 16
+
+    def cpu_baseline(self, sample_rows):
+        return c4_cpu_baseline(sample_scale=1.0 / 64)
+
+    def traffic_key(self):
+        return "c4_sf100"
+
+
+def c4_cpu_baseline(sample_scale, seed=103):
+    import torch
+    from galaxysql_amd import abi
+    from galaxysql_amd.queries import run_q18, gen_q18_numpy
+
+    lib = abi.load_oracle()
+    rng = np.random.default_rng(seed)
+    data = gen_q18_numpy(rng, n_cust=int(C4.CUST * sample_scale),
+                         n_orders=int(C4.ORDERS * sample_scale),
+                         having_frac=1e-5)
+    t = [[torch.from_numpy(a) for a in cols] for cols in data]
+    n_li = data[2][0].shape[0]
+    t0 = time.perf_counter()
+    run_q18(lib, -1, t[0], t[1], t[2])
+    t1 = time.perf_counter()
+    return {"value": n_li / (t1 - t0), "unit": "rows/s", "cores": 1,
+            "kind": "port",
+            "sample": f"full Q18 chain on a 1/{int(1/sample_scale)} SF100 "
+                      f"sample ({n_li} lineitem rows), agg+joins timed, "
+                      "1 thread, -O3 -march=native"}
+
+
+WORKLOADS = {"c2": C2, "c3": C3, "c4": C4}
 
 
 def main():
@@ -330,6 +440,8 @@ def main():
     ap.add_argument("--sf", type=int, default=None, help="C2 scale factor")
     ap.add_argument("--c3-scale", type=float, default=1.0,
                     help="C3 size fraction of SF100 (1.0 = full)")
+    ap.add_argument("--c4-scale", type=float, default=1.0,
+                    help="C4 size fraction of SF100 (1.0 = full)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--cpu-probe-rows", type=int, default=8_000_000)
     args = ap.parse_args()
@@ -393,7 +505,11 @@ def main():
 
     value = total_probe_rows_per_step * args.steps / elapsed
     n_build_avg = acc["n_build"] / args.steps
-    bpr = probe_bytes_per_row(int(n_build_avg), acc["matches"], acc["probe_rows"])
+    if hasattr(wl, "roofline_bytes_per_row"):
+        bpr = wl.roofline_bytes_per_row(acc)
+    else:
+        bpr = probe_bytes_per_row(int(n_build_avg), acc["matches"],
+                                  acc["probe_rows"])
     probe_ms = acc["probe_kernel_ms"]
     achieved = (bpr * acc["probe_rows"] / (probe_ms / 1e3)) if probe_ms else 0.0
 
