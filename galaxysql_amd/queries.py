@@ -191,7 +191,21 @@ def run_q18(lib, device, cust, orders, lineitem, having=300,
     out-of-scope rows), the operator work is the aggregate + joins.
 
     Returns (n_final_rows, info)."""
+    import os
+    import time
     import torch
+    dbg = os.environ.get("GX_DEBUG_TIMING")
+    tmark = [time.perf_counter()]
+
+    def mark(label):
+        if dbg:
+            if lineitem[0].is_cuda:
+                torch.cuda.synchronize()
+            now = time.perf_counter()
+            print(f"[q18] {label}: {(now - tmark[0]) * 1e3:.1f} ms",
+                  flush=True)
+            tmark[0] = now
+
     agg = HashAggExec(lib, group_cols=[0], aggs=[(abi.SUM_I64, 1)],
                       input_types=Q18_LINEITEM_TYPES,
                       expected_groups=expected_groups or lineitem[0].numel() // 4,
@@ -199,7 +213,9 @@ def run_q18(lib, device, cust, orders, lineitem, having=300,
     try:
         ka = []
         gc = chunk_from_torch(lib, list(lineitem), Q18_LINEITEM_TYPES, ka)
+        mark("agg create+stage")
         lib.check(lib.lib.gxop_agg_consume(agg._op, C.byref(gc)), "agg_consume")
+        mark("agg consume")
         agg.build_consume()
         agg_stats = agg.stats()
         parts = []
@@ -213,8 +229,10 @@ def run_q18(lib, device, cust, orders, lineitem, having=300,
             parts.append(result_to_tensors(lib, out, [I64, I64],
                                            lineitem[0].device))
             lib.lib.gxop_result_release(out)
+        mark("agg emit+copy")
     finally:
         agg.close()
+    mark("agg close")
 
     keys = torch.cat([p[0] for p in parts]) if parts else \
         torch.empty(0, dtype=torch.int64, device=lineitem[0].device)
@@ -224,6 +242,7 @@ def run_q18(lib, device, cust, orders, lineitem, having=300,
     skeys = keys[mask].contiguous()
     ssums = sums[mask].contiguous()
     n_surv = skeys.numel()
+    mark("having filter")
 
     # survivors ⋈ orders on orderkey (build = tiny survivors, probe = orders)
     ja = ParallelHashJoinExec(lib, abi.INNER, [EquiJoinKey(0, 0, I64)],
@@ -259,6 +278,7 @@ def run_q18(lib, device, cust, orders, lineitem, having=300,
     finally:
         jb.close()
 
+    mark("joins")
     info = {"groups": n_groups, "survivors": n_surv, "after_orders": n_r1,
             "final_rows": n_final, "agg_stats": agg_stats}
     return n_final, info
