@@ -204,69 +204,77 @@ def c2_cpu_baseline(n_orders, n_probe_sample, seed=99):
 # --------------------------------------------------------------------------
 
 class C3:
+    """HONEST SF100 Q3: unfiltered tables in HBM; the timed step runs the
+    device-side vectorized filter+project scans (customer segment, orders
+    date, lineitem shipdate; revenue = extendedprice*(1-discount) both as
+    f64 and exact scaled-int), then the join chain + 3-key aggregate."""
     name = "c3"
-    # SF100 pre-filtered cardinalities:
-    # customer 15M * 1/5 = 3M build; orders 150M * 48.6%; lineitem 600M * 54%
     CUST_TOTAL = 15_000_000
     ORDERS_TOTAL = 150_000_000
-    CUST_SEL = 0.2
-    ORDERS_SEL = 0.486
-    LINEITEM = 324_000_000
+    LINEITEM = 600_000_000
 
     def __init__(self, args, device, rank):
         import torch
+        from galaxysql_amd.queries import Q3_SHIP_CUTOFF
         scale = args.c3_scale
         g = torch.Generator(device=device)
         g.manual_seed(4321 + rank)
         ct = int(self.CUST_TOTAL * scale)
         ot = int(self.ORDERS_TOTAL * scale)
         li = int(self.LINEITEM * scale)
-        n_cust = int(ct * self.CUST_SEL)
-        n_orders = int(ot * self.ORDERS_SEL)
-        self.cust = [torch.randperm(ct, generator=g, device=device,
-                                    dtype=torch.int64)[:n_cust].contiguous()]
+        self.cust = [torch.arange(ct, dtype=torch.int64, device=device),
+                     torch.randint(0, 5, (ct,), generator=g, device=device,
+                                   dtype=torch.int32)]
         okeys_all = 4 * torch.randperm(ot, generator=g, device=device,
                                        dtype=torch.int64)
         self.orders = [
-            torch.randint(0, ct, (n_orders,), generator=g, device=device,
+            torch.randint(0, ct, (ot,), generator=g, device=device,
                           dtype=torch.int64),
-            okeys_all[:n_orders].contiguous(),
-            torch.randint(8000, 9500, (n_orders,), generator=g, device=device,
+            okeys_all,
+            torch.randint(8000, 9500, (ot,), generator=g, device=device,
                           dtype=torch.int32),
-            torch.zeros(n_orders, dtype=torch.int32, device=device),
+            torch.zeros(ot, dtype=torch.int32, device=device),
         ]
         lkeys = okeys_all[torch.randint(0, ot, (li,), generator=g,
-                                        device=device)]
+                                        device=device)].contiguous()
+        ship = torch.randint(8000, 9500, (li,), generator=g, device=device,
+                             dtype=torch.int32)
         cents = torch.randint(100, 10_000_000, (li,), generator=g,
                               device=device, dtype=torch.int64)
-        self.lineitem = [lkeys.contiguous(), cents.double() / 100.0, cents]
-        self.n_probe = li
-        self.n_build = n_orders  # join2 build upper bound (pre-SEMI)
-        self.expected_groups = int(n_orders * self.CUST_SEL) + 1024
+        disc = torch.randint(0, 11, (li,), generator=g, device=device,
+                             dtype=torch.int64)
+        self.lineitem = [lkeys, ship, cents.double() / 100.0,
+                         disc.double() / 100.0, cents, disc]
+        # metric unit = join-probe rows = lineitem rows surviving the scan
+        self.n_probe = int((ship > Q3_SHIP_CUTOFF).sum().item())
+        self.n_scan = li
+        self.expected_groups = int(ot * 0.486 * 0.2) + 1024
         self.last_info = None
 
     def step(self, lib, local_rank, distributed):
-        from galaxysql_amd.chunk import I64, I32, F64
         from galaxysql_amd.exchange import shuffle_columns
-        from galaxysql_amd.queries import run_q3, CUST_TYPES, ORDERS_TYPES, \
-            LINEITEM_TYPES
-
-        cust, orders, lineitem = self.cust, self.orders, self.lineitem
-        if distributed:
-            cust = shuffle_columns(lib, cust, CUST_TYPES, [0], device=local_rank)
+        from galaxysql_amd.queries import (run_q3, run_q3_honest, CUST_TYPES,
+                                           ORDERS_TYPES, LINEITEM_TYPES)
+        if not distributed:
+            _, info = run_q3_honest(lib, local_rank, self.cust, self.orders,
+                                    self.lineitem,
+                                    expected_groups=self.expected_groups,
+                                    to_host=False)
+        else:
+            (cust, orders, lineitem), scanned = run_q3_honest(
+                lib, local_rank, self.cust, self.orders, self.lineitem,
+                as_tensors=True)
+            cust = shuffle_columns(lib, cust, CUST_TYPES, [0],
+                                   device=local_rank)
             orders = shuffle_columns(lib, orders, ORDERS_TYPES, [0],
                                      device=local_rank)
-            # join2 side shuffles by orderkey: orders result rows move with
-            # their orderkey hash; Q3's group key is orderkey-led, so the
-            # local aggregate after this shuffle is final.
             lineitem = shuffle_columns(lib, lineitem, LINEITEM_TYPES, [0],
                                        device=local_rank)
-            # join1 runs on custkey-sharded data; run_q3 re-shuffles its
-            # RESULT by o_orderkey (col 1) before join2.
-        _, info = run_q3(lib, local_rank, cust, orders, lineitem,
-                         expected_groups=self.expected_groups, to_host=False,
-                         reshuffle_by_orderkey=distributed,
-                         local_rank=local_rank)
+            _, info = run_q3(lib, local_rank, cust, orders, lineitem,
+                             expected_groups=self.expected_groups,
+                             to_host=False, reshuffle_by_orderkey=True,
+                             local_rank=local_rank)
+            info.update(scanned)
         self.last_info = info
         st = dict(info["join2_stats"])
         st["n_build"] = info["orders_kept"]
@@ -274,13 +282,16 @@ class C3:
         return st
 
     def config(self, world):
+        li = self.last_info or {}
         return {"workload": "C3_tpch_sf100_q3_join_chain_groupby",
-                "customer_build_rows_per_gpu": self.cust[0].numel(),
+                "customer_rows_per_gpu": self.cust[0].numel(),
                 "orders_rows_per_gpu": self.orders[0].numel(),
-                "lineitem_rows_per_gpu": self.n_probe,
-                "groups_last_step": self.last_info["groups"] if self.last_info else None,
-                "aggregates": "SUM(revenue) f64 + SUM(cents) i64 + COUNT(*)",
-                "inputs": "pre-filtered, revenue pre-projected (DESIGN.md)",
+                "lineitem_rows_scanned_per_gpu": self.n_scan,
+                "lineitem_rows_joined_per_gpu": self.n_probe,
+                "groups_last_step": li.get("groups"),
+                "aggregates": "SUM(revenue) f64 + SUM(rev_scaled4) i64 + COUNT(*)",
+                "inputs": "UNFILTERED; segment/date/shipdate filters + "
+                          "revenue projection run on device in the timed step",
                 "exchange": "rccl_all_to_allv" if world > 1 else "none",
                 "parallelism": f"hash_shuffle_dp{world}"}
 
@@ -292,28 +303,29 @@ class C3:
 
 
 def c3_cpu_baseline(sample_scale, seed=101):
-    """Oracle Q3 chain on a bounded sample (~1/64 of SF100), whole chain
-    timed (both builds + probes + aggregate), single thread."""
+    """Oracle honest-Q3 chain on a bounded sample (~1/64 of SF100): scans +
+    builds + probes + aggregate all timed, single thread."""
     import torch
     from galaxysql_amd import abi
-    from galaxysql_amd.queries import run_q3, gen_q3_numpy
+    from galaxysql_amd.queries import run_q3_honest, gen_q3_raw_numpy
 
     lib = abi.load_oracle()
     rng = np.random.default_rng(seed)
-    data = gen_q3_numpy(rng,
-                        n_cust_total=int(C3.CUST_TOTAL * sample_scale),
-                        n_orders_total=int(C3.ORDERS_TOTAL * sample_scale),
-                        n_lineitem=int(C3.LINEITEM * sample_scale))
+    data = gen_q3_raw_numpy(rng,
+                            n_cust=int(C3.CUST_TOTAL * sample_scale),
+                            n_orders=int(C3.ORDERS_TOTAL * sample_scale),
+                            n_lineitem=int(C3.LINEITEM * sample_scale))
     t = [[torch.from_numpy(a) for a in cols] for cols in data]
     t0 = time.perf_counter()
-    _, info = run_q3(lib, -1, t[0], t[1], t[2], to_host=False)
+    _, info = run_q3_honest(lib, -1, t[0], t[1], t[2], to_host=False)
     t1 = time.perf_counter()
-    n_li = data[2][0].shape[0]
+    n_li = info["lineitem_kept"]
     return {"value": n_li / (t1 - t0), "unit": "rows/s", "cores": 1,
             "kind": "port",
-            "sample": f"full Q3 chain on a 1/{int(1/sample_scale)} SF100 "
-                      f"sample ({n_li} lineitem rows), builds+probes+agg all "
-                      "timed, 1 thread, -O3 -march=native"}
+            "sample": f"honest Q3 chain (scans+joins+agg timed) on a "
+                      f"1/{int(1/sample_scale)} SF100 sample "
+                      f"({data[2][0].shape[0]} lineitem rows scanned, "
+                      f"{n_li} joined), 1 thread, -O3 -march=native"}
 
 
 # --------------------------------------------------------------------------

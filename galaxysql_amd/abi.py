@@ -118,8 +118,30 @@ class GxPartCfg(C.Structure):
     ]
 
 
+class GxPred(C.Structure):
+    _fields_ = [("col", C.c_int32), ("cmp", C.c_int32),
+                ("v_i64", C.c_int64), ("v_f64", C.c_double)]
+
+
+class GxProj(C.Structure):
+    _fields_ = [("op", C.c_int32), ("a", C.c_int32), ("b", C.c_int32)]
+
+
+class GxScanCfg(C.Structure):
+    _fields_ = [
+        ("n_preds", C.c_int32), ("preds", C.POINTER(GxPred)),
+        ("n_projs", C.c_int32), ("projs", C.POINTER(GxProj)),
+        ("n_input_cols", C.c_int32), ("input_types", C.POINTER(C.c_int32)),
+        ("device", C.c_int32), ("stream", C.c_uint64),
+    ]
+
+
 # Join types (gx_join_type)
 INNER, LEFT, RIGHT, SEMI, ANTI = 0, 1, 2, 3, 4
+# Comparisons (gx_cmp)
+LT, LE, GT, GE, EQ, NE = 0, 1, 2, 3, 4, 5
+# Projections (gx_proj_op)
+PROJ_COPY, PROJ_REV_F64, PROJ_REV_SCALED4 = 0, 1, 2
 # Agg funcs (gx_agg_func)
 COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64, MAX_F64 = range(8)
 
@@ -162,6 +184,11 @@ class GxLib:
                                                C.POINTER(C.POINTER(GxResult)),
                                                C.POINTER(C.c_int64)]
         L.gxop_part_close.argtypes = [C.c_void_p]
+        L.gxop_scan_create.restype = C.c_void_p
+        L.gxop_scan_create.argtypes = [C.POINTER(GxScanCfg)]
+        L.gxop_scan_consume.argtypes = [C.c_void_p, C.POINTER(GxChunk),
+                                        C.POINTER(C.POINTER(GxResult))]
+        L.gxop_scan_close.argtypes = [C.c_void_p]
         L.gxop_result_copy_col.argtypes = [C.POINTER(GxResult), C.c_int32,
                                            C.c_void_p, C.c_void_p]
         L.gxop_join_get_stats.argtypes = [C.c_void_p, C.POINTER(GxJoinStats)]

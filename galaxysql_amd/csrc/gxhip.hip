@@ -427,7 +427,7 @@ static inline int gx_grid(int64_t n, int block = 256) {
 
 /* ========================= operator base =============================== */
 
-enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3 };
+enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4 };
 
 struct gx_op {
     int kind;
@@ -904,6 +904,7 @@ struct JoinOp : gx_op {
 /* ======================= agg + partition in gxhip_agg.inc ============== */
 #include "gxhip_agg.inc"
 #include "gxhip_part.inc"
+#include "gxhip_scan.inc"
 
 /* ========================= C ABI ======================================= */
 

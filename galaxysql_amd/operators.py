@@ -210,6 +210,59 @@ class PartitioningExchanger:
             pass
 
 
+class ScanExec:
+    """Vectorized filter + project (executor/vectorized/ stage,
+    SURVEY.md §8f row 1). preds: (col, abi.LT.., const); projs:
+    (abi.PROJ_COPY, a, -1) / (abi.PROJ_REV_F64, a, b) / ..."""
+
+    def __init__(self, lib, preds, projs, input_types, device=-1, stream=0):
+        from .abi import GxPred, GxProj, GxScanCfg
+        self._lib = lib
+        self._keep = []
+        pa = (GxPred * max(1, len(preds)))()
+        for i, (col, cmp, const) in enumerate(preds):
+            pa[i] = GxPred(col, cmp,
+                           int(const) if not isinstance(const, float) else 0,
+                           float(const))
+        pj = (GxProj * len(projs))()
+        for i, (op, a, b) in enumerate(projs):
+            pj[i] = GxProj(op, a, b)
+        it = (C.c_int32 * len(input_types))(*input_types)
+        cfg = GxScanCfg(n_preds=len(preds), preds=pa, n_projs=len(projs),
+                        projs=pj, n_input_cols=len(input_types),
+                        input_types=it, device=device, stream=stream)
+        self._keep += [pa, pj, it, cfg]
+        self._op = lib.lib.gxop_scan_create(C.byref(cfg))
+        if not self._op:
+            raise RuntimeError(f"gxop_scan_create: {lib.error()}")
+
+    def consume_chunk(self, chunk: Chunk) -> Chunk:
+        ka = []
+        gc = self._lib.to_gx_chunk(chunk, ka)
+        out = C.POINTER(GxResult)()
+        self._lib.check(self._lib.lib.gxop_scan_consume(self._op, C.byref(gc),
+                                                        C.byref(out)),
+                        "scan_consume")
+        return self._lib.result_to_chunk(out) if out else None
+
+    def consume_raw(self, gx_chunk_ref):
+        out = C.POINTER(GxResult)()
+        self._lib.check(self._lib.lib.gxop_scan_consume(
+            self._op, gx_chunk_ref, C.byref(out)), "scan_consume")
+        return out
+
+    def close(self):
+        if self._op:
+            self._lib.lib.gxop_scan_close(self._op)
+            self._op = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
 def run_join(lib, join_type, join_keys, build_chunks, probe_chunks,
              outer_types, inner_types, **kw):
     """Convenience: full consume -> build -> probe -> tail pass; returns the

@@ -45,6 +45,31 @@ def _probe_tensors(lib, op, tensors, types):
     return out
 
 
+def _consume_src(lib, op, src, types):
+    """src: list of torch tensors OR a gx_result pointer (device-chained)."""
+    if isinstance(src, list):
+        _consume_tensors(lib, op, src, types)
+    else:
+        lib.check(lib.lib.gxop_join_consume(op._op,
+                                            C.byref(src.contents.chunk)),
+                  "join_consume")
+
+
+def _probe_src(lib, op, src, types):
+    if isinstance(src, list):
+        return _probe_tensors(lib, op, src, types)
+    out = C.POINTER(GxResult)()
+    lib.check(lib.lib.gxop_join_probe(op._op, C.byref(src.contents.chunk),
+                                      C.byref(out)), "join_probe")
+    return out
+
+
+def _src_rows(src):
+    if isinstance(src, list):
+        return src[0].numel()
+    return src.contents.chunk.n_rows
+
+
 def _probe_result(lib, op, res):
     out = C.POINTER(GxResult)()
     lib.check(lib.lib.gxop_join_probe(op._op, C.byref(res.contents.chunk),
@@ -87,14 +112,16 @@ def run_q3(lib, device, cust, orders, lineitem, expected_groups=0,
     Returns (result_rows_chunk, info) where info carries join2 kernel stats
     + intermediate cardinalities. The caller owns nothing to free."""
     # join1: SEMI — build customer, probe orders on o_custkey
+    # (cust/orders/lineitem may be tensor lists or device-resident
+    #  gx_results, e.g. scan outputs — run_q3_honest)
     j1 = ParallelHashJoinExec(
         lib, abi.SEMI, [EquiJoinKey(0, 0, I64)],
         outer_types=ORDERS_TYPES, inner_types=CUST_TYPES,
-        device=device, expected_build_rows=cust[0].numel())
+        device=device, expected_build_rows=_src_rows(cust))
     try:
-        _consume_tensors(lib, j1, list(cust), CUST_TYPES)
+        _consume_src(lib, j1, cust, CUST_TYPES)
         j1.build_consume()
-        r1 = _probe_tensors(lib, j1, list(orders), ORDERS_TYPES)
+        r1 = _probe_src(lib, j1, orders, ORDERS_TYPES)
         n_orders_kept = r1.contents.chunk.n_rows if r1 else 0
 
         # join2: INNER — build the surviving orders (key col 1 = o_orderkey),
@@ -126,7 +153,7 @@ def run_q3(lib, device, cust, orders, lineitem, expected_groups=0,
             if r1:
                 lib.lib.gxop_result_release(r1)
                 r1 = None
-            r2 = _probe_tensors(lib, j2, list(lineitem), LINEITEM_TYPES)
+            r2 = _probe_src(lib, j2, lineitem, LINEITEM_TYPES)
             n_joined = r2.contents.chunk.n_rows if r2 else 0
             j2_stats = j2.stats()
 
@@ -174,6 +201,96 @@ def run_q3(lib, device, cust, orders, lineitem, expected_groups=0,
             j2.close()
     finally:
         j1.close()
+
+
+# ---- honest Q3: unfiltered inputs, filters+projection run ON DEVICE ----
+
+RAW_CUST_TYPES = [I64, I32]                       # c_custkey, c_mktsegment
+RAW_ORDERS_TYPES = [I64, I64, I32, I32]           # o_custkey, o_orderkey, o_orderdate, o_shippriority
+RAW_LINEITEM_TYPES = [I64, I32, F64, F64, I64, I64]
+# l_orderkey, l_shipdate, l_extendedprice, l_discount, price_cents, disc_hundredths
+
+Q3_DATE_CUTOFF = 8729      # o_orderdate < cutoff  (~48.6 % of [8000,9500))
+Q3_SHIP_CUTOFF = 8690      # l_shipdate  > cutoff  (~54 %)
+Q3_SEGMENT = 0             # c_mktsegment == 0     (1 of 5)
+
+
+def run_q3_honest(lib, device, raw_cust, raw_orders, raw_lineitem,
+                  expected_groups=0, to_host=True, as_tensors=False,
+                  local_rank=0):
+    """Q3 with the vectorized filter/projection stage ON DEVICE (SURVEY.md
+    §8f row 1): scans filter customer by segment, orders by date, lineitem
+    by shipdate — and project revenue = extendedprice*(1-discount) as f64
+    and as exact scale-4 scaled-int — then the join chain + aggregate run
+    on the scan outputs without leaving HBM.
+
+    as_tensors: return the scan outputs as tensors for the N>1 shuffle
+    path instead of chaining device results (costs one copy)."""
+    from .operators import ScanExec
+
+    sc = ScanExec(lib, [(1, abi.EQ, Q3_SEGMENT)], [(abi.PROJ_COPY, 0, -1)],
+                  RAW_CUST_TYPES, device=device)
+    so = ScanExec(lib, [(2, abi.LT, Q3_DATE_CUTOFF)],
+                  [(abi.PROJ_COPY, 0, -1), (abi.PROJ_COPY, 1, -1),
+                   (abi.PROJ_COPY, 2, -1), (abi.PROJ_COPY, 3, -1)],
+                  RAW_ORDERS_TYPES, device=device)
+    sl = ScanExec(lib, [(1, abi.GT, Q3_SHIP_CUTOFF)],
+                  [(abi.PROJ_COPY, 0, -1), (abi.PROJ_REV_F64, 2, 3),
+                   (abi.PROJ_REV_SCALED4, 4, 5)],
+                  RAW_LINEITEM_TYPES, device=device)
+    rc = ro = rl = None
+    try:
+        ka = []
+        rc = sc.consume_raw(C.byref(chunk_from_torch(lib, list(raw_cust),
+                                                     RAW_CUST_TYPES, ka)))
+        ro = so.consume_raw(C.byref(chunk_from_torch(lib, list(raw_orders),
+                                                     RAW_ORDERS_TYPES, ka)))
+        rl = sl.consume_raw(C.byref(chunk_from_torch(lib, list(raw_lineitem),
+                                                     RAW_LINEITEM_TYPES, ka)))
+        scanned = {"cust_kept": rc.contents.chunk.n_rows if rc else 0,
+                   "orders_kept_scan": ro.contents.chunk.n_rows if ro else 0,
+                   "lineitem_kept": rl.contents.chunk.n_rows if rl else 0}
+        if as_tensors:
+            dev = raw_cust[0].device
+            cust = result_to_tensors(lib, rc, CUST_TYPES, dev)
+            orders = result_to_tensors(lib, ro, ORDERS_TYPES, dev)
+            lineitem = result_to_tensors(lib, rl, LINEITEM_TYPES, dev)
+            lib.lib.gxop_result_release(rc)
+            lib.lib.gxop_result_release(ro)
+            lib.lib.gxop_result_release(rl)
+            rc = ro = rl = None
+            return (cust, orders, lineitem), scanned
+        chunks, info = run_q3(lib, device, rc, ro, rl,
+                              expected_groups=expected_groups,
+                              to_host=to_host, local_rank=local_rank)
+        info.update(scanned)
+        return chunks, info
+    finally:
+        for r in (rc, ro, rl):
+            if r:
+                lib.lib.gxop_result_release(r)
+        sc.close()
+        so.close()
+        sl.close()
+
+
+def gen_q3_raw_numpy(rng, n_cust, n_orders, n_lineitem):
+    """UNFILTERED Q3-shaped inputs for the honest path (numpy)."""
+    import numpy as np
+    cust = [np.arange(n_cust, dtype=np.int64),
+            rng.integers(0, 5, n_cust).astype(np.int32)]
+    okeys = 4 * rng.permutation(n_orders).astype(np.int64)
+    orders = [rng.integers(0, n_cust, n_orders).astype(np.int64),
+              okeys,
+              rng.integers(8000, 9500, n_orders).astype(np.int32),
+              np.zeros(n_orders, dtype=np.int32)]
+    lkey = okeys[rng.integers(0, n_orders, n_lineitem)]
+    ship = rng.integers(8000, 9500, n_lineitem).astype(np.int32)
+    cents = rng.integers(100, 10_000_000, n_lineitem).astype(np.int64)
+    disc = rng.integers(0, 11, n_lineitem).astype(np.int64)
+    lineitem = [lkey, ship, cents.astype(np.float64) / 100.0,
+                disc.astype(np.float64) / 100.0, cents, disc]
+    return cust, orders, lineitem
 
 
 Q18_LINEITEM_TYPES = [I64, I64]          # l_orderkey, l_quantity

@@ -208,6 +208,52 @@ int gxop_part_consume_concat(gx_op *op, const gx_chunk *chunk,
                              gx_result **out, int64_t *counts);
 int gxop_part_close(gx_op *op);
 
+/* ---- scan: vectorized filter + project --------------------------------- */
+
+/* Mirrors the vectorized filter/projection stage (executor/vectorized/,
+ * VectorizedFilterExec + projections — SURVEY.md §8f row 1): one stateless
+ * pass per chunk, AND-of-predicates selection + projected output columns.
+ * Null comparison semantics are SQL: a NULL operand fails every predicate. */
+
+typedef enum gx_cmp {
+    GX_CMP_LT = 0, GX_CMP_LE, GX_CMP_GT, GX_CMP_GE, GX_CMP_EQ, GX_CMP_NE
+} gx_cmp;
+
+typedef struct gx_pred {
+    int32_t col;
+    int32_t cmp;       /* gx_cmp */
+    int64_t v_i64;     /* constant for I64/I32 columns */
+    double  v_f64;     /* constant for F64 columns */
+} gx_pred;
+
+typedef enum gx_proj_op {
+    GX_PROJ_COPY = 0,      /* out = col a */
+    GX_PROJ_REV_F64 = 1,   /* out = a * (1 - b), doubles (Q3/Q9 revenue) */
+    GX_PROJ_REV_SCALED4 = 2/* out = a * (100 - b), i64 cents x hundredths ->
+                              DECIMAL scale-4, exact */
+} gx_proj_op;
+
+typedef struct gx_proj {
+    int32_t op;        /* gx_proj_op */
+    int32_t a, b;      /* input columns */
+} gx_proj;
+
+typedef struct gx_scan_cfg {
+    int32_t n_preds;
+    const gx_pred *preds;
+    int32_t n_projs;
+    const gx_proj *projs;
+    int32_t n_input_cols;
+    const int32_t *input_types;
+    int32_t device;
+    uint64_t stream;
+} gx_scan_cfg;
+
+gx_op *gxop_scan_create(const gx_scan_cfg *cfg);
+/* filter+project one chunk; *out receives the surviving projected rows. */
+int gxop_scan_consume(gx_op *op, const gx_chunk *chunk, gx_result **out);
+int gxop_scan_close(gx_op *op);
+
 /* ---- results / errors -------------------------------------------------- */
 
 int gxop_result_to_host(gx_result *res);

@@ -349,7 +349,7 @@ static gx_result *make_result(std::vector<OutCol> &&cols, int32_t n_rows) {
 
 /* ---- operator base ----------------------------------------------------- */
 
-enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3 };
+enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4 };
 
 } // anonymous namespace
 
@@ -919,6 +919,99 @@ struct PartOp : gx_op {
     }
 };
 
+/* ---- scan (vectorized filter + project) ---------------------------------
+ * Restates the vectorized filter/projection stage (executor/vectorized/,
+ * SURVEY.md §8f row 1): AND of predicates, SQL NULL-fails semantics,
+ * projected output. */
+struct ScanOp : gx_op {
+    gx_scan_cfg cfg;
+    std::vector<gx_pred> preds;
+    std::vector<gx_proj> projs;
+    std::vector<int32_t> input_types, out_types;
+
+    ScanOp(const gx_scan_cfg *c) : gx_op(OP_SCAN), cfg(*c) {
+        preds.assign(c->preds, c->preds + c->n_preds);
+        projs.assign(c->projs, c->projs + c->n_projs);
+        input_types.assign(c->input_types, c->input_types + c->n_input_cols);
+        for (auto &p : projs) {
+            switch (p.op) {
+            case GX_PROJ_COPY: out_types.push_back(input_types[p.a]); break;
+            case GX_PROJ_REV_F64: out_types.push_back(GX_F64); break;
+            default: out_types.push_back(GX_I64); break;
+            }
+        }
+    }
+
+    static bool cmp_ok_i(int64_t a, int32_t cmp, int64_t b) {
+        switch (cmp) {
+        case GX_CMP_LT: return a < b;
+        case GX_CMP_LE: return a <= b;
+        case GX_CMP_GT: return a > b;
+        case GX_CMP_GE: return a >= b;
+        case GX_CMP_EQ: return a == b;
+        default: return a != b;
+        }
+    }
+    static bool cmp_ok_f(double a, int32_t cmp, double b) {
+        switch (cmp) {
+        case GX_CMP_LT: return a < b;
+        case GX_CMP_LE: return a <= b;
+        case GX_CMP_GT: return a > b;
+        case GX_CMP_GE: return a >= b;
+        case GX_CMP_EQ: return a == b;
+        default: return a != b;
+        }
+    }
+
+    int consume(const gx_chunk *ch, gx_result **out) {
+        *out = nullptr;
+        Store in;
+        in.init((int32_t)input_types.size(), input_types.data());
+        if (in.append(ch) != 0) return -1;
+        std::vector<OutCol> cols(out_types.size());
+        for (size_t c = 0; c < out_types.size(); c++) cols[c].type = out_types[c];
+        int32_t kept = 0;
+        for (size_t r = 0; r < in.n_rows; r++) {
+            bool pass = true;
+            for (auto &p : preds) {
+                const Column &c = in.cols[p.col];
+                if (c.is_null(r)) { pass = false; break; }
+                bool ok;
+                switch (c.type) {
+                case GX_I64: ok = cmp_ok_i(c.i64v[r], p.cmp, p.v_i64); break;
+                case GX_I32: ok = cmp_ok_i((int64_t)c.i32v[r], p.cmp, p.v_i64); break;
+                default: ok = cmp_ok_f(c.f64v[r], p.cmp, p.v_f64); break;
+                }
+                if (!ok) { pass = false; break; }
+            }
+            if (!pass) continue;
+            kept++;
+            for (size_t c = 0; c < projs.size(); c++) {
+                const gx_proj &pj = projs[c];
+                const Column &a = in.cols[pj.a];
+                switch (pj.op) {
+                case GX_PROJ_COPY:
+                    cols[c].append_from(a, r);
+                    break;
+                case GX_PROJ_REV_F64: {
+                    const Column &b = in.cols[pj.b];
+                    if (a.is_null(r) || b.is_null(r)) cols[c].append_null();
+                    else cols[c].append_f64(a.f64v[r] * (1.0 - b.f64v[r]));
+                    break; }
+                default: { /* REV_SCALED4 */
+                    const Column &b = in.cols[pj.b];
+                    if (a.is_null(r) || b.is_null(r)) cols[c].append_null();
+                    else cols[c].append_i64((int64_t)((uint64_t)a.i64v[r] *
+                                            (uint64_t)(100 - b.i64v[r])));
+                    break; }
+                }
+            }
+        }
+        *out = make_result(std::move(cols), kept);
+        return 0;
+    }
+};
+
 } // anonymous namespace
 
 /* ---- C ABI ------------------------------------------------------------- */
@@ -1008,6 +1101,16 @@ int gxop_agg_get_stats(gx_op *op, gx_agg_stats *out) {
     out->groups = static_cast<AggOp *>(op)->n_groups();
     return 0;
 }
+
+gx_op *gxop_scan_create(const gx_scan_cfg *cfg) {
+    if (!cfg || cfg->n_projs <= 0) { set_err("bad scan cfg"); return nullptr; }
+    return new ScanOp(cfg);
+}
+int gxop_scan_consume(gx_op *op, const gx_chunk *c, gx_result **out) {
+    if (!op || op->kind != OP_SCAN) { set_err("not a scan op"); return -1; }
+    return static_cast<ScanOp *>(op)->consume(c, out);
+}
+int gxop_scan_close(gx_op *op) { delete op; return 0; }
 
 int gxop_result_to_host(gx_result *) { return 0; /* oracle results are host */ }
 void gxop_result_release(gx_result *res) {

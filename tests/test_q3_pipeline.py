@@ -83,3 +83,70 @@ def test_q3_hip_vs_oracle():
     strip_f64 = lambda rows: [(a, b, c, e, f) for a, b, c, d, e, f in rows]
     assert multiset(strip_f64(got)) == multiset(strip_f64(ref))
     assert multiset(got, f64_round=4) == multiset(ref, f64_round=4)
+
+
+# ---- honest Q3 (device-side filter+project) ----
+
+def numpy_q3_honest(cust, orders, lineitem):
+    from galaxysql_amd.queries import Q3_DATE_CUTOFF, Q3_SHIP_CUTOFF, Q3_SEGMENT
+    ck, cseg = cust
+    keep_c = set(int(k) for k, s in zip(ck, cseg) if s == Q3_SEGMENT)
+    ocust, okey, odate, oprio = orders
+    order_info = {}
+    for c, k, d, p in zip(ocust, okey, odate, oprio):
+        if d < Q3_DATE_CUTOFF and int(c) in keep_c:
+            order_info[int(k)] = (int(d), int(p))
+    lkey, ship, pf, df, cents, disc = lineitem
+    groups = {}
+    for k, sdt, a, b, cc, dd in zip(lkey, ship, pf, df, cents, disc):
+        if not sdt > Q3_SHIP_CUTOFF:
+            continue
+        info = order_info.get(int(k))
+        if info is None:
+            continue
+        key = (int(k), info[0], info[1])
+        s = groups.setdefault(key, [0.0, 0, 0])
+        s[0] += float(a) * (1.0 - float(b))
+        s[1] += int(cc) * (100 - int(dd))
+        s[2] += 1
+    return [(k[0], k[1], k[2], v[0], v[1], v[2]) for k, v in groups.items()]
+
+
+def honest_rows(lib, device, data):
+    from galaxysql_amd.queries import run_q3_honest
+    t = [[torch.from_numpy(a) for a in cols] for cols in data]
+    if device >= 0:
+        t = [[x.cuda(device) for x in cols] for cols in t]
+    chunks, info = run_q3_honest(lib, device, t[0], t[1], t[2], to_host=True)
+    rows = []
+    for c in chunks:
+        rows.extend(c.rows())
+    return rows, info
+
+
+def test_q3_honest_oracle_vs_numpy(oracle):
+    from galaxysql_amd.queries import gen_q3_raw_numpy
+    rng = np.random.default_rng(81)
+    data = gen_q3_raw_numpy(rng, n_cust=3000, n_orders=25000, n_lineitem=100000)
+    got, info = honest_rows(oracle, -1, data)
+    exp = numpy_q3_honest(*data)
+    assert info["groups"] == len(exp)
+    assert multiset(got, f64_round=6) == multiset(exp, f64_round=6)
+
+
+@pytest.mark.gpu
+def test_q3_honest_hip_vs_oracle():
+    from galaxysql_amd.queries import gen_q3_raw_numpy
+    oracle = abi.load_oracle()
+    hip = abi.load_hip()
+    rng = np.random.default_rng(82)
+    data = gen_q3_raw_numpy(rng, n_cust=30000, n_orders=400000,
+                            n_lineitem=1_500_000)
+    ref, ri = honest_rows(oracle, -1, data)
+    got, gi = honest_rows(hip, 0, data)
+    for k in ("cust_kept", "orders_kept_scan", "lineitem_kept",
+              "orders_kept", "joined_rows", "groups"):
+        assert gi[k] == ri[k], k
+    strip_f64 = lambda rows: [(a, b, c, e, f) for a, b, c, d, e, f in rows]
+    assert multiset(strip_f64(got)) == multiset(strip_f64(ref))
+    assert multiset(got, f64_round=4) == multiset(ref, f64_round=4)
