@@ -195,6 +195,13 @@ struct DevColView {
     const uint8_t *bytes;
 };
 
+/* rebase chunk-local slice end-offsets to store-global offsets */
+__global__ void k_rebase_i32(int32_t *off, int64_t n, int32_t base) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        off[i] += base;
+}
+
 struct DevStore {
     std::vector<DevColumn> cols;
     int64_t n_rows = 0;
@@ -253,7 +260,15 @@ struct DevStore {
                     HIP_OK(hipMemcpyAsync((char *)c.bytes.p + c.byte_len, b->data,
                                           (size_t)blen, kind, stream));
                 }
-                c.byte_len += blen; /* caller must rebase offsets; see append_fixup */
+                if (c.byte_len > 0 && n > 0) {
+                    int64_t g = (n + 255) / 256;
+                    if (g > 4096) g = 4096;
+                    hipLaunchKernelGGL(k_rebase_i32, dim3((uint32_t)g), dim3(256),
+                                       0, stream,
+                                       (int32_t *)c.offsets.p + old, n,
+                                       (int32_t)c.byte_len);
+                }
+                c.byte_len += blen;
             }
             /* nulls: u8 per row; lazily materialize zeros when first needed */
             if (b->nulls) {

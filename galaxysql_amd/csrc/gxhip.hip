@@ -50,6 +50,63 @@ struct ColViews {
     DevColView col[GX_MAX_COLS];
 };
 
+
+/* airlift Slice.hashCode = (int) XxHash64(seed 0) over raw bytes
+ * (chunk/SliceBlock.java:183-195, non-"compatible" path) — device
+ * restatement of the published algorithm; placement-only. */
+#define XXP1 0x9E3779B185EBCA87ull
+#define XXP2 0xC2B2AE3D27D4EB4Full
+#define XXP3 0x165667B19E3779F9ull
+#define XXP4 0x85EBCA77C2B2AE63ull
+#define XXP5 0x27D4EB2F165667C5ull
+
+__device__ static inline uint64_t xx_rotl(uint64_t x, int r) {
+    return (x << r) | (x >> (64 - r));
+}
+__device__ static uint64_t xxhash64_dev(const uint8_t *data, int64_t len) {
+    const uint8_t *p = data, *end = data + len;
+    uint64_t h;
+    if (len >= 32) {
+        uint64_t v1 = XXP1 + XXP2, v2 = XXP2, v3 = 0, v4 = (uint64_t)0 - XXP1;
+        const uint8_t *limit = end - 32;
+        do {
+            uint64_t a, b, c, d;
+            __builtin_memcpy(&a, p, 8); __builtin_memcpy(&b, p + 8, 8);
+            __builtin_memcpy(&c, p + 16, 8); __builtin_memcpy(&d, p + 24, 8);
+            v1 = xx_rotl(v1 + a * XXP2, 31) * XXP1;
+            v2 = xx_rotl(v2 + b * XXP2, 31) * XXP1;
+            v3 = xx_rotl(v3 + c * XXP2, 31) * XXP1;
+            v4 = xx_rotl(v4 + d * XXP2, 31) * XXP1;
+            p += 32;
+        } while (p <= limit);
+        h = xx_rotl(v1, 1) + xx_rotl(v2, 7) + xx_rotl(v3, 12) + xx_rotl(v4, 18);
+        h ^= xx_rotl(v1 * XXP2, 31) * XXP1; h = h * XXP1 + XXP4;
+        h ^= xx_rotl(v2 * XXP2, 31) * XXP1; h = h * XXP1 + XXP4;
+        h ^= xx_rotl(v3 * XXP2, 31) * XXP1; h = h * XXP1 + XXP4;
+        h ^= xx_rotl(v4 * XXP2, 31) * XXP1; h = h * XXP1 + XXP4;
+    } else {
+        h = XXP5;
+    }
+    h += (uint64_t)len;
+    while (p + 8 <= end) {
+        uint64_t k; __builtin_memcpy(&k, p, 8);
+        h ^= xx_rotl(k * XXP2, 31) * XXP1;
+        h = xx_rotl(h, 27) * XXP1 + XXP4; p += 8;
+    }
+    if (p + 4 <= end) {
+        uint32_t k; __builtin_memcpy(&k, p, 4);
+        h ^= (uint64_t)k * XXP1;
+        h = xx_rotl(h, 23) * XXP2 + XXP3; p += 4;
+    }
+    while (p < end) { h ^= (*p) * XXP5; h = xx_rotl(h, 11) * XXP1; p++; }
+    h ^= h >> 33; h *= XXP2; h ^= h >> 29; h *= XXP3; h ^= h >> 32;
+    return h;
+}
+
+__device__ static inline int32_t slice_begin(const DevColView &c, int64_t i) {
+    return i > 0 ? c.offsets[i - 1] : 0;
+}
+
 __device__ static inline bool col_is_null(const DevColView &c, int64_t i) {
     return c.has_nulls && c.nulls[i];
 }
@@ -60,6 +117,10 @@ __device__ static inline int32_t col_hash(const DevColView &c, int64_t i) {
     case GX_I64: return gx_hash_i64(((const int64_t *)c.values)[i]);
     case GX_I32: return ((const int32_t *)c.values)[i];
     case GX_F64: return gx_hash_f64(((const double *)c.values)[i]);
+    case GX_SLICE: {
+        int32_t b = slice_begin(c, i), e = c.offsets[i];
+        return (int32_t)xxhash64_dev(c.bytes + b, e - b);
+    }
     }
     return 0;
 }
@@ -73,6 +134,14 @@ __device__ static inline bool col_eq(const DevColView &a, int64_t i,
     case GX_I64: return ((const int64_t *)a.values)[i] == ((const int64_t *)b.values)[j];
     case GX_I32: return ((const int32_t *)a.values)[i] == ((const int32_t *)b.values)[j];
     case GX_F64: return ((const double *)a.values)[i] == ((const double *)b.values)[j];
+    case GX_SLICE: {
+        int32_t ab = slice_begin(a, i), ae = a.offsets[i];
+        int32_t bb = slice_begin(b, j), be = b.offsets[j];
+        if (ae - ab != be - bb) return false;
+        for (int32_t k = 0; k < ae - ab; k++)
+            if (a.bytes[ab + k] != b.bytes[bb + k]) return false;
+        return true;
+    }
     }
     return false;
 }
@@ -402,6 +471,38 @@ __global__ void k_gather(DevColView src, const uint32_t *idx, int64_t n,
     }
 }
 
+/* ---- SLICE (varlen) gather: lens -> inclusive scan -> byte copy ------- */
+
+__global__ void k_slice_lens(DevColView src, const uint32_t *idx,
+                             int fill_null_only, int64_t n, int32_t *lens,
+                             uint8_t *out_nulls) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t s = fill_null_only ? 0xFFFFFFFFu : (idx ? idx[i] : (uint32_t)i);
+        if (s == 0xFFFFFFFFu || col_is_null(src, s)) {
+            out_nulls[i] = 1;
+            lens[i] = 0;
+        } else {
+            out_nulls[i] = 0;
+            lens[i] = src.offsets[s] - slice_begin(src, s);
+        }
+    }
+}
+
+__global__ void k_slice_copy(DevColView src, const uint32_t *idx,
+                             int fill_null_only, int64_t n,
+                             const int32_t *out_off, uint8_t *out_bytes) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t s = fill_null_only ? 0xFFFFFFFFu : (idx ? idx[i] : (uint32_t)i);
+        if (s == 0xFFFFFFFFu || col_is_null(src, s)) continue;
+        int32_t sb = slice_begin(src, s), se = src.offsets[s];
+        int32_t db = i > 0 ? out_off[i - 1] : 0;
+        for (int32_t k = 0; k < se - sb; k++)
+            out_bytes[db + k] = src.bytes[sb + k];
+    }
+}
+
 /* ======================= result holder ================================= */
 
 struct HipResult {
@@ -448,7 +549,8 @@ int ensure_device(int device) {
 }
 
 /* Build a gx_result from exact-size device buffers. Each column owns a
- * values buf + nulls buf. */
+ * values buf + nulls buf; SLICE columns use slot 0 for END-OFFSETS and a
+ * data buf appended later by attach_slice_data (lengths known post-scan). */
 HipResult *alloc_result_cols(const std::vector<int32_t> &types, int64_t n,
                              int device, hipStream_t stream) {
     auto *h = new HipResult();
@@ -456,8 +558,7 @@ HipResult *alloc_result_cols(const std::vector<int32_t> &types, int64_t n,
     h->blocks.resize(types.size());
     h->bufs.resize(types.size() * 2);
     for (size_t c = 0; c < types.size(); c++) {
-        size_t es = (types[c] == GX_I32) ? 4 : 8;
-        if (types[c] == GX_SLICE) { gx_set_err("SLICE output not yet on device path"); free_result(h); return nullptr; }
+        size_t es = (types[c] == GX_I32 || types[c] == GX_SLICE) ? 4 : 8;
         if (n > 0) {
             if (h->bufs[c * 2].grow((size_t)n * es, stream) ||
                 h->bufs[c * 2 + 1].grow((size_t)n, stream)) {
@@ -469,7 +570,10 @@ HipResult *alloc_result_cols(const std::vector<int32_t> &types, int64_t n,
         std::memset(&b, 0, sizeof(b));
         b.type = types[c];
         b.mem = GX_MEM_DEVICE;
-        b.values = h->bufs[c * 2].p;
+        if (types[c] == GX_SLICE)
+            b.offsets = (const int32_t *)h->bufs[c * 2].p;
+        else
+            b.values = h->bufs[c * 2].p;
         b.nulls = (const uint8_t *)h->bufs[c * 2 + 1].p;
     }
     h->res.chunk.n_rows = (int32_t)n;
@@ -477,6 +581,45 @@ HipResult *alloc_result_cols(const std::vector<int32_t> &types, int64_t n,
     h->res.chunk.blocks = h->blocks.data();
     h->res.opaque = h;
     return h;
+}
+
+/* allocate and register the byte payload for a SLICE result column */
+int attach_slice_data(HipResult *h, size_t col, int64_t data_len,
+                      hipStream_t stream) {
+    h->bufs.emplace_back();
+    if (data_len > 0 && h->bufs.back().grow((size_t)data_len, stream)) return -1;
+    h->blocks[col].data = (const uint8_t *)h->bufs.back().p;
+    h->blocks[col].data_len = data_len;
+    return 0;
+}
+
+/* gather one SLICE column into result column `col`: lens + nulls, scan to
+ * end-offsets (in place), attach data, copy bytes. idx==nullptr -> identity;
+ * fill_null_only -> all-null column. */
+int gather_slice_col(const DevColView &src, const uint32_t *idx,
+                     int fill_null_only, int64_t n, HipResult *h, size_t col,
+                     DevBuf &scan_tmp, hipStream_t stream) {
+    if (n == 0) return attach_slice_data(h, col, 0, stream);
+    int32_t *off = (int32_t *)h->bufs[col * 2].p;
+    uint8_t *nulls = (uint8_t *)h->bufs[col * 2 + 1].p;
+    hipLaunchKernelGGL(k_slice_lens, dim3(gx_grid(n)), dim3(256), 0, stream,
+                       src, idx, fill_null_only, n, off, nulls);
+    size_t tmp_bytes = 0;
+    HIP_OK(hipcub::DeviceScan::InclusiveSum(nullptr, tmp_bytes, off, off, n,
+                                            stream));
+    if (scan_tmp.grow(tmp_bytes, stream)) return -1;
+    HIP_OK(hipcub::DeviceScan::InclusiveSum(scan_tmp.p, tmp_bytes, off, off,
+                                            n, stream));
+    int32_t total = 0;
+    HIP_OK(hipMemcpyAsync(&total, off + (n - 1), 4, hipMemcpyDeviceToHost,
+                          stream));
+    HIP_OK(hipStreamSynchronize(stream));
+    if (attach_slice_data(h, col, total, stream)) return -1;
+    if (total > 0)
+        hipLaunchKernelGGL(k_slice_copy, dim3(gx_grid(n)), dim3(256), 0,
+                           stream, src, idx, fill_null_only, n, off,
+                           (uint8_t *)h->blocks[col].data);
+    return 0;
 }
 
 /* stage an input chunk: if every block is device-resident we use the
@@ -494,7 +637,39 @@ struct StagedChunk {
             DevColView &v = views[c];
             v.type = b->type;
             v.offsets = nullptr; v.bytes = nullptr;
-            if (b->type == GX_SLICE) { gx_set_err("SLICE input not yet on device path"); return -1; }
+            if (b->type == GX_SLICE) {
+                v.has_nulls = b->nulls != nullptr;
+                if (b->mem == GX_MEM_DEVICE) {
+                    v.offsets = b->offsets;
+                    v.bytes = b->data;
+                    v.nulls = (const uint8_t *)b->nulls;
+                } else {
+                    int64_t blen = n_rows > 0 ? b->offsets[n_rows - 1] : 0;
+                    owned.emplace_back();
+                    DevBuf &ob = owned.back();
+                    if (ob.grow((size_t)std::max<int64_t>(n_rows, 1) * 4, stream)) return -1;
+                    HIP_OK(hipMemcpyAsync(ob.p, b->offsets, (size_t)n_rows * 4,
+                                          hipMemcpyHostToDevice, stream));
+                    v.offsets = (const int32_t *)ob.p;
+                    owned.emplace_back();
+                    DevBuf &db = owned.back();
+                    if (blen > 0) {
+                        if (db.grow((size_t)blen, stream)) return -1;
+                        HIP_OK(hipMemcpyAsync(db.p, b->data, (size_t)blen,
+                                              hipMemcpyHostToDevice, stream));
+                    }
+                    v.bytes = (const uint8_t *)db.p;
+                    if (b->nulls) {
+                        owned.emplace_back();
+                        DevBuf &nb = owned.back();
+                        if (nb.grow((size_t)n_rows, stream)) return -1;
+                        HIP_OK(hipMemcpyAsync(nb.p, b->nulls, (size_t)n_rows,
+                                              hipMemcpyHostToDevice, stream));
+                        v.nulls = (const uint8_t *)nb.p;
+                    }
+                }
+                continue;
+            }
             size_t es = (b->type == GX_I32) ? 4 : 8;
             if (b->mem == GX_MEM_DEVICE) {
                 v.values = b->values;
@@ -712,11 +887,25 @@ struct JoinOp : gx_op {
         G.bidx = d_bpos;
         G.n = n_out;
         int col = 0;
+        int n_fused = 0;
+        struct SliceJob { DevColView src; const uint32_t *idx; int null_fill; size_t col; };
+        std::vector<SliceJob> slice_jobs;
         auto add_col = [&](bool from_build, int src_col) {
-            G.side[col] = from_build ? (d_bpos ? 1 : 2) : (d_pidx ? 0 : 2);
-            G.src[col] = from_build ? build.view(src_col) : probe.views[src_col];
-            G.out_vals[col] = h->bufs[col * 2].p;
-            G.out_nulls[col] = (uint8_t *)h->bufs[col * 2 + 1].p;
+            DevColView src = from_build ? build.view(src_col)
+                                        : probe.views[src_col];
+            const uint32_t *idx = from_build ? d_bpos : d_pidx;
+            if (src.type == GX_SLICE) {
+                slice_jobs.push_back({src, idx, idx == nullptr ? 1 : 0,
+                                      (size_t)col});
+                col++;
+                return;
+            }
+            /* compact fused slots but keep out buffers per logical col */
+            G.src[n_fused] = src;
+            G.side[n_fused] = from_build ? (d_bpos ? 1 : 2) : (d_pidx ? 0 : 2);
+            G.out_vals[n_fused] = h->bufs[col * 2].p;
+            G.out_nulls[n_fused] = (uint8_t *)h->bufs[col * 2 + 1].p;
+            n_fused++;
             col++;
         };
         bool semi = cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI;
@@ -732,9 +921,18 @@ struct JoinOp : gx_op {
             for (size_t i = 0; i < inner_types.size(); i++) add_col(!outer_from_build, (int)i);
             for (size_t i = 0; i < n_outer; i++) add_col(outer_from_build, (int)i);
         }
-        G.n_cols = col;
-        hipLaunchKernelGGL(k_gather_multi, dim3(gx_grid(n_out)), dim3(256), 0,
-                           stream, G);
+        G.n_cols = n_fused;
+        if (n_fused > 0)
+            hipLaunchKernelGGL(k_gather_multi, dim3(gx_grid(n_out)), dim3(256),
+                               0, stream, G);
+        for (auto &sj : slice_jobs) {
+            if (gather_slice_col(sj.src, sj.null_fill ? nullptr : sj.idx,
+                                 sj.null_fill, n_out, h, sj.col, d_scan_tmp,
+                                 stream)) {
+                free_result(h);
+                return nullptr;
+            }
+        }
         return h;
     }
 
@@ -874,10 +1072,19 @@ struct JoinOp : gx_op {
         HipResult *h = alloc_result_cols(otypes, cnt, device, stream);
         if (!h) { d_idx.release(); d_cnt.release(); return -1; }
         size_t col = 0;
+        int rc_col = 0;
         auto emit_col = [&](bool from_build, int src_col, bool null_fill) {
             DevColView src = from_build ? build.view(src_col)
                                         : build.view(0) /* placeholder for null fill */;
             if (null_fill) src.type = otypes[col];
+            if (otypes[col] == GX_SLICE) {
+                if (gather_slice_col(src, (const uint32_t *)d_idx.p,
+                                     (int)null_fill, cnt, h, col, d_scan_tmp,
+                                     stream))
+                    rc_col = -1;
+                col++;
+                return;
+            }
             hipLaunchKernelGGL(k_gather, dim3(gx_grid(cnt)), dim3(256), 0, stream,
                                src, (const uint32_t *)d_idx.p, (int64_t)cnt,
                                h->bufs[col * 2].p, (uint8_t *)h->bufs[col * 2 + 1].p,
@@ -894,6 +1101,7 @@ struct JoinOp : gx_op {
         }
         HIP_OK(hipStreamSynchronize(stream));
         d_idx.release(); d_cnt.release();
+        if (rc_col != 0) { free_result(h); return -1; }
         *out = &h->res;
         return 0;
     }
@@ -949,12 +1157,25 @@ int gxop_result_to_host(gx_result *res) {
     for (size_t c = 0; c < h->blocks.size(); c++) {
         gx_block &b = h->blocks[c];
         if (b.mem != GX_MEM_DEVICE) continue;
-        size_t es = (b.type == GX_I32) ? 4 : 8;
-        h->host.emplace_back((size_t)n * es);
-        if (n > 0)
-            HIP_OK(hipMemcpy(h->host.back().data(), b.values, (size_t)n * es,
-                             hipMemcpyDeviceToHost));
-        b.values = h->host.back().data();
+        if (b.type == GX_SLICE) {
+            h->host.emplace_back((size_t)n * 4);
+            if (n > 0)
+                HIP_OK(hipMemcpy(h->host.back().data(), (const void *)b.offsets,
+                                 (size_t)n * 4, hipMemcpyDeviceToHost));
+            b.offsets = (const int32_t *)h->host.back().data();
+            h->host.emplace_back((size_t)(b.data_len > 0 ? b.data_len : 1));
+            if (b.data_len > 0)
+                HIP_OK(hipMemcpy(h->host.back().data(), (const void *)b.data,
+                                 (size_t)b.data_len, hipMemcpyDeviceToHost));
+            b.data = (const uint8_t *)h->host.back().data();
+        } else {
+            size_t es = (b.type == GX_I32) ? 4 : 8;
+            h->host.emplace_back((size_t)n * es);
+            if (n > 0)
+                HIP_OK(hipMemcpy(h->host.back().data(), b.values, (size_t)n * es,
+                                 hipMemcpyDeviceToHost));
+            b.values = h->host.back().data();
+        }
         h->host.emplace_back((size_t)n);
         if (n > 0)
             HIP_OK(hipMemcpy(h->host.back().data(), (void *)b.nulls, (size_t)n,
