@@ -265,3 +265,23 @@ def test_partition_routing_parity(libs):
                 assert multiset(rh) == multiset(ro), f"partition {p}/{n_parts}"
         exo.close()
         exh.close()
+
+
+def test_agg_small_chunk_stream(libs):
+    """CHUNK_SIZE-style streaming (the CN's real 1000-row push cadence,
+    ConnectionParams.java:1088): many small chunks take the inline-gid path,
+    then a big chunk switches to the compaction path — gids must stay
+    consistent across both."""
+    rng = np.random.default_rng(25)
+    n_small, n_big = 60_000, 400_000
+    gk_small = rng.integers(0, 5000, size=n_small, dtype=np.int64)
+    gk_big = rng.integers(0, 9000, size=n_big, dtype=np.int64)
+    v_small = rng.integers(-100, 100, n_small, dtype=np.int64)
+    v_big = rng.integers(-100, 100, n_big, dtype=np.int64)
+    chunks = chunks_from_columns([I64, I64], [(gk_small, None), (v_small, None)],
+                                 chunk_size=1000)
+    chunks += chunks_from_columns([I64, I64], [(gk_big, None), (v_big, None)],
+                                  chunk_size=n_big)
+    both_agg(libs, [0], [(abi.SUM_I64, 1), (abi.COUNT_ROW, -1),
+                         (abi.MIN_I64, 1)], [I64, I64], chunks,
+             expected_groups=64)
