@@ -120,7 +120,8 @@ class GxPartCfg(C.Structure):
 
 class GxPred(C.Structure):
     _fields_ = [("col", C.c_int32), ("cmp", C.c_int32),
-                ("v_i64", C.c_int64), ("v_f64", C.c_double)]
+                ("v_i64", C.c_int64), ("v_f64", C.c_double),
+                ("v_bytes", C.c_void_p), ("v_len", C.c_int32)]
 
 
 class GxProj(C.Structure):
@@ -140,6 +141,7 @@ class GxScanCfg(C.Structure):
 INNER, LEFT, RIGHT, SEMI, ANTI = 0, 1, 2, 3, 4
 # Comparisons (gx_cmp)
 LT, LE, GT, GE, EQ, NE = 0, 1, 2, 3, 4, 5
+CONTAINS = 6  # SLICE LIKE '%pat%'
 # Projections (gx_proj_op)
 PROJ_COPY, PROJ_REV_F64, PROJ_REV_SCALED4 = 0, 1, 2
 # Agg funcs (gx_agg_func)

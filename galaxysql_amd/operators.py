@@ -220,10 +220,18 @@ class ScanExec:
         self._lib = lib
         self._keep = []
         pa = (GxPred * max(1, len(preds)))()
+        self._pats = []
         for i, (col, cmp, const) in enumerate(preds):
-            pa[i] = GxPred(col, cmp,
-                           int(const) if not isinstance(const, float) else 0,
-                           float(const))
+            if isinstance(const, (str, bytes)):
+                pat = const.encode() if isinstance(const, str) else bytes(const)
+                buf = C.create_string_buffer(pat, len(pat))
+                self._pats.append(buf)
+                pa[i] = GxPred(col, cmp, 0, 0.0,
+                               C.cast(buf, C.c_void_p), len(pat))
+            else:
+                pa[i] = GxPred(col, cmp,
+                               int(const) if not isinstance(const, float) else 0,
+                               float(const), None, 0)
         pj = (GxProj * len(projs))()
         for i, (op, a, b) in enumerate(projs):
             pj[i] = GxProj(op, a, b)

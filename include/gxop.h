@@ -216,7 +216,8 @@ int gxop_part_close(gx_op *op);
  * Null comparison semantics are SQL: a NULL operand fails every predicate. */
 
 typedef enum gx_cmp {
-    GX_CMP_LT = 0, GX_CMP_LE, GX_CMP_GT, GX_CMP_GE, GX_CMP_EQ, GX_CMP_NE
+    GX_CMP_LT = 0, GX_CMP_LE, GX_CMP_GT, GX_CMP_GE, GX_CMP_EQ, GX_CMP_NE,
+    GX_CMP_CONTAINS = 6   /* SLICE columns: LIKE '%pattern%' byte scan */
 } gx_cmp;
 
 typedef struct gx_pred {
@@ -224,6 +225,8 @@ typedef struct gx_pred {
     int32_t cmp;       /* gx_cmp */
     int64_t v_i64;     /* constant for I64/I32 columns */
     double  v_f64;     /* constant for F64 columns */
+    const uint8_t *v_bytes; /* CONTAINS pattern (host memory) */
+    int32_t v_len;
 } gx_pred;
 
 typedef enum gx_proj_op {

@@ -929,8 +929,15 @@ struct ScanOp : gx_op {
     std::vector<gx_proj> projs;
     std::vector<int32_t> input_types, out_types;
 
+    std::vector<std::vector<uint8_t>> pat_store;
     ScanOp(const gx_scan_cfg *c) : gx_op(OP_SCAN), cfg(*c) {
         preds.assign(c->preds, c->preds + c->n_preds);
+        for (auto &p : preds) {
+            if (p.cmp == GX_CMP_CONTAINS && p.v_bytes && p.v_len > 0) {
+                pat_store.emplace_back(p.v_bytes, p.v_bytes + p.v_len);
+                p.v_bytes = pat_store.back().data();
+            }
+        }
         projs.assign(c->projs, c->projs + c->n_projs);
         input_types.assign(c->input_types, c->input_types + c->n_input_cols);
         for (auto &p : projs) {
@@ -980,6 +987,15 @@ struct ScanOp : gx_op {
                 switch (c.type) {
                 case GX_I64: ok = cmp_ok_i(c.i64v[r], p.cmp, p.v_i64); break;
                 case GX_I32: ok = cmp_ok_i((int64_t)c.i32v[r], p.cmp, p.v_i64); break;
+                case GX_SLICE: {
+                    /* LIKE '%pat%' naive byte scan over the slice */
+                    int32_t b = c.begin_off(r), e = c.off[r];
+                    int32_t len = e - b, pl = p.v_len;
+                    ok = false;
+                    for (int32_t s = 0; s + pl <= len && !ok; s++)
+                        ok = std::memcmp(c.bytes.data() + b + s, p.v_bytes,
+                                         (size_t)pl) == 0;
+                    break; }
                 default: ok = cmp_ok_f(c.f64v[r], p.cmp, p.v_f64); break;
                 }
                 if (!ok) { pass = false; break; }

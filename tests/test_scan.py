@@ -8,7 +8,7 @@ import numpy as np
 import pytest
 
 from galaxysql_amd import abi
-from galaxysql_amd.chunk import Chunk, Block, I64, I32, F64, chunks_from_columns, multiset
+from galaxysql_amd.chunk import Chunk, Block, I64, I32, F64, SLICE, chunks_from_columns, multiset
 from galaxysql_amd.operators import ScanExec
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
@@ -86,4 +86,61 @@ def test_scan_hip_vs_oracle():
     ref = run_scan(oracle, chunks, -1)
     got = run_scan(hip, chunks, 0)
     # per-row arithmetic: fully exact, including the doubles
+    assert multiset(got) == multiset(ref)
+
+
+# ---- LIKE '%pat%' on SLICE columns (the Q9/C5 p_name filter) ----
+
+def make_string_inputs(rng, n):
+    words = ["green", "blue", "lime", "forest", "salmon", "puff", "dim"]
+    vals = []
+    for _ in range(n):
+        if rng.random() < 0.03:
+            vals.append(None)
+        else:
+            k = rng.integers(1, 4)
+            vals.append(" ".join(words[int(i)]
+                                 for i in rng.integers(0, len(words), k)))
+    keys = rng.integers(0, 1 << 40, n).astype(np.int64)
+    return chunks_from_columns(
+        [I64, SLICE],
+        [(keys, None), Block.of(SLICE, vals)], chunk_size=911)
+
+
+def run_like_scan(lib, chunks, device):
+    op = ScanExec(lib, [(1, abi.CONTAINS, "green")],
+                  [(abi.PROJ_COPY, 0, -1), (abi.PROJ_COPY, 1, -1)],
+                  [I64, SLICE], device=device)
+    try:
+        rows = []
+        for ch in chunks:
+            out = op.consume_chunk(ch)
+            if out is not None:
+                rows.extend(out.rows())
+        return rows
+    finally:
+        op.close()
+
+
+def test_like_scan_oracle_vs_numpy(oracle):
+    rng = np.random.default_rng(95)
+    chunks = make_string_inputs(rng, 12000)
+    got = run_like_scan(oracle, chunks, -1)
+    exp = []
+    for ch in chunks:
+        for i in range(ch.n_rows):
+            s = ch.blocks[1].get(i)
+            if s is not None and b"green" in s:
+                exp.append((ch.blocks[0].get(i), s))
+    assert multiset(got) == multiset(exp)
+
+
+@pytest.mark.gpu
+def test_like_scan_hip_vs_oracle():
+    oracle = abi.load_oracle()
+    hip = abi.load_hip()
+    rng = np.random.default_rng(96)
+    chunks = make_string_inputs(rng, 300_000)
+    ref = run_like_scan(oracle, chunks, -1)
+    got = run_like_scan(hip, chunks, 0)
     assert multiset(got) == multiset(ref)
