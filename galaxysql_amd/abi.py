@@ -125,7 +125,8 @@ class GxPred(C.Structure):
 
 
 class GxProj(C.Structure):
-    _fields_ = [("op", C.c_int32), ("a", C.c_int32), ("b", C.c_int32)]
+    _fields_ = [("op", C.c_int32), ("a", C.c_int32), ("b", C.c_int32),
+                ("c", C.c_int32), ("d", C.c_int32)]
 
 
 class GxScanCfg(C.Structure):
@@ -143,7 +144,7 @@ INNER, LEFT, RIGHT, SEMI, ANTI = 0, 1, 2, 3, 4
 LT, LE, GT, GE, EQ, NE = 0, 1, 2, 3, 4, 5
 CONTAINS = 6  # SLICE LIKE '%pat%'
 # Projections (gx_proj_op)
-PROJ_COPY, PROJ_REV_F64, PROJ_REV_SCALED4 = 0, 1, 2
+PROJ_COPY, PROJ_REV_F64, PROJ_REV_SCALED4, PROJ_Q9_AMOUNT4 = 0, 1, 2, 3
 # Agg funcs (gx_agg_func)
 COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64, MAX_F64 = range(8)
 

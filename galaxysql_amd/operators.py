@@ -233,8 +233,9 @@ class ScanExec:
                                int(const) if not isinstance(const, float) else 0,
                                float(const), None, 0)
         pj = (GxProj * len(projs))()
-        for i, (op, a, b) in enumerate(projs):
-            pj[i] = GxProj(op, a, b)
+        for i, spec in enumerate(projs):
+            spec = tuple(spec) + (-1,) * (5 - len(spec))
+            pj[i] = GxProj(*spec[:5])
         it = (C.c_int32 * len(input_types))(*input_types)
         cfg = GxScanCfg(n_preds=len(preds), preds=pa, n_projs=len(projs),
                         projs=pj, n_input_cols=len(input_types),

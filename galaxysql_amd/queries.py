@@ -21,7 +21,7 @@ import ctypes as C
 
 from . import abi
 from .abi import GxResult
-from .chunk import I64, I32, F64
+from .chunk import I64, I32, F64, SLICE
 from .exchange import chunk_from_torch
 from .operators import ParallelHashJoinExec, HashAggExec, EquiJoinKey
 
@@ -438,3 +438,269 @@ def gen_q3_numpy(rng, n_cust_total, n_orders_total, n_lineitem, cust_sel=0.2,
     revenue = cents.astype(np.float64) / 100.0
     return ([cust_keys], [o_custkey, o_orderkey, o_date, o_prio],
             [l_orderkey, revenue, cents])
+
+
+# ---- Q9: 6-way join chain with LIKE + DECIMAL sum ----
+
+Q9_PART_TYPES = [I64, SLICE]             # p_partkey, p_name
+Q9_SUPP_TYPES = [I64, I64]               # s_suppkey, s_nationkey
+Q9_PARTSUPP_TYPES = [I64, I64, I64]      # ps_partkey, ps_suppkey, ps_supplycost_cents
+Q9_ORDERS_TYPES = [I64, I32]             # o_orderkey, o_year
+Q9_LINEITEM_TYPES = [I64, I64, I64, I64, I64, I64]
+# l_partkey, l_suppkey, l_orderkey, l_quantity, l_extendedprice_cents, l_discount_hundredths
+
+
+def _scan_raw(lib, op, tensors, types):
+    ka = []
+    return op.consume_raw(C.byref(chunk_from_torch(lib, list(tensors), types,
+                                                   ka)))
+
+
+def stage_table(lib, chunk, types, device):
+    """Upload a host Chunk (any column types incl. SLICE) into a
+    device-resident gx_result once, outside the timed region — a
+    1-partition pass through the exchange kernels. Caller releases."""
+    from .operators import PartitioningExchanger
+    ex = PartitioningExchanger(lib, 1, [0], types, device=device)
+    try:
+        ka = []
+        gc = lib.to_gx_chunk(chunk, ka)
+        out = C.POINTER(GxResult)()
+        counts = (C.c_int64 * 1)()
+        lib.check(lib.lib.gxop_part_consume_concat(ex._op, C.byref(gc),
+                                                   C.byref(out), counts),
+                  "stage_table")
+        return out
+    finally:
+        ex.close()
+
+
+def run_q9(lib, device, part_res, supplier, partsupp, orders, lineitem,
+           pattern=b"green", world=1, local_rank=0):
+    """Q9 chain: part LIKE-filter -> SEMI partsupp -> 2-key INNER lineitem ->
+    amount projection (exact scale-4 DECIMAL: extprice*(1-disc) -
+    supplycost*qty) -> INNER orders -> INNER supplier -> GROUP BY
+    (nationkey, o_year) SUM(amount). At world>1 each stage's inputs are
+    hash-shuffled on its join key (partkey / orderkey / suppkey) and the
+    aggregate is TWO-PHASE (partial, exchange on group hash, final) — the
+    reference's MppHashAggConvertRule split.
+
+    Returns (rows, info); rows = final (nationkey, o_year, sum4, count)."""
+    import torch
+    from .operators import ScanExec
+    from .exchange import shuffle_columns
+    dev_t = supplier[0].device
+    dist = world > 1
+
+    if dist:
+        partsupp = shuffle_columns(lib, partsupp, Q9_PARTSUPP_TYPES, [0],
+                                   device=device)
+        lineitem = shuffle_columns(lib, lineitem, Q9_LINEITEM_TYPES, [0],
+                                   device=device)
+
+    # 1. part scan: p_name LIKE '%pattern%' -> partkey list (filter runs
+    # BEFORE the exchange, as the reference pushes filters below shuffles;
+    # only the surviving i64 keys travel)
+    sc = ScanExec(lib, [(1, abi.CONTAINS, pattern)], [(abi.PROJ_COPY, 0)],
+                  Q9_PART_TYPES, device=device)
+    try:
+        r_part = sc.consume_raw(C.byref(part_res.contents.chunk))
+        n_part = r_part.contents.chunk.n_rows if r_part else 0
+    finally:
+        sc.close()
+
+    part_src = r_part
+    if dist:
+        cols = result_to_tensors(lib, r_part, [I64], dev_t)
+        lib.lib.gxop_result_release(r_part)
+        part_src = shuffle_columns(lib, cols, [I64], [0], device=device)
+
+    # 2. SEMI: partsupp rows whose partkey passed
+    ja = ParallelHashJoinExec(lib, abi.SEMI, [EquiJoinKey(0, 0, I64)],
+                              outer_types=Q9_PARTSUPP_TYPES,
+                              inner_types=[I64], device=device,
+                              expected_build_rows=max(1, n_part))
+    try:
+        _consume_src(lib, ja, part_src, [I64])
+        if not isinstance(part_src, list):
+            lib.lib.gxop_result_release(part_src)
+        ja.build_consume()
+        r_ps = _probe_tensors(lib, ja, list(partsupp), Q9_PARTSUPP_TYPES)
+        n_ps = r_ps.contents.chunk.n_rows if r_ps else 0
+    finally:
+        ja.close()
+
+    # 3. 2-key INNER: lineitem x partsupp on (partkey, suppkey)
+    jb = ParallelHashJoinExec(lib, abi.INNER,
+                              [EquiJoinKey(0, 0, I64), EquiJoinKey(1, 1, I64)],
+                              outer_types=Q9_LINEITEM_TYPES,
+                              inner_types=Q9_PARTSUPP_TYPES, device=device,
+                              expected_build_rows=max(1, n_ps))
+    try:
+        lib.check(lib.lib.gxop_join_consume(jb._op,
+                                            C.byref(r_ps.contents.chunk)),
+                  "join_consume")
+        jb.build_consume()
+        lib.lib.gxop_result_release(r_ps)
+        r_li = _probe_tensors(lib, jb, list(lineitem), Q9_LINEITEM_TYPES)
+        n_li = r_li.contents.chunk.n_rows if r_li else 0
+        jb_stats = jb.stats()
+    finally:
+        jb.close()
+
+    # 4. projection: amount4 + carry (l_orderkey, l_suppkey)
+    # r_li cols: [l_partkey,l_suppkey,l_orderkey,qty,extprice,disc,
+    #             ps_partkey,ps_suppkey,ps_cost]
+    t9 = Q9_LINEITEM_TYPES + Q9_PARTSUPP_TYPES
+    sp = ScanExec(lib, [], [(abi.PROJ_COPY, 2), (abi.PROJ_COPY, 1),
+                            (abi.PROJ_Q9_AMOUNT4, 4, 5, 8, 3)],
+                  t9, device=device)
+    try:
+        out = C.POINTER(GxResult)()
+        lib.check(lib.lib.gxop_scan_consume(sp._op,
+                                            C.byref(r_li.contents.chunk),
+                                            C.byref(out)), "scan_consume")
+        lib.lib.gxop_result_release(r_li)
+        r_amt = out  # [l_orderkey, l_suppkey, amount4]
+    finally:
+        sp.close()
+
+    amt_types = [I64, I64, I64]
+    if dist:
+        cols = result_to_tensors(lib, r_amt, amt_types, dev_t)
+        lib.lib.gxop_result_release(r_amt)
+        cols = shuffle_columns(lib, cols, amt_types, [0], device=device)
+        orders = shuffle_columns(lib, orders, Q9_ORDERS_TYPES, [0],
+                                 device=device)
+        r_amt_src = cols
+    else:
+        r_amt_src = r_amt
+
+    # 5. INNER orders on orderkey (build = the amount rows, probe = orders)
+    jc = ParallelHashJoinExec(lib, abi.INNER, [EquiJoinKey(0, 0, I64)],
+                              outer_types=Q9_ORDERS_TYPES,
+                              inner_types=amt_types, device=device,
+                              expected_build_rows=max(1, n_li))
+    try:
+        _consume_src(lib, jc, r_amt_src, amt_types)
+        if not isinstance(r_amt_src, list):
+            lib.lib.gxop_result_release(r_amt_src)
+        jc.build_consume()
+        r_ord = _probe_tensors(lib, jc, list(orders), Q9_ORDERS_TYPES)
+        n_ord = r_ord.contents.chunk.n_rows if r_ord else 0
+    finally:
+        jc.close()
+
+    # r_ord cols: [o_orderkey, o_year, l_orderkey, l_suppkey, amount4]
+    ord_types = Q9_ORDERS_TYPES + amt_types
+    if dist:
+        cols = result_to_tensors(lib, r_ord, ord_types, dev_t)
+        lib.lib.gxop_result_release(r_ord)
+        cols = shuffle_columns(lib, cols, ord_types, [3], device=device)
+        supplier = shuffle_columns(lib, supplier, Q9_SUPP_TYPES, [0],
+                                   device=device)
+        r_ord_src = cols
+    else:
+        r_ord_src = r_ord
+
+    # 6. INNER supplier on suppkey (build = supplier, probe = result)
+    jd = ParallelHashJoinExec(lib, abi.INNER, [EquiJoinKey(3, 0, I64)],
+                              outer_types=ord_types,
+                              inner_types=Q9_SUPP_TYPES, device=device,
+                              expected_build_rows=supplier[0].numel())
+    try:
+        _consume_tensors(lib, jd, list(supplier), Q9_SUPP_TYPES)
+        jd.build_consume()
+        r_fin = _probe_src(lib, jd, r_ord_src, ord_types)
+        if not isinstance(r_ord_src, list):
+            lib.lib.gxop_result_release(r_ord_src)
+        n_fin = r_fin.contents.chunk.n_rows if r_fin else 0
+    finally:
+        jd.close()
+
+    # 7. GROUP BY (s_nationkey, o_year): cols of r_fin =
+    #    [o_orderkey,o_year,l_orderkey,l_suppkey,amount4,s_suppkey,s_nationkey]
+    fin_types = ord_types + Q9_SUPP_TYPES
+    agg = HashAggExec(lib, group_cols=[6, 1],
+                      aggs=[(abi.SUM_I64, 4), (abi.COUNT_ROW, -1)],
+                      input_types=fin_types, expected_groups=4096,
+                      device=device)
+    try:
+        lib.check(lib.lib.gxop_agg_consume(agg._op,
+                                           C.byref(r_fin.contents.chunk)),
+                  "agg_consume")
+        agg.build_consume()
+        lib.lib.gxop_result_release(r_fin)
+        parts_out = agg.result_chunks()
+    finally:
+        agg.close()
+
+    rows = []
+    for c in parts_out:
+        rows.extend(c.rows())
+
+    if dist:
+        # two-phase agg: exchange partial rows on the group-key hash, then a
+        # final SUM of partial sums/counts
+        pk = [torch.tensor([r[0] for r in rows], dtype=torch.int64,
+                           device=dev_t),
+              torch.tensor([r[1] for r in rows], dtype=torch.int32,
+                           device=dev_t),
+              torch.tensor([r[2] for r in rows], dtype=torch.int64,
+                           device=dev_t),
+              torch.tensor([r[3] for r in rows], dtype=torch.int64,
+                           device=dev_t)]
+        ptypes = [I64, I32, I64, I64]
+        cols = shuffle_columns(lib, pk, ptypes, [0, 1], device=device)
+        fagg = HashAggExec(lib, group_cols=[0, 1],
+                           aggs=[(abi.SUM_I64, 2), (abi.SUM_I64, 3)],
+                           input_types=ptypes, expected_groups=4096,
+                           device=device)
+        try:
+            ka = []
+            gc2 = chunk_from_torch(lib, cols, ptypes, ka)
+            lib.check(lib.lib.gxop_agg_consume(fagg._op, C.byref(gc2)),
+                      "agg_consume")
+            fagg.build_consume()
+            rows = []
+            for c in fagg.result_chunks():
+                rows.extend(c.rows())
+        finally:
+            fagg.close()
+
+    info = {"part_kept": n_part, "partsupp_kept": n_ps, "lineitem_joined": n_li,
+            "after_orders": n_ord, "final_rows": n_fin, "groups": len(rows),
+            "join2_stats": jb_stats}
+    return rows, info
+
+
+def gen_q9_numpy(rng, n_part, n_supp, n_orders, n_lineitem, n_nation=25):
+    import numpy as np
+    words = ["green", "blue", "lime", "forest", "salmon", "navy", "puff"]
+
+    def name():
+        return " ".join(words[i] for i in rng.integers(0, len(words), 3))
+
+    part = [np.arange(n_part, dtype=np.int64),
+            [name() for _ in range(n_part)]]
+    supplier = [np.arange(n_supp, dtype=np.int64),
+                rng.integers(0, n_nation, n_supp).astype(np.int64)]
+    # partsupp: 4 DISTINCT suppliers per part (TPC-H shape: the
+    # (partkey, suppkey) pair is the table's primary key)
+    ps_part = np.repeat(np.arange(n_part, dtype=np.int64), 4)
+    first = rng.integers(0, n_supp, n_part)
+    ps_supp = ((np.repeat(first, 4) +
+                np.tile(np.arange(4), n_part) * max(1, n_supp // 5))
+               % n_supp).astype(np.int64)
+    ps_cost = rng.integers(100, 100_000, ps_part.shape[0]).astype(np.int64)
+    partsupp = [ps_part, ps_supp, ps_cost]
+    orders = [4 * rng.permutation(n_orders).astype(np.int64),
+              rng.integers(1992, 1999, n_orders).astype(np.int32)]
+    # lineitem rows reference REAL (partkey, suppkey) partsupp pairs
+    pick = rng.integers(0, ps_part.shape[0], n_lineitem)
+    lineitem = [ps_part[pick], ps_supp[pick],
+                orders[0][rng.integers(0, n_orders, n_lineitem)],
+                rng.integers(1, 51, n_lineitem).astype(np.int64),
+                rng.integers(100, 10_000_000, n_lineitem).astype(np.int64),
+                rng.integers(0, 11, n_lineitem).astype(np.int64)]
+    return part, supplier, partsupp, orders, lineitem

@@ -232,13 +232,17 @@ typedef struct gx_pred {
 typedef enum gx_proj_op {
     GX_PROJ_COPY = 0,      /* out = col a */
     GX_PROJ_REV_F64 = 1,   /* out = a * (1 - b), doubles (Q3/Q9 revenue) */
-    GX_PROJ_REV_SCALED4 = 2/* out = a * (100 - b), i64 cents x hundredths ->
+    GX_PROJ_REV_SCALED4 = 2,/* out = a * (100 - b), i64 cents x hundredths ->
                               DECIMAL scale-4, exact */
+    GX_PROJ_Q9_AMOUNT4 = 3 /* out = a*(100-b) - c*d*100: Q9 amount =
+                              extprice*(1-disc) - supplycost*qty, all i64
+                              cents/hundredths -> DECIMAL scale-4, exact */
 } gx_proj_op;
 
 typedef struct gx_proj {
     int32_t op;        /* gx_proj_op */
     int32_t a, b;      /* input columns */
+    int32_t c, d;      /* extra inputs (Q9_AMOUNT4) */
 } gx_proj;
 
 typedef struct gx_scan_cfg {

@@ -1014,11 +1014,23 @@ struct ScanOp : gx_op {
                     if (a.is_null(r) || b.is_null(r)) cols[c].append_null();
                     else cols[c].append_f64(a.f64v[r] * (1.0 - b.f64v[r]));
                     break; }
-                default: { /* REV_SCALED4 */
+                case GX_PROJ_REV_SCALED4: {
                     const Column &b = in.cols[pj.b];
                     if (a.is_null(r) || b.is_null(r)) cols[c].append_null();
                     else cols[c].append_i64((int64_t)((uint64_t)a.i64v[r] *
                                             (uint64_t)(100 - b.i64v[r])));
+                    break; }
+                default: { /* Q9_AMOUNT4 */
+                    const Column &b = in.cols[pj.b];
+                    const Column &cc = in.cols[pj.c];
+                    const Column &dd = in.cols[pj.d];
+                    if (a.is_null(r) || b.is_null(r) || cc.is_null(r) ||
+                        dd.is_null(r))
+                        cols[c].append_null();
+                    else
+                        cols[c].append_i64(
+                            (int64_t)((uint64_t)a.i64v[r] * (uint64_t)(100 - b.i64v[r]))
+                            - (int64_t)((uint64_t)cc.i64v[r] * (uint64_t)dd.i64v[r] * 100u));
                     break; }
                 }
             }
