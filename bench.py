@@ -440,7 +440,152 @@ def c4_cpu_baseline(sample_scale, seed=103):
                       "1 thread, -O3 -march=native"}
 
 
-WORKLOADS = {"c2": C2, "c3": C3, "c4": C4}
+# --------------------------------------------------------------------------
+# C5: SF300 Q9-shaped 6-way join chain, sharded 8-ways (per-GPU = SF37.5)
+# --------------------------------------------------------------------------
+
+class C5:
+    name = "c5"
+    # per-GPU shard of SF300/8 (weak scaling; BASELINE.md C5 = 8 GPUs)
+    PART = 7_500_000
+    SUPP = 375_000
+    ORDERS = 56_250_000
+    LINEITEM = 225_000_000
+    N_NATION = 25
+
+    def __init__(self, args, device, rank):
+        import torch
+        from galaxysql_amd.chunk import Block, Chunk, I64, SLICE
+        from galaxysql_amd import abi as _abi
+        from galaxysql_amd.queries import stage_table, Q9_PART_TYPES
+        scale = args.c5_scale
+        g = torch.Generator(device=device)
+        g.manual_seed(6789 + rank)
+        npart = int(self.PART * scale)
+        nsupp = int(self.SUPP * scale)
+        nord = int(self.ORDERS * scale)
+        nli = int(self.LINEITEM * scale)
+
+        # p_name: one 8-char word from a 92-word vocab, 5 of which contain
+        # "green" -> ~5.4 % LIKE selectivity
+        rng = np.random.default_rng(6789 + rank)
+        vocab = [f"w{i:02d}filler"[:8] for i in range(87)] + \
+                ["green00", "greenish", "00green0", "agreenxx", "greeny00"]
+        vocab = [w.ljust(8, "x")[:8] for w in vocab]
+        vb = np.frombuffer("".join(vocab).encode(), dtype=np.uint8).reshape(-1, 8)
+        idx = rng.integers(0, len(vocab), npart)
+        data = vb[idx].reshape(-1)
+        offsets = (np.arange(1, npart + 1, dtype=np.int32) * 8)
+        pkeys = np.arange(npart, dtype=np.int64)
+        part_chunk = Chunk([Block(I64, values=pkeys),
+                            Block(SLICE, offsets=offsets, data=data)])
+        self.lib = None  # staged lazily (needs the lib handle)
+        self._part_chunk = part_chunk
+        self.part_res = None
+        self.device = device
+
+        self.supplier = [torch.arange(nsupp, dtype=torch.int64, device=device),
+                         torch.randint(0, self.N_NATION, (nsupp,), generator=g,
+                                       device=device, dtype=torch.int64)]
+        ps_part = torch.repeat_interleave(
+            torch.arange(npart, dtype=torch.int64, device=device), 4)
+        first = torch.randint(0, nsupp, (npart,), generator=g, device=device,
+                              dtype=torch.int64)
+        step4 = torch.arange(4, device=device, dtype=torch.int64) * max(1, nsupp // 5)
+        ps_supp = ((first.unsqueeze(1) + step4.unsqueeze(0)).reshape(-1)) % nsupp
+        ps_cost = torch.randint(100, 100_000, (ps_part.numel(),), generator=g,
+                                device=device, dtype=torch.int64)
+        self.partsupp = [ps_part.contiguous(), ps_supp.contiguous(), ps_cost]
+
+        okeys = 4 * torch.randperm(nord, generator=g, device=device,
+                                   dtype=torch.int64)
+        self.orders = [okeys,
+                       torch.randint(1992, 1999, (nord,), generator=g,
+                                     device=device, dtype=torch.int32)]
+        pick = torch.randint(0, ps_part.numel(), (nli,), generator=g,
+                             device=device)
+        self.lineitem = [ps_part[pick].contiguous(), ps_supp[pick].contiguous(),
+                         okeys[torch.randint(0, nord, (nli,), generator=g,
+                                             device=device)].contiguous(),
+                         torch.randint(1, 51, (nli,), generator=g,
+                                       device=device, dtype=torch.int64),
+                         torch.randint(100, 10_000_000, (nli,), generator=g,
+                                       device=device, dtype=torch.int64),
+                         torch.randint(0, 11, (nli,), generator=g,
+                                       device=device, dtype=torch.int64)]
+        self.n_probe = nli
+        self.n_build = ps_part.numel()
+        self.last_info = None
+
+    def step(self, lib, local_rank, distributed):
+        from galaxysql_amd.queries import run_q9, stage_table, Q9_PART_TYPES
+        if self.part_res is None:
+            # one-time device upload of the part table (strings), untimed
+            # relative to steady state (first call is warmup)
+            self.part_res = stage_table(lib, self._part_chunk, Q9_PART_TYPES,
+                                        local_rank)
+        world = int(os.environ.get("WORLD_SIZE", "1"))
+        rows, info = run_q9(lib, local_rank, self.part_res, self.supplier,
+                            self.partsupp, self.orders, self.lineitem,
+                            world=world, local_rank=local_rank)
+        self.last_info = info
+        st = dict(info["join2_stats"])
+        st["n_build"] = info["partsupp_kept"]
+        st["groups"] = info["groups"]
+        return st
+
+    def config(self, world):
+        li = self.last_info or {}
+        return {"workload": "C5_tpch_sf300_q9_6way_join_like_decimal",
+                "part_rows_per_gpu": self._part_chunk.n_rows,
+                "partsupp_rows_per_gpu": self.n_build,
+                "orders_rows_per_gpu": self.orders[0].numel(),
+                "lineitem_rows_per_gpu": self.n_probe,
+                "part_like_kept": li.get("part_kept"),
+                "lineitem_joined": li.get("lineitem_joined"),
+                "groups_last_step": li.get("groups"),
+                "aggregates": "SUM(amount scale-4 scaled-int, exact DECIMAL) "
+                              "GROUP BY (nation, year), two-phase at N>1",
+                "exchange": "rccl_all_to_allv" if world > 1 else "none",
+                "parallelism": f"hash_shuffle_dp{world}"}
+
+    def cpu_baseline(self, sample_rows):
+        return c5_cpu_baseline(sample_scale=1.0 / 32)
+
+    def traffic_key(self):
+        return "c5_sf300"
+
+
+def c5_cpu_baseline(sample_scale, seed=107):
+    import torch
+    from galaxysql_amd import abi
+    from galaxysql_amd.chunk import Block, Chunk, I64, SLICE
+    from galaxysql_amd.queries import (run_q9, gen_q9_numpy, stage_table,
+                                       Q9_PART_TYPES)
+
+    lib = abi.load_oracle()
+    rng = np.random.default_rng(seed)
+    data = gen_q9_numpy(rng, n_part=int(C5.PART * sample_scale),
+                        n_supp=int(C5.SUPP * sample_scale),
+                        n_orders=int(C5.ORDERS * sample_scale),
+                        n_lineitem=int(C5.LINEITEM * sample_scale))
+    part_chunk = Chunk([Block(I64, values=data[0][0]),
+                        Block.of(SLICE, data[0][1])])
+    part_res = stage_table(lib, part_chunk, Q9_PART_TYPES, -1)
+    t = [[torch.from_numpy(a) for a in cols] for cols in data[1:]]
+    n_li = data[4][0].shape[0]
+    t0 = time.perf_counter()
+    run_q9(lib, -1, part_res, t[0], t[1], t[2], t[3])
+    t1 = time.perf_counter()
+    lib.lib.gxop_result_release(part_res)
+    return {"value": n_li / (t1 - t0), "unit": "rows/s", "cores": 1,
+            "kind": "port",
+            "sample": f"full Q9 chain on a 1/{int(1/sample_scale)} per-GPU "
+                      f"shard ({n_li} lineitem rows), 1 thread, "
+                      "-O3 -march=native"}
+
+
+WORKLOADS = {"c2": C2, "c3": C3, "c4": C4, "c5": C5}
 
 
 def main():
@@ -454,6 +599,8 @@ def main():
                     help="C3 size fraction of SF100 (1.0 = full)")
     ap.add_argument("--c4-scale", type=float, default=1.0,
                     help="C4 size fraction of SF100 (1.0 = full)")
+    ap.add_argument("--c5-scale", type=float, default=1.0,
+                    help="C5 size fraction of the per-GPU SF300/8 shard")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--cpu-probe-rows", type=int, default=8_000_000)
     args = ap.parse_args()
