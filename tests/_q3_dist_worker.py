@@ -12,7 +12,8 @@ sys.path.insert(0, os.environ["PYTHONPATH"])
 
 from galaxysql_amd import abi
 from galaxysql_amd.exchange import shuffle_columns
-from galaxysql_amd.queries import (run_q3, gen_q3_numpy, CUST_TYPES,
+from galaxysql_amd.queries import (run_q3, run_q3_honest, gen_q3_numpy,
+                                   gen_q3_raw_numpy, CUST_TYPES,
                                    ORDERS_TYPES, LINEITEM_TYPES)
 
 
@@ -49,6 +50,31 @@ def main():
     np.savez(os.path.join(outdir, f"q3dist_{rank}.npz"),
              groups=info["groups"], rows=rows, cents=cents,
              joined=info["joined_rows"])
+
+    # HONEST distributed path — exactly what bench.py C3 runs at N>1:
+    # local scans -> as_tensors -> shuffles -> run_q3 with orderkey reshuffle
+    rng2 = np.random.default_rng(777)
+    raw = gen_q3_raw_numpy(rng2, n_cust=1500, n_orders=15000, n_lineitem=60000)
+    raww = [[torch.from_numpy(a) for a in cols] for cols in raw]
+    mycust = [c[rank::world].contiguous() for c in raww[0]]
+    myord = [c[rank::world].contiguous() for c in raww[1]]
+    myli = [c[rank::world].contiguous() for c in raww[2]]
+    (cust2, orders2, lineitem2), scanned = run_q3_honest(
+        lib, -1, mycust, myord, myli, as_tensors=True)
+    cust2 = shuffle_columns(lib, cust2, CUST_TYPES, [0], device=-1)
+    orders2 = shuffle_columns(lib, orders2, ORDERS_TYPES, [0], device=-1)
+    lineitem2 = shuffle_columns(lib, lineitem2, LINEITEM_TYPES, [0], device=-1)
+    chunks2, info2 = run_q3(lib, -1, cust2, orders2, lineitem2, to_host=True,
+                            reshuffle_by_orderkey=True)
+    cents2 = 0
+    rows2 = 0
+    for c in chunks2:
+        for r in c.rows():
+            cents2 += r[4]
+            rows2 += 1
+    np.savez(os.path.join(outdir, f"q3hdist_{rank}.npz"),
+             groups=info2["groups"], rows=rows2, cents=cents2,
+             joined=info2["joined_rows"])
     dist.destroy_process_group()
 
 

@@ -49,3 +49,16 @@ def test_q3_distributed_matches_single(tmp_path):
     assert int(d0["joined"]) + int(d1["joined"]) == info["joined_rows"]
     assert int(d0["groups"]) + int(d1["groups"]) == info["groups"]
     assert int(d0["cents"]) + int(d1["cents"]) == cents
+
+    # honest distributed (the bench's N>1 path) vs single-process honest
+    from galaxysql_amd.queries import run_q3_honest, gen_q3_raw_numpy
+    h0 = np.load(tmp_path / "q3hdist_0.npz")
+    h1 = np.load(tmp_path / "q3hdist_1.npz")
+    rng2 = np.random.default_rng(777)
+    raw = gen_q3_raw_numpy(rng2, n_cust=1500, n_orders=15000, n_lineitem=60000)
+    tr = [[torch.from_numpy(a) for a in cols] for cols in raw]
+    chunks, hinfo = run_q3_honest(lib, -1, tr[0], tr[1], tr[2], to_host=True)
+    hcents = sum(r[4] for c in chunks for r in c.rows())
+    assert int(h0["joined"]) + int(h1["joined"]) == hinfo["joined_rows"]
+    assert int(h0["groups"]) + int(h1["groups"]) == hinfo["groups"]
+    assert int(h0["cents"]) + int(h1["cents"]) == hcents
