@@ -31,6 +31,7 @@
 #include "gx_common.h"
 
 #include <algorithm>
+#include <cstdlib>
 #include <hipcub/hipcub.hpp>
 
 thread_local std::string gx_err;
@@ -223,6 +224,48 @@ __global__ void k_join_scatter(const int32_t *hashes, const uint8_t *keynull,
     }
 }
 
+/* ---- radix-staged probe ------------------------------------------------
+ * For LARGE builds the table (starts+entries) far exceeds the 256 MiB
+ * Infinity Cache, so every probe costs 1-2 random HBM lines. Staging the
+ * probe rows by BUCKET RANGE first makes each range's table slice
+ * L3-resident while its rows are probed: partition p owns buckets
+ * [p<<shift, (p+1)<<shift). The staged 16-B row carries everything the
+ * fast-path probe needs (rowid, hash, inline key), so the probe reads are
+ * payload-sequential + table-slice-local. keynull rows carry the flag in
+ * the rowid's top bit (they still emit LEFT/ANTI rows, match nothing). */
+struct __align__(16) RadixRow {
+    uint32_t rowid;   /* bit31 = keynull */
+    int32_t hash;
+    int64_t key;
+};
+
+__global__ void k_radix_count(const int32_t *hashes, const uint8_t *keynull,
+                              int64_t n, uint32_t mask, int shift,
+                              uint32_t *counts) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t b = keynull[i] ? 0u : ((uint32_t)gx_mix(hashes[i]) & mask);
+        atomicAdd(&counts[b >> shift], 1u);
+    }
+}
+
+__global__ void k_radix_scatter(const int32_t *hashes, const uint8_t *keynull,
+                                DevColView key0, int fast_i64, int64_t n,
+                                uint32_t mask, int shift, uint32_t *cursors,
+                                RadixRow *out) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        bool kn = keynull[i];
+        uint32_t b = kn ? 0u : ((uint32_t)gx_mix(hashes[i]) & mask);
+        uint32_t at = atomicAdd(&cursors[b >> shift], 1u);
+        RadixRow r;
+        r.rowid = (uint32_t)i | (kn ? 0x80000000u : 0u);
+        r.hash = hashes[i];
+        r.key = (!kn && fast_i64) ? ((const int64_t *)key0.values)[i] : 0;
+        out[at] = r;
+    }
+}
+
 /* ========================== join: probe =============================== */
 
 /* Output pair: probe row index + matched build position (0xFFFFFFFF for the
@@ -235,6 +278,7 @@ struct ProbeParams {
     int64_t n_probe;
     const int32_t *hashes;      /* probe row hashes */
     const uint8_t *keynull;     /* probe row has a NULL key */
+    const RadixRow *staged;     /* radix-staged rows (fast path); else NULL */
     int fast_i64;               /* entries carry the key inline */
     KeyViews build_keys;        /* for generic compare */
     KeyViews probe_keys;
@@ -314,16 +358,28 @@ __global__ void k_probe(ProbeParams P) {
         bool active = base_i < P.n_probe;
         if (!__ballot(active)) break;
 
-        int64_t i = active ? base_i : 0;
+        int64_t i = 0;
         bool matched = false;
         uint32_t it = 0, end = 0;
         int64_t want_key = 0;
-        if (active && !P.keynull[i]) {
-            uint32_t b = (uint32_t)gx_mix(P.hashes[i]) & P.mask;
-            it = P.starts[b];
-            end = P.starts[b + 1];
-            if (P.fast_i64)
-                want_key = ((const int64_t *)P.probe_keys.col[0].values)[i];
+        if (P.staged) {
+            RadixRow r = active ? P.staged[base_i] : RadixRow{0x80000000u, 0, 0};
+            i = r.rowid & 0x7FFFFFFFu;
+            if (active && !(r.rowid & 0x80000000u)) {
+                uint32_t b = (uint32_t)gx_mix(r.hash) & P.mask;
+                it = P.starts[b];
+                end = P.starts[b + 1];
+                want_key = r.key;
+            }
+        } else {
+            i = active ? base_i : 0;
+            if (active && !P.keynull[i]) {
+                uint32_t b = (uint32_t)gx_mix(P.hashes[i]) & P.mask;
+                it = P.starts[b];
+                end = P.starts[b + 1];
+                if (P.fast_i64)
+                    want_key = ((const int64_t *)P.probe_keys.col[0].values)[i];
+            }
         }
 
         /* walk candidates; lanes iterate together so emissions batch */
@@ -713,6 +769,7 @@ struct JoinOp : gx_op {
     DevBuf d_bitmap;         /* build_outer matched bitmap */
     DevBuf d_scan_tmp;
     DevBuf d_pidx, d_bpos, d_meta, d_ph, d_pn; /* reused across probe calls */
+    DevBuf d_staged, d_radix_cnt;               /* radix-staged probe rows */
     uint32_t mask = 0;
     int64_t n_buckets = 0;
     bool fast_i64 = false;
@@ -744,6 +801,7 @@ struct JoinOp : gx_op {
         d_scan_tmp.release();
         d_pidx.release(); d_bpos.release(); d_meta.release();
         d_ph.release(); d_pn.release();
+        d_staged.release(); d_radix_cnt.release();
         if (ev0) (void)hipEventDestroy(ev0);
         if (ev1) (void)hipEventDestroy(ev1);
     }
@@ -979,6 +1037,53 @@ struct JoinOp : gx_op {
             uint32_t cap = (uint32_t)std::min<int64_t>(
                 std::max<int64_t>((int64_t)n * 2, 1 << 16), INT64_C(1) << 31);
             if (d_meta.grow(8, stream)) break;
+
+            /* radix-stage the probe rows by bucket range when the table far
+             * exceeds the Infinity Cache (fast path only) — slices of
+             * starts+entries become L3-resident while their rows probe. */
+            const int64_t table_bytes =
+                (n_buckets + 1) * 4 + build.n_rows * (int64_t)sizeof(JoinEntry);
+            int64_t n_part_radix = 1;
+            int radix_shift = 0;
+            const bool radix_force = getenv("GX_RADIX_FORCE") != nullptr;
+            if (fast_i64 && (radix_force ||
+                             (n >= (1 << 22) && table_bytes > (192 << 20)))) {
+                int64_t target = radix_force ? std::max<int64_t>(table_bytes / 4, 1)
+                                             : (96 << 20);
+                n_part_radix = gx_pow2((table_bytes + target - 1) / target);
+                if (n_part_radix > n_buckets) n_part_radix = 1;
+                while ((1 << radix_shift) < n_buckets / n_part_radix)
+                    radix_shift++;
+            }
+            std::vector<uint32_t> radix_starts;
+            if (n_part_radix > 1) {
+                if (d_staged.grow((size_t)n * sizeof(RadixRow), stream) ||
+                    d_radix_cnt.grow((size_t)n_part_radix * 4, stream)) break;
+                HIP_OK(hipMemsetAsync(d_radix_cnt.p, 0, (size_t)n_part_radix * 4,
+                                      stream));
+                hipLaunchKernelGGL(k_radix_count, dim3(gx_grid(n)), dim3(256), 0,
+                                   stream, (const int32_t *)d_ph.p,
+                                   (const uint8_t *)d_pn.p, n, mask, radix_shift,
+                                   (uint32_t *)d_radix_cnt.p);
+                std::vector<uint32_t> counts((size_t)n_part_radix);
+                HIP_OK(hipMemcpyAsync(counts.data(), d_radix_cnt.p,
+                                      (size_t)n_part_radix * 4,
+                                      hipMemcpyDeviceToHost, stream));
+                HIP_OK(hipStreamSynchronize(stream));
+                radix_starts.assign((size_t)n_part_radix + 1, 0);
+                for (int64_t p = 0; p < n_part_radix; p++)
+                    radix_starts[p + 1] = radix_starts[p] + counts[p];
+                HIP_OK(hipMemcpyAsync(d_radix_cnt.p, radix_starts.data(),
+                                      (size_t)n_part_radix * 4,
+                                      hipMemcpyHostToDevice, stream));
+                hipLaunchKernelGGL(k_radix_scatter, dim3(gx_grid(n)), dim3(256),
+                                   0, stream, (const int32_t *)d_ph.p,
+                                   (const uint8_t *)d_pn.p,
+                                   pk.col[0], 1, n, mask, radix_shift,
+                                   (uint32_t *)d_radix_cnt.p,
+                                   (RadixRow *)d_staged.p);
+            }
+
             for (int attempt = 0; attempt < 4; attempt++) {
                 if (d_pidx.grow((size_t)cap * 4, stream) ||
                     d_bpos.grow((size_t)cap * 4, stream)) { attempt = 99; break; }
@@ -990,6 +1095,7 @@ struct JoinOp : gx_op {
                 P.n_probe = n;
                 P.hashes = (const int32_t *)d_ph.p;
                 P.keynull = (const uint8_t *)d_pn.p;
+                P.staged = nullptr;
                 P.fast_i64 = (int)fast_i64;
                 P.build_keys = key_views(build, build_key_cols);
                 P.probe_keys = pk;
@@ -1016,7 +1122,20 @@ struct JoinOp : gx_op {
                     HIP_OK(hipEventCreate(&ev1));
                 }
                 HIP_OK(hipEventRecord(ev0, stream));
-                hipLaunchKernelGGL(k_probe, dim3(gx_grid(n)), dim3(256), 0, stream, P);
+                if (n_part_radix > 1) {
+                    for (int64_t p = 0; p < n_part_radix; p++) {
+                        int64_t cnt = (int64_t)radix_starts[p + 1] - radix_starts[p];
+                        if (cnt == 0) continue;
+                        ProbeParams Pp = P;
+                        Pp.staged = (const RadixRow *)d_staged.p + radix_starts[p];
+                        Pp.n_probe = cnt;
+                        hipLaunchKernelGGL(k_probe, dim3(gx_grid(cnt)), dim3(256),
+                                           0, stream, Pp);
+                    }
+                } else {
+                    hipLaunchKernelGGL(k_probe, dim3(gx_grid(n)), dim3(256), 0,
+                                       stream, P);
+                }
                 HIP_OK(hipEventRecord(ev1, stream));
                 uint32_t meta[2];
                 HIP_OK(hipMemcpyAsync(meta, d_meta.p, 8, hipMemcpyDeviceToHost, stream));

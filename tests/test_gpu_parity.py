@@ -285,3 +285,31 @@ def test_agg_small_chunk_stream(libs):
     both_agg(libs, [0], [(abi.SUM_I64, 1), (abi.COUNT_ROW, -1),
                          (abi.MIN_I64, 1)], [I64, I64], chunks,
              expected_groups=64)
+
+
+def test_join_radix_staged_probe(libs):
+    """Force the radix-staged probe path (bucket-range partitioning for
+    L3-resident table slices) and check exact parity vs the oracle."""
+    import os
+    rng = np.random.default_rng(33)
+    n_build, n_probe = 300_000, 900_000
+    bkeys = rng.integers(0, 200_000, size=n_build, dtype=np.int64)
+    build = chunks_from_columns(
+        [I64, I64], [(bkeys, None),
+                     (rng.integers(0, 1 << 30, n_build, dtype=np.int64), None)],
+        chunk_size=50_000)
+    pkeys = rng.integers(0, 220_000, size=n_probe, dtype=np.int64)
+    nulls = (rng.random(n_probe) < 0.01).astype(np.uint8)
+    probe = chunks_from_columns(
+        [I64, I64], [(pkeys, nulls),
+                     (rng.integers(0, 1 << 30, n_probe, dtype=np.int64), None)],
+        chunk_size=300_000)
+    keys = [EquiJoinKey(0, 0, I64)]
+    os.environ["GX_RADIX_FORCE"] = "1"
+    try:
+        for jt in (abi.INNER, abi.LEFT, abi.SEMI, abi.ANTI):
+            ref, got = both_join(libs, jt, keys, build, probe,
+                                 [I64, I64], [I64, I64])
+            assert multiset(got) == multiset(ref), f"join_type={jt}"
+    finally:
+        del os.environ["GX_RADIX_FORCE"]
