@@ -46,11 +46,6 @@ struct KeyViews {
     DevColView col[GX_MAX_KEYS];
 };
 
-struct ColViews {
-    int32_t n;
-    DevColView col[GX_MAX_COLS];
-};
-
 
 /* airlift Slice.hashCode = (int) XxHash64(seed 0) over raw bytes
  * (chunk/SliceBlock.java:183-195, non-"compatible" path) — device

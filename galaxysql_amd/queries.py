@@ -70,13 +70,6 @@ def _src_rows(src):
     return src.contents.chunk.n_rows
 
 
-def _probe_result(lib, op, res):
-    out = C.POINTER(GxResult)()
-    lib.check(lib.lib.gxop_join_probe(op._op, C.byref(res.contents.chunk),
-                                      C.byref(out)), "join_probe")
-    return out
-
-
 _TORCH_DT = None
 
 
