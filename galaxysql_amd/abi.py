@@ -67,6 +67,8 @@ class GxJoinCfg(C.Structure):
         ("device", C.c_int32),
         ("stream", C.c_uint64),
         ("expected_build_rows", C.c_int64),
+        ("n_out_proj", C.c_int32),
+        ("out_proj", C.POINTER(C.c_int32)),
     ]
 
 

@@ -788,6 +788,7 @@ struct StagedChunk {
 struct JoinOp : gx_op {
     gx_join_cfg cfg;
     std::vector<gx_equi_key> keys;
+    std::vector<int32_t> out_proj;  /* projection pushdown; empty = full */
     std::vector<int32_t> outer_types, inner_types;
     std::vector<int> build_key_cols, probe_key_cols;
 
@@ -812,6 +813,8 @@ struct JoinOp : gx_op {
 
     JoinOp(const gx_join_cfg *c) : gx_op(OP_JOIN, c->device, c->stream), cfg(*c) {
         keys.assign(c->keys, c->keys + c->n_keys);
+        if (c->n_out_proj > 0)
+            out_proj.assign(c->out_proj, c->out_proj + c->n_out_proj);
         outer_types.assign(c->outer_types, c->outer_types + c->n_outer_cols);
         inner_types.assign(c->inner_types, c->inner_types + c->n_inner_cols);
         for (auto &k : keys) {
@@ -937,20 +940,43 @@ struct JoinOp : gx_op {
         return 0;
     }
 
-    std::vector<int32_t> output_types() const {
-        if (cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI)
-            return outer_types;
-        std::vector<int32_t> inner_out = cfg.single_join
-            ? std::vector<int32_t>(inner_types.begin(), inner_types.begin() + 1)
-            : inner_types;
-        std::vector<int32_t> t;
-        if (cfg.join_type != GX_JOIN_RIGHT) {
-            t = outer_types;
-            t.insert(t.end(), inner_out.begin(), inner_out.end());
-        } else {
-            t = inner_types;
-            t.insert(t.end(), outer_types.begin(), outer_types.end());
+    struct OutSpec { bool outer; int src; int32_t type; };
+
+    /* full output schema in reference order (AbstractJoinExec.java:102-227) */
+    std::vector<OutSpec> full_specs() const {
+        std::vector<OutSpec> s;
+        auto add = [&](bool is_outer, int i, int32_t t) { s.push_back({is_outer, i, t}); };
+        if (cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI) {
+            for (size_t i = 0; i < outer_types.size(); i++)
+                add(true, (int)i, outer_types[i]);
+            return s;
         }
+        size_t n_inner_out = cfg.single_join ? 1 : inner_types.size();
+        if (cfg.join_type != GX_JOIN_RIGHT) {
+            for (size_t i = 0; i < outer_types.size(); i++)
+                add(true, (int)i, outer_types[i]);
+            for (size_t i = 0; i < n_inner_out; i++)
+                add(false, (int)i, inner_types[i]);
+        } else {
+            for (size_t i = 0; i < inner_types.size(); i++)
+                add(false, (int)i, inner_types[i]);
+            for (size_t i = 0; i < outer_types.size(); i++)
+                add(true, (int)i, outer_types[i]);
+        }
+        return s;
+    }
+
+    std::vector<OutSpec> projected_specs() const {
+        std::vector<OutSpec> full = full_specs();
+        if (out_proj.empty()) return full;
+        std::vector<OutSpec> sel;
+        for (int32_t i : out_proj) sel.push_back(full[(size_t)i]);
+        return sel;
+    }
+
+    std::vector<int32_t> output_types() const {
+        std::vector<int32_t> t;
+        for (auto &s : projected_specs()) t.push_back(s.type);
         return t;
     }
 
@@ -976,36 +1002,26 @@ struct JoinOp : gx_op {
         int n_fused = 0;
         struct SliceJob { DevColView src; const uint32_t *idx; int null_fill; size_t col; };
         std::vector<SliceJob> slice_jobs;
-        auto add_col = [&](bool from_build, int src_col) {
-            DevColView src = from_build ? build.view(src_col)
-                                        : probe.views[src_col];
+        /* spec.outer refers to the logical OUTER input; map to probe/build
+         * honoring build_outer (ParallelHashJoinExec buildJoinRow overrides) */
+        for (auto &spec : projected_specs()) {
+            const bool from_build = (spec.outer && cfg.build_outer) ||
+                                    (!spec.outer && !cfg.build_outer);
+            DevColView src = from_build ? build.view(spec.src)
+                                        : probe.views[spec.src];
             const uint32_t *idx = from_build ? d_bpos : d_pidx;
             if (src.type == GX_SLICE) {
                 slice_jobs.push_back({src, idx, idx == nullptr ? 1 : 0,
                                       (size_t)col});
                 col++;
-                return;
+                continue;
             }
-            /* compact fused slots but keep out buffers per logical col */
             G.src[n_fused] = src;
             G.side[n_fused] = from_build ? (d_bpos ? 1 : 2) : (d_pidx ? 0 : 2);
             G.out_vals[n_fused] = h->bufs[col * 2].p;
             G.out_nulls[n_fused] = (uint8_t *)h->bufs[col * 2 + 1].p;
             n_fused++;
             col++;
-        };
-        bool semi = cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI;
-        size_t n_outer = outer_types.size();
-        size_t n_inner_out = cfg.single_join ? 1 : inner_types.size();
-        bool outer_from_build = cfg.build_outer;
-        if (semi) {
-            for (size_t i = 0; i < n_outer; i++) add_col(outer_from_build, (int)i);
-        } else if (cfg.join_type != GX_JOIN_RIGHT) {
-            for (size_t i = 0; i < n_outer; i++) add_col(outer_from_build, (int)i);
-            for (size_t i = 0; i < n_inner_out; i++) add_col(!outer_from_build, (int)i);
-        } else {
-            for (size_t i = 0; i < inner_types.size(); i++) add_col(!outer_from_build, (int)i);
-            for (size_t i = 0; i < n_outer; i++) add_col(outer_from_build, (int)i);
         }
         G.n_cols = n_fused;
         if (n_fused > 0)
@@ -1229,31 +1245,26 @@ struct JoinOp : gx_op {
         if (!h) { d_idx.release(); d_cnt.release(); return -1; }
         size_t col = 0;
         int rc_col = 0;
-        auto emit_col = [&](bool from_build, int src_col, bool null_fill) {
-            DevColView src = from_build ? build.view(src_col)
-                                        : build.view(0) /* placeholder for null fill */;
-            if (null_fill) src.type = otypes[col];
-            if (otypes[col] == GX_SLICE) {
+        for (auto &spec : projected_specs()) {
+            /* buildOuter tail: outer columns come from the BUILD store,
+             * inner columns are null-filled (nextJoinNullRows:372-401) */
+            const bool from_build = spec.outer;
+            const bool null_fill = !spec.outer;
+            DevColView src = from_build ? build.view(spec.src) : build.view(0);
+            if (null_fill) src.type = spec.type;
+            if (spec.type == GX_SLICE) {
                 if (gather_slice_col(src, (const uint32_t *)d_idx.p,
                                      (int)null_fill, cnt, h, col, d_scan_tmp,
                                      stream))
                     rc_col = -1;
                 col++;
-                return;
+                continue;
             }
             hipLaunchKernelGGL(k_gather, dim3(gx_grid(cnt)), dim3(256), 0, stream,
                                src, (const uint32_t *)d_idx.p, (int64_t)cnt,
                                h->bufs[col * 2].p, (uint8_t *)h->bufs[col * 2 + 1].p,
                                (int)null_fill);
             col++;
-        };
-        size_t n_inner_out = cfg.single_join ? 1 : inner_types.size();
-        if (cfg.join_type != GX_JOIN_RIGHT) {
-            for (size_t i = 0; i < outer_types.size(); i++) emit_col(true, (int)i, false);
-            for (size_t i = 0; i < n_inner_out; i++) emit_col(false, 0, true);
-        } else {
-            for (size_t i = 0; i < inner_types.size(); i++) emit_col(false, 0, true);
-            for (size_t i = 0; i < outer_types.size(); i++) emit_col(true, (int)i, false);
         }
         HIP_OK(hipStreamSynchronize(stream));
         d_idx.release(); d_cnt.release();

@@ -36,7 +36,7 @@ class ParallelHashJoinExec:
 
     def __init__(self, lib, join_type, join_keys, outer_types, inner_types,
                  max_one_row=False, build_outer=False, anti_null_col=-1,
-                 device=-1, stream=0, expected_build_rows=0):
+                 device=-1, stream=0, expected_build_rows=0, out_proj=None):
         self._lib = lib
         self._keep = []
         keys = (GxEquiKey * len(join_keys))()
@@ -44,14 +44,16 @@ class ParallelHashJoinExec:
             keys[i] = GxEquiKey(k.outer_index, k.inner_index, k.unified_type, 0)
         ot = (C.c_int32 * len(outer_types))(*outer_types)
         it = (C.c_int32 * len(inner_types))(*inner_types)
+        op_arr = (C.c_int32 * max(1, len(out_proj or [])))(*(out_proj or [0]))
         cfg = GxJoinCfg(
             join_type=join_type, single_join=int(max_one_row),
             build_outer=int(build_outer), n_keys=len(join_keys), keys=keys,
             n_outer_cols=len(outer_types), outer_types=ot,
             n_inner_cols=len(inner_types), inner_types=it,
             anti_null_col=anti_null_col, device=device, stream=stream,
-            expected_build_rows=expected_build_rows)
-        self._keep += [keys, ot, it, cfg]
+            expected_build_rows=expected_build_rows,
+            n_out_proj=len(out_proj or []), out_proj=op_arr)
+        self._keep += [keys, ot, it, op_arr, cfg]
         self._op = lib.lib.gxop_join_create(C.byref(cfg))
         if not self._op:
             raise RuntimeError(f"gxop_join_create: {lib.error()}")

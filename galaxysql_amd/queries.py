@@ -119,10 +119,14 @@ def run_q3(lib, device, cust, orders, lineitem, expected_groups=0,
 
         # join2: INNER — build the surviving orders (key col 1 = o_orderkey),
         # probe lineitem on l_orderkey
+        # projection pushdown: of the full join2 schema
+        # [l_orderkey, revenue, cents, o_custkey, o_orderkey, o_date, o_prio]
+        # the aggregate needs only [0, 1, 2, 5, 6]
         j2 = ParallelHashJoinExec(
             lib, abi.INNER, [EquiJoinKey(0, 1, I64)],
             outer_types=LINEITEM_TYPES, inner_types=ORDERS_TYPES,
-            device=device, expected_build_rows=n_orders_kept)
+            device=device, expected_build_rows=n_orders_kept,
+            out_proj=[0, 1, 2, 5, 6])
         try:
             if reshuffle_by_orderkey:
                 # N>1: join1 ran custkey-sharded; its result re-shards by
@@ -152,12 +156,11 @@ def run_q3(lib, device, cust, orders, lineitem, expected_groups=0,
 
             # agg: GROUP BY (l_orderkey, o_orderdate, o_shippriority)
             #      SUM(revenue) f64, SUM(revenue_cents) i64, COUNT(*)
-            # join2 output columns: [l_orderkey, revenue, revenue_cents,
-            #                        o_custkey, o_orderkey, o_orderdate, o_prio]
+            # projected join2 output: [l_orderkey, revenue, cents, date, prio]
             agg = HashAggExec(
-                lib, group_cols=[0, 5, 6],
+                lib, group_cols=[0, 3, 4],
                 aggs=[(abi.SUM_F64, 1), (abi.SUM_I64, 2), (abi.COUNT_ROW, -1)],
-                input_types=LINEITEM_TYPES + ORDERS_TYPES,
+                input_types=[I64, F64, I64, I32, I32],
                 expected_groups=expected_groups or max(1024, n_orders_kept),
                 device=device)
             try:

@@ -125,6 +125,11 @@ typedef struct gx_join_cfg {
     int32_t device;           /* HIP device ordinal; -1 only in the CPU oracle */
     uint64_t stream;          /* hipStream_t, 0 = default stream */
     int64_t expected_build_rows; /* size hint; 0 = unknown */
+    /* projection pushdown (the planner's Project-above-join collapse):
+     * indices into the join's FULL output schema; n_out_proj=0 keeps the
+     * reference schema unchanged. */
+    int32_t n_out_proj;
+    const int32_t *out_proj;
 } gx_join_cfg;
 
 typedef struct gx_op gx_op;  /* opaque operator instance */

@@ -371,6 +371,7 @@ namespace {
 struct JoinOp : gx_op {
     gx_join_cfg cfg;
     std::vector<gx_equi_key> keys;
+    std::vector<int32_t> out_proj;  /* projection pushdown; empty = full */
     std::vector<int32_t> outer_types, inner_types;
 
     Store build;                 /* build-side input columns */
@@ -394,6 +395,8 @@ struct JoinOp : gx_op {
 
     JoinOp(const gx_join_cfg *c) : gx_op(OP_JOIN), cfg(*c) {
         keys.assign(c->keys, c->keys + c->n_keys);
+        if (c->n_out_proj > 0)
+            out_proj.assign(c->out_proj, c->out_proj + c->n_out_proj);
         outer_types.assign(c->outer_types, c->outer_types + c->n_outer_cols);
         inner_types.assign(c->inner_types, c->inner_types + c->n_inner_cols);
         for (auto &k : keys) {
@@ -528,6 +531,14 @@ struct JoinOp : gx_op {
         return t;
     }
 
+    gx_result *project_result(std::vector<OutCol> &&cols, int32_t n_rows) {
+        if (out_proj.empty()) return make_result(std::move(cols), n_rows);
+        std::vector<OutCol> sel;
+        sel.reserve(out_proj.size());
+        for (int32_t idx : out_proj) sel.push_back(std::move(cols[idx]));
+        return make_result(std::move(sel), n_rows);
+    }
+
     int probe(const gx_chunk *ch, gx_result **out) {
         *out = nullptr;
         if (!built) { set_err("probe before build"); return -1; }
@@ -542,13 +553,13 @@ struct JoinOp : gx_op {
         std::vector<OutCol> cols(otypes.size());
         for (size_t c = 0; c < otypes.size(); c++) cols[c].type = otypes[c];
 
-        if (pass_nothing) { *out = make_result(std::move(cols), 0); return 0; }
+        if (pass_nothing) { *out = project_result(std::move(cols), 0); return 0; }
         if (pass_through) {
             /* ANTI with empty build: every probe row passes, no operand check
              * (AbstractBufferedJoinExec.doSpecialCheckForSemiJoin:296-301) */
             for (size_t r = 0; r < probe_store.n_rows; r++)
                 emit_match(cols, probe_store, r, -1);
-            *out = make_result(std::move(cols), (int32_t)probe_store.n_rows);
+            *out = project_result(std::move(cols), (int32_t)probe_store.n_rows);
             return 0;
         }
 
@@ -609,7 +620,7 @@ struct JoinOp : gx_op {
                 }
             }
         }
-        *out = make_result(std::move(cols), n_rows);
+        *out = project_result(std::move(cols), n_rows);
         return 0;
     }
 
@@ -642,7 +653,7 @@ struct JoinOp : gx_op {
         }
         tail_cursor = build.n_rows;
         tail_done = true;
-        if (n_rows > 0) *out = make_result(std::move(cols), n_rows);
+        if (n_rows > 0) *out = project_result(std::move(cols), n_rows);
         return 0;
     }
 };

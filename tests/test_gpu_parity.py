@@ -313,3 +313,23 @@ def test_join_radix_staged_probe(libs):
             assert multiset(got) == multiset(ref), f"join_type={jt}"
     finally:
         del os.environ["GX_RADIX_FORCE"]
+
+
+def test_join_projection_pushdown(libs):
+    """out_proj: the planner's Project-above-join collapse — output only a
+    subset/permutation of the full join schema."""
+    rng = np.random.default_rng(35)
+    build = make_chunks(rng, [I64, I64, F64], 2000, key_space=900, null_frac=0.05)
+    probe = make_chunks(rng, [I64, I32], 6000, key_space=1000, null_frac=0.05)
+    keys = [EquiJoinKey(0, 0, I64)]
+    # full schema: [probe0 I64, probe1 I32, build0 I64, build1 I64, build2 F64]
+    proj = [3, 0, 4]
+    for jt in (abi.INNER, abi.LEFT):
+        ref, got = both_join(libs, jt, keys, build, probe,
+                             [I64, I32], [I64, I64, F64], out_proj=proj)
+        assert multiset(got, f64_round=9) == multiset(ref, f64_round=9), jt
+    # buildOuter tail with projection
+    ref, got = both_join(libs, abi.LEFT, keys, build, probe,
+                         [I64, I32], [I64, I64, F64], out_proj=[0, 3],
+                         build_outer=True)
+    assert multiset(got) == multiset(ref)
