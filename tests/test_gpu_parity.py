@@ -328,8 +328,12 @@ def test_join_projection_pushdown(libs):
         ref, got = both_join(libs, jt, keys, build, probe,
                              [I64, I32], [I64, I64, F64], out_proj=proj)
         assert multiset(got, f64_round=9) == multiset(ref, f64_round=9), jt
-    # buildOuter tail with projection
-    ref, got = both_join(libs, abi.LEFT, keys, build, probe,
+    # buildOuter tail with projection: the BUILD side is the OUTER input,
+    # so consume receives outer-typed chunks and probe inner-typed ones
+    bo_build = make_chunks(rng, [I64, I32], 1500, key_space=700, null_frac=0.05)
+    bo_probe = make_chunks(rng, [I64, I64, F64], 4000, key_space=800,
+                           null_frac=0.05)
+    ref, got = both_join(libs, abi.LEFT, keys, bo_build, bo_probe,
                          [I64, I32], [I64, I64, F64], out_proj=[0, 3],
                          build_outer=True)
     assert multiset(got) == multiset(ref)
