@@ -14,7 +14,8 @@ import os
 
 import numpy as np
 
-from .chunk import Block, Chunk, I64, I32, F64, SLICE, _NP_DTYPES
+from .chunk import (Block, Chunk, I64, I32, F64, SLICE, DECIMAL,
+                    _NP_DTYPES)
 
 _REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
@@ -147,6 +148,7 @@ LT, LE, GT, GE, EQ, NE = 0, 1, 2, 3, 4, 5
 CONTAINS = 6  # SLICE LIKE '%pat%'
 # Projections (gx_proj_op)
 PROJ_COPY, PROJ_REV_F64, PROJ_REV_SCALED4, PROJ_Q9_AMOUNT4 = 0, 1, 2, 3
+PROJ_DEC_TO_SCALED, PROJ_SCALED_TO_DEC = 4, 5  # gx_proj.c = decimal scale
 # Agg funcs (gx_agg_func)
 COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64, MAX_F64 = range(8)
 
@@ -265,6 +267,12 @@ class GxLib:
                     C.cast(gb.data, C.POINTER(C.c_uint8)), shape=(dlen,)).copy() \
                     if dlen else np.zeros(0, np.uint8)
                 blocks.append(Block(SLICE, nulls=nulls, offsets=offsets, data=data))
+            elif gb.type == DECIMAL:
+                vals = np.ctypeslib.as_array(
+                    C.cast(gb.values, C.POINTER(C.c_uint8)),
+                    shape=(n, 40)).copy() \
+                    if n else np.zeros((0, 40), np.uint8)
+                blocks.append(Block(DECIMAL, values=vals, nulls=nulls))
             else:
                 ctype = {I64: C.c_int64, I32: C.c_int32, F64: C.c_double}[gb.type]
                 vals = np.ctypeslib.as_array(

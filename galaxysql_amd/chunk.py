@@ -15,7 +15,7 @@ import numpy as np
 
 CHUNK_SIZE = 1000
 
-I64, I32, F64, SLICE = 0, 1, 2, 3
+I64, I32, F64, SLICE, DECIMAL = 0, 1, 2, 3, 4
 
 _NP_DTYPES = {I64: np.int64, I32: np.int32, F64: np.float64}
 
@@ -23,7 +23,9 @@ _NP_DTYPES = {I64: np.int64, I32: np.int32, F64: np.float64}
 class Block:
     """One column of a batch.
 
-    values: numpy array (i64/i32/f64); None for SLICE.
+    values: numpy array (i64/i32/f64); (n, 40) uint8 for DECIMAL
+    (DecimalBlock "simple" 40-B records, chunk/DecimalBlock.java /
+    DecimalBox layout — see include/gxop.h); None for SLICE.
     nulls:  uint8 array (1 = NULL) or None.
     offsets/data: SLICE only — int32 END offsets per value + byte buffer.
     """
@@ -60,6 +62,15 @@ class Block:
                 offsets[i] = len(data)
             return Block(SLICE, nulls=nulls if has_null else None,
                          offsets=offsets, data=np.frombuffer(bytes(data), dtype=np.uint8))
+        if btype == DECIMAL:
+            vals = np.zeros((n, 40), dtype=np.uint8)
+            for i, v in enumerate(pyvals):
+                if v is not None:
+                    scaled, scale = v  # (scaled int, decimal scale)
+                    vals[i] = np.frombuffer(dec40_encode(scaled, scale),
+                                            dtype=np.uint8)
+            return Block(DECIMAL, values=vals,
+                         nulls=nulls if has_null else None)
         vals = np.array([0 if v is None else v for v in pyvals], dtype=_NP_DTYPES[btype])
         return Block(btype, values=vals, nulls=nulls if has_null else None)
 
@@ -69,6 +80,8 @@ class Block:
         if self.type == SLICE:
             start = int(self.offsets[i - 1]) if i > 0 else 0
             return bytes(self.data[start:int(self.offsets[i])])
+        if self.type == DECIMAL:
+            return bytes(self.values[i])
         v = self.values[i]
         if self.type == F64:
             return float(v)
@@ -132,6 +145,43 @@ def chunks_from_columns(btypes, columns, chunk_size=CHUNK_SIZE):
                                  nulls=None if b.nulls is None else b.nulls[start:end]))
         out.append(Chunk(sub))
     return out
+
+
+def dec40_encode(scaled: int, scale: int) -> bytes:
+    """Python mirror of oracle scaled_to_dec40 (DecimalBox simple layout:
+    base-1e9 words w0[,w1[,w2]], integers@36, fractions@37,
+    derivedFractions@38, isNeg@39)."""
+    import struct
+    neg = scaled < 0
+    a = -scaled if neg else scaled
+    ip, rem = divmod(a, 10 ** scale)
+    fr = rem * 10 ** (9 - scale)
+    if ip >= 10 ** 9:
+        w = (ip // 10 ** 9, ip % 10 ** 9, fr)
+        integers = 18
+    else:
+        w = (ip, fr, 0)
+        integers = 9
+    p = bytearray(40)
+    struct.pack_into("<iii", p, 0, *[int(x) for x in w])
+    p[36] = integers
+    p[37] = scale
+    p[38] = scale
+    p[39] = 1 if neg else 0
+    return bytes(p)
+
+
+def dec40_decode(p, scale: int) -> int:
+    """Python mirror of oracle dec40_to_scaled."""
+    import struct
+    w0, w1, w2 = struct.unpack_from("<iii", bytes(p), 0)
+    integers, isneg = p[36], p[39]
+    if integers > 9:
+        ip, fr = w0 * 10 ** 9 + w1, w2
+    else:
+        ip, fr = w0, w1
+    v = ip * 10 ** scale + fr // 10 ** (9 - scale)
+    return -v if isneg else v
 
 
 def multiset(rows, f64_round=None):

@@ -180,6 +180,7 @@ struct DevColumn {
         switch (type) {
         case GX_I64: case GX_F64: return 8;
         case GX_I32: return 4;
+        case GX_DECIMAL: return 40;
         default: return 0;
         }
     }

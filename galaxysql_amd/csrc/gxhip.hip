@@ -38,6 +38,15 @@ thread_local std::string gx_err;
 
 /* ================= kernel-side column descriptors ====================== */
 
+static inline size_t gx_fixed_size(int32_t t) {
+    switch (t) {
+    case GX_I64: case GX_F64: return 8;
+    case GX_I32: return 4;
+    case GX_DECIMAL: return 40;
+    default: return 0;
+    }
+}
+
 #define GX_MAX_KEYS 4
 #define GX_MAX_COLS 16
 
@@ -515,6 +524,10 @@ __global__ void k_gather_multi(GatherParams G) {
                 case GX_I64: ((int64_t *)G.out_vals[c])[i] = 0; break;
                 case GX_I32: ((int32_t *)G.out_vals[c])[i] = 0; break;
                 case GX_F64: ((double *)G.out_vals[c])[i] = 0; break;
+                case GX_DECIMAL:
+                    for (int k = 0; k < 5; k++)
+                        ((uint64_t *)G.out_vals[c])[i * 5 + k] = 0;
+                    break;
                 }
                 continue;
             }
@@ -524,6 +537,11 @@ __global__ void k_gather_multi(GatherParams G) {
             case GX_I64: ((int64_t *)G.out_vals[c])[i] = nn ? 0 : ((const int64_t *)src.values)[s]; break;
             case GX_I32: ((int32_t *)G.out_vals[c])[i] = nn ? 0 : ((const int32_t *)src.values)[s]; break;
             case GX_F64: ((double *)G.out_vals[c])[i] = nn ? 0 : ((const double *)src.values)[s]; break;
+            case GX_DECIMAL:
+                for (int k = 0; k < 5; k++)
+                    ((uint64_t *)G.out_vals[c])[i * 5 + k] = nn ? 0
+                        : ((const uint64_t *)src.values)[(int64_t)s * 5 + k];
+                break;
             }
         }
     }
@@ -542,6 +560,10 @@ __global__ void k_gather(DevColView src, const uint32_t *idx, int64_t n,
             case GX_I64: ((int64_t *)out_vals)[i] = 0; break;
             case GX_I32: ((int32_t *)out_vals)[i] = 0; break;
             case GX_F64: ((double *)out_vals)[i] = 0; break;
+            case GX_DECIMAL:
+                for (int k = 0; k < 5; k++)
+                    ((uint64_t *)out_vals)[i * 5 + k] = 0;
+                break;
             }
             continue;
         }
@@ -551,6 +573,11 @@ __global__ void k_gather(DevColView src, const uint32_t *idx, int64_t n,
         case GX_I64: ((int64_t *)out_vals)[i] = nn ? 0 : ((const int64_t *)src.values)[s]; break;
         case GX_I32: ((int32_t *)out_vals)[i] = nn ? 0 : ((const int32_t *)src.values)[s]; break;
         case GX_F64: ((double *)out_vals)[i] = nn ? 0 : ((const double *)src.values)[s]; break;
+        case GX_DECIMAL:
+            for (int k = 0; k < 5; k++)
+                ((uint64_t *)out_vals)[i * 5 + k] = nn ? 0
+                    : ((const uint64_t *)src.values)[(int64_t)s * 5 + k];
+            break;
         }
     }
 }
@@ -642,7 +669,7 @@ HipResult *alloc_result_cols(const std::vector<int32_t> &types, int64_t n,
     h->blocks.resize(types.size());
     h->bufs.resize(types.size() * 2);
     for (size_t c = 0; c < types.size(); c++) {
-        size_t es = (types[c] == GX_I32 || types[c] == GX_SLICE) ? 4 : 8;
+        size_t es = types[c] == GX_SLICE ? 4 : gx_fixed_size(types[c]);
         if (n > 0) {
             if (h->bufs[c * 2].grow((size_t)n * es, stream) ||
                 h->bufs[c * 2 + 1].grow((size_t)n, stream)) {
@@ -754,7 +781,7 @@ struct StagedChunk {
                 }
                 continue;
             }
-            size_t es = (b->type == GX_I32) ? 4 : 8;
+            size_t es = gx_fixed_size(b->type);
             if (b->mem == GX_MEM_DEVICE) {
                 v.values = b->values;
                 v.nulls = (const uint8_t *)b->nulls;
@@ -1336,7 +1363,7 @@ int gxop_result_to_host(gx_result *res) {
                                  (size_t)b.data_len, hipMemcpyDeviceToHost));
             b.data = (const uint8_t *)h->host.back().data();
         } else {
-            size_t es = (b.type == GX_I32) ? 4 : 8;
+            size_t es = gx_fixed_size(b.type);
             h->host.emplace_back((size_t)n * es);
             if (n > 0)
                 HIP_OK(hipMemcpy(h->host.back().data(), b.values, (size_t)n * es,
@@ -1368,7 +1395,7 @@ int gxop_result_copy_col(const gx_result *res, int32_t col, void *dst_values,
     const gx_block *b = &res->chunk.blocks[col];
     int64_t n = res->chunk.n_rows;
     if (n == 0) return 0;
-    size_t es = (b->type == GX_I32) ? 4 : 8;
+    size_t es = gx_fixed_size(b->type);
     HIP_OK(hipMemcpy(dst_values, b->values, (size_t)n * es, hipMemcpyDefault));
     if (dst_nulls && b->nulls)
         HIP_OK(hipMemcpy(dst_nulls, (const void *)b->nulls, (size_t)n,

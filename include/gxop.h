@@ -47,7 +47,16 @@ typedef enum gx_type {
     GX_I64 = 0,   /* LongBlock:    int64 values[] + nulls[]                */
     GX_I32 = 1,   /* IntegerBlock: int32 values[] + nulls[]                */
     GX_F64 = 2,   /* DoubleBlock:  double values[] + nulls[]               */
-    GX_SLICE = 3  /* SliceBlock:   int32 end-offsets[] + byte data         */
+    GX_SLICE = 3, /* SliceBlock:   int32 end-offsets[] + byte data         */
+    GX_DECIMAL = 4/* DecimalBlock: fixed 40 B/value slices
+                     (polardbx-common DecimalTypeBase.DECIMAL_MEMORY_SIZE=40:
+                      9 x int32 base-1e9 words + integers/fractions/
+                      derivedFractions/isNeg bytes at offsets 36..39).
+                     The kernels compute on exact scaled-int64; the
+                     DEC<->scaled conversions mirror the reference's own
+                     "simple" fast path (DecimalBox.doAddToSum1/2 layout:
+                     w0=int (or w0=hi,w1=lo for 18 digits), then the frac
+                     word holding the fraction digits x 10^(9-f)). */
 } gx_type;
 
 typedef enum gx_mem {
@@ -239,9 +248,13 @@ typedef enum gx_proj_op {
     GX_PROJ_REV_F64 = 1,   /* out = a * (1 - b), doubles (Q3/Q9 revenue) */
     GX_PROJ_REV_SCALED4 = 2,/* out = a * (100 - b), i64 cents x hundredths ->
                               DECIMAL scale-4, exact */
-    GX_PROJ_Q9_AMOUNT4 = 3 /* out = a*(100-b) - c*d*100: Q9 amount =
+    GX_PROJ_Q9_AMOUNT4 = 3,/* out = a*(100-b) - c*d*100: Q9 amount =
                               extprice*(1-disc) - supplycost*qty, all i64
                               cents/hundredths -> DECIMAL scale-4, exact */
+    GX_PROJ_DEC_TO_SCALED = 4, /* a = GX_DECIMAL col (simple format),
+                                  c = scale -> exact scaled i64 */
+    GX_PROJ_SCALED_TO_DEC = 5  /* a = scaled i64 col, c = scale ->
+                                  GX_DECIMAL (DecimalBox simple layout) */
 } gx_proj_op;
 
 typedef struct gx_proj {

@@ -159,7 +159,7 @@ struct Column {
     std::vector<int32_t> i32v;
     std::vector<double>  f64v;
     std::vector<int32_t> off;   /* slice end-offsets (global) */
-    std::vector<uint8_t> bytes; /* slice data */
+    std::vector<uint8_t> bytes; /* slice data / decimal 40B slots */
     std::vector<uint8_t> null_; /* 1 = NULL */
 
     size_t size() const { return null_.size(); }
@@ -187,6 +187,10 @@ struct Column {
             bytes.insert(bytes.end(), b->data, b->data + blen);
             off.reserve(off.size() + n);
             for (int32_t i = 0; i < n; i++) off.push_back(base + b->offsets[i]);
+            break; }
+        case GX_DECIMAL: {
+            const uint8_t *src = (const uint8_t *)b->values;
+            bytes.insert(bytes.end(), src, src + (size_t)n * 40);
             break; }
         }
     }
@@ -292,6 +296,7 @@ struct OutCol {
         case GX_I32: i32v.push_back(0); break;
         case GX_F64: f64v.push_back(0); break;
         case GX_SLICE: off.push_back((int32_t)bytes.size()); break;
+        case GX_DECIMAL: bytes.resize(bytes.size() + 40, 0); break;
         }
     }
     void append_from(const Column &c, size_t i) {
@@ -306,9 +311,17 @@ struct OutCol {
             bytes.insert(bytes.end(), c.bytes.data() + b, c.bytes.data() + e);
             off.push_back((int32_t)bytes.size());
             break; }
+        case GX_DECIMAL:
+            bytes.insert(bytes.end(), c.bytes.data() + (size_t)i * 40,
+                         c.bytes.data() + (size_t)i * 40 + 40);
+            break;
         }
     }
     void append_i64(int64_t v) { null_.push_back(0); i64v.push_back(v); }
+    void append_dec40(const uint8_t *p) {
+        null_.push_back(0);
+        bytes.insert(bytes.end(), p, p + 40);
+    }
     void append_f64(double v)  { null_.push_back(0); f64v.push_back(v); }
 };
 
@@ -337,6 +350,9 @@ static gx_result *make_result(std::vector<OutCol> &&cols, int32_t n_rows) {
             b.offsets = oc.off.data();
             b.data = oc.bytes.data();
             b.data_len = (int64_t)oc.bytes.size();
+            break;
+        case GX_DECIMAL:
+            b.values = oc.bytes.data();
             break;
         }
     }
@@ -930,6 +946,59 @@ struct PartOp : gx_op {
     }
 };
 
+/* DecimalBox "simple" layout conversions (DecimalBox.java:43-71,
+ * doAddToSum1/2; DecimalTypeBase offsets 36..39). Exact for values with
+ * <= 18 integer digits and fraction scale <= 9. */
+static const int64_t POW10[19] = {1LL,10LL,100LL,1000LL,10000LL,100000LL,
+    1000000LL,10000000LL,100000000LL,1000000000LL,10000000000LL,
+    100000000000LL,1000000000000LL,10000000000000LL,100000000000000LL,
+    1000000000000000LL,10000000000000000LL,100000000000000000LL,
+    1000000000000000000LL};
+
+static inline int64_t dec40_to_scaled(const uint8_t *p, int scale) {
+    int32_t w[3];
+    std::memcpy(w, p, 12);
+    uint8_t integers = p[36];
+    uint8_t isneg = p[39];
+    int64_t ip, fr;
+    if (integers > 9) {
+        ip = (int64_t)w[0] * 1000000000LL + w[1];
+        fr = w[2];
+    } else {
+        ip = w[0];
+        fr = w[1];
+    }
+    /* frac word holds the fraction digits x 10^(9-scale) */
+    int64_t v = ip * POW10[scale] + fr / POW10[9 - scale];
+    return isneg ? -v : v;
+}
+
+static inline void scaled_to_dec40(int64_t v, int scale, uint8_t *p) {
+    std::memset(p, 0, 40);
+    uint8_t isneg = v < 0;
+    uint64_t a = isneg ? (uint64_t)(-v) : (uint64_t)v;
+    int64_t ip = (int64_t)(a / (uint64_t)POW10[scale]);
+    int64_t fr = (int64_t)(a % (uint64_t)POW10[scale]) * POW10[9 - scale];
+    int32_t w[3];
+    uint8_t integers;
+    if (ip >= 1000000000LL) {
+        w[0] = (int32_t)(ip / 1000000000LL);
+        w[1] = (int32_t)(ip % 1000000000LL);
+        w[2] = (int32_t)fr;
+        integers = 18;
+    } else {
+        w[0] = (int32_t)ip;
+        w[1] = (int32_t)fr;
+        w[2] = 0;
+        integers = 9;
+    }
+    std::memcpy(p, w, 12);
+    p[36] = integers;
+    p[37] = (uint8_t)scale;
+    p[38] = (uint8_t)scale;
+    p[39] = isneg;
+}
+
 /* ---- scan (vectorized filter + project) ---------------------------------
  * Restates the vectorized filter/projection stage (executor/vectorized/,
  * SURVEY.md §8f row 1): AND of predicates, SQL NULL-fails semantics,
@@ -955,6 +1024,7 @@ struct ScanOp : gx_op {
             switch (p.op) {
             case GX_PROJ_COPY: out_types.push_back(input_types[p.a]); break;
             case GX_PROJ_REV_F64: out_types.push_back(GX_F64); break;
+            case GX_PROJ_SCALED_TO_DEC: out_types.push_back(GX_DECIMAL); break;
             default: out_types.push_back(GX_I64); break;
             }
         }
@@ -1030,6 +1100,19 @@ struct ScanOp : gx_op {
                     if (a.is_null(r) || b.is_null(r)) cols[c].append_null();
                     else cols[c].append_i64((int64_t)((uint64_t)a.i64v[r] *
                                             (uint64_t)(100 - b.i64v[r])));
+                    break; }
+                case GX_PROJ_DEC_TO_SCALED: {
+                    if (a.is_null(r)) cols[c].append_null();
+                    else cols[c].append_i64(dec40_to_scaled(
+                        a.bytes.data() + (size_t)r * 40, pj.c));
+                    break; }
+                case GX_PROJ_SCALED_TO_DEC: {
+                    if (a.is_null(r)) cols[c].append_null();
+                    else {
+                        uint8_t buf[40];
+                        scaled_to_dec40(a.i64v[r], pj.c, buf);
+                        cols[c].append_dec40(buf);
+                    }
                     break; }
                 default: { /* Q9_AMOUNT4 */
                     const Column &b = in.cols[pj.b];
