@@ -109,9 +109,9 @@ def test_gpu_decimal_copy_through_scan():
     """DECIMAL survives a filtered COPY projection (the gather/staging path)."""
     hip = abi.load_hip()
     keys = np.arange(100, dtype=np.int64)
-    vals = np.array([dec40_encode(int(k) * 7 - 50, 2) for k in keys])
-    vals = np.frombuffer(b"".join(bytes(v) for v in vals),
-                         dtype=np.uint8).reshape(100, 40).copy()
+    vals = np.zeros((100, 40), dtype=np.uint8)
+    for k in range(100):
+        vals[k] = np.frombuffer(dec40_encode(k * 7 - 50, 2), dtype=np.uint8)
     chunk = Chunk([Block(I64, values=keys),
                    Block(DECIMAL, values=vals)])
     sc = ScanExec(hip, preds=[(0, abi.LT, 40)],
