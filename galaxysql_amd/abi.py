@@ -70,6 +70,7 @@ class GxJoinCfg(C.Structure):
         ("expected_build_rows", C.c_int64),
         ("n_out_proj", C.c_int32),
         ("out_proj", C.POINTER(C.c_int32)),
+        ("memory_budget_bytes", C.c_int64),
     ]
 
 

@@ -32,6 +32,7 @@
 
 #include <algorithm>
 #include <cstdlib>
+#include <deque>
 #include <hipcub/hipcub.hpp>
 
 thread_local std::string gx_err;
@@ -643,6 +644,7 @@ enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4 };
 
 struct gx_op {
     int kind;
+    int variant = 0;  /* OP_JOIN: 0 = in-HBM JoinOp, 1 = HybridJoinOp */
     int device;
     hipStream_t stream;
     virtual ~gx_op() = default;
@@ -1306,6 +1308,7 @@ struct JoinOp : gx_op {
 /* ======================= agg + partition in gxhip_agg.inc ============== */
 #include "gxhip_agg.inc"
 #include "gxhip_part.inc"
+#include "gxhip_hybrid.inc"
 #include "gxhip_scan.inc"
 
 /* ========================= C ABI ======================================= */
@@ -1319,22 +1322,40 @@ gx_op *gxop_join_create(const gx_join_cfg *cfg) {
     }
     if (cfg->device < 0) { gx_set_err("gxhip requires a GPU device"); return nullptr; }
     HIP_OK_NULL(hipSetDevice(cfg->device));
+    /* out-of-core decision (HybridHashJoinExec vs ParallelHashJoinExec):
+     * needs both a budget and a size hint; otherwise stay in-HBM */
+    if (cfg->memory_budget_bytes > 0 && cfg->expected_build_rows > 0) {
+        std::vector<int32_t> bt(cfg->build_outer ? cfg->outer_types
+                                                 : cfg->inner_types,
+                                (cfg->build_outer ? cfg->outer_types
+                                                  : cfg->inner_types) +
+                                (cfg->build_outer ? cfg->n_outer_cols
+                                                  : cfg->n_inner_cols));
+        int64_t est = cfg->expected_build_rows * join_row_bytes(bt);
+        if (est > cfg->memory_budget_bytes)
+            return new HybridJoinOp(cfg, hybrid_partition_count(
+                                             est, cfg->memory_budget_bytes));
+    }
     return new JoinOp(cfg);
 }
 int gxop_join_consume(gx_op *op, const gx_chunk *c) {
     if (!op || op->kind != OP_JOIN) { gx_set_err("not a join op"); return -1; }
+    if (op->variant == 1) return static_cast<HybridJoinOp *>(op)->consume(c);
     return static_cast<JoinOp *>(op)->consume(c);
 }
 int gxop_join_build(gx_op *op) {
     if (!op || op->kind != OP_JOIN) { gx_set_err("not a join op"); return -1; }
+    if (op->variant == 1) return static_cast<HybridJoinOp *>(op)->do_build();
     return static_cast<JoinOp *>(op)->do_build();
 }
 int gxop_join_probe(gx_op *op, const gx_chunk *c, gx_result **out) {
     if (!op || op->kind != OP_JOIN) { gx_set_err("not a join op"); return -1; }
+    if (op->variant == 1) return static_cast<HybridJoinOp *>(op)->probe(c, out);
     return static_cast<JoinOp *>(op)->probe(c, out);
 }
 int gxop_join_tail(gx_op *op, gx_result **out) {
     if (!op || op->kind != OP_JOIN) { gx_set_err("not a join op"); return -1; }
+    if (op->variant == 1) return static_cast<HybridJoinOp *>(op)->tail(out);
     return static_cast<JoinOp *>(op)->tail(out);
 }
 int gxop_join_close(gx_op *op) { delete op; return 0; }
@@ -1405,6 +1426,14 @@ int gxop_result_copy_col(const gx_result *res, int32_t col, void *dst_values,
 
 int gxop_join_get_stats(gx_op *op, gx_join_stats *out) {
     if (!op || op->kind != OP_JOIN || !out) { gx_set_err("not a join op"); return -1; }
+    if (op->variant == 1) {
+        HybridJoinOp *h = static_cast<HybridJoinOp *>(op);
+        out->probe_kernel_ms = h->probe_kernel_ms;
+        out->probe_launches = h->probe_launches;
+        out->probe_rows = h->probe_rows_total;
+        out->matches = h->matches_total;
+        return 0;
+    }
     JoinOp *j = static_cast<JoinOp *>(op);
     out->probe_kernel_ms = j->probe_kernel_ms;
     out->probe_launches = j->probe_launches;

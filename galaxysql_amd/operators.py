@@ -36,7 +36,8 @@ class ParallelHashJoinExec:
 
     def __init__(self, lib, join_type, join_keys, outer_types, inner_types,
                  max_one_row=False, build_outer=False, anti_null_col=-1,
-                 device=-1, stream=0, expected_build_rows=0, out_proj=None):
+                 device=-1, stream=0, expected_build_rows=0, out_proj=None,
+                 memory_budget_bytes=0):
         self._lib = lib
         self._keep = []
         keys = (GxEquiKey * len(join_keys))()
@@ -52,7 +53,8 @@ class ParallelHashJoinExec:
             n_inner_cols=len(inner_types), inner_types=it,
             anti_null_col=anti_null_col, device=device, stream=stream,
             expected_build_rows=expected_build_rows,
-            n_out_proj=len(out_proj or []), out_proj=op_arr)
+            n_out_proj=len(out_proj or []), out_proj=op_arr,
+            memory_budget_bytes=memory_budget_bytes)
         self._keep += [keys, ot, it, op_arr, cfg]
         self._op = lib.lib.gxop_join_create(C.byref(cfg))
         if not self._op:

@@ -139,6 +139,15 @@ typedef struct gx_join_cfg {
      * reference schema unchanged. */
     int32_t n_out_proj;
     const int32_t *out_proj;
+    /* out-of-core partitioned ("hybrid") join, modeled on
+     * HybridHashJoinExec (operator/HybridHashJoinExec.java): when
+     * expected_build_rows' estimated bytes exceed this budget, the library
+     * radix-partitions BOTH sides to host memory and joins partition by
+     * partition (probe() defers all output; gxop_join_tail drains it).
+     * 0 = unlimited (always in-HBM). The CN factory sets this from its
+     * MemoryAllocatorCtx the same way the reference picks
+     * HybridHashJoinExec over ParallelHashJoinExec at plan time. */
+    int64_t memory_budget_bytes;
 } gx_join_cfg;
 
 typedef struct gx_op gx_op;  /* opaque operator instance */
