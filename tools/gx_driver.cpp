@@ -20,6 +20,7 @@
 #include <cstdlib>
 #include <cstring>
 #include <cstdint>
+#include <algorithm>
 #include <chrono>
 #include <string>
 #include <vector>
@@ -405,13 +406,15 @@ static void t_scan(int device) {
     if (res2) {
         api.result_to_host(res2);
         CHECK(res2->chunk.n_rows == kept, "dec rows");
-        for (int r = 0; r < res2->chunk.n_rows; r++) {
-            if (cell_i64(res2, 0, r) != scaled[r]) {
-                CHECK(false, "decimal roundtrip row %d: %lld != %lld", r,
-                      (long long)cell_i64(res2, 0, r), (long long)scaled[r]);
-                break;
-            }
-        }
+        /* scan output order is nondeterministic (atomic emit staging) --
+         * the roundtrip check is on the sorted multisets */
+        std::vector<int64_t> back(kept);
+        for (int r = 0; r < res2->chunk.n_rows; r++)
+            back[r] = cell_i64(res2, 0, r);
+        std::vector<int64_t> want = scaled;
+        std::sort(back.begin(), back.end());
+        std::sort(want.begin(), want.end());
+        CHECK(back == want, "decimal roundtrip multiset mismatch");
         api.result_release(res2);
     }
     api.scan_close(op2);
