@@ -110,6 +110,25 @@ class GxAggStats(C.Structure):
     ]
 
 
+class GxGroupJoinCfg(C.Structure):
+    _fields_ = [
+        ("join_type", C.c_int32),
+        ("n_keys", C.c_int32),
+        ("keys", C.POINTER(GxEquiKey)),
+        ("n_build_cols", C.c_int32),
+        ("build_types", C.POINTER(C.c_int32)),
+        ("n_probe_cols", C.c_int32),
+        ("probe_types", C.POINTER(C.c_int32)),
+        ("n_group_cols", C.c_int32),
+        ("group_cols", C.POINTER(C.c_int32)),
+        ("n_aggs", C.c_int32),
+        ("aggs", C.POINTER(GxAggSpec)),
+        ("device", C.c_int32),
+        ("stream", C.c_uint64),
+        ("expected_build_rows", C.c_int64),
+    ]
+
+
 class GxPartCfg(C.Structure):
     _fields_ = [
         ("n_parts", C.c_int32),
@@ -184,6 +203,14 @@ class GxLib:
         L.gxop_agg_build.argtypes = [C.c_void_p]
         L.gxop_agg_next.argtypes = [C.c_void_p, C.POINTER(C.POINTER(GxResult))]
         L.gxop_agg_close.argtypes = [C.c_void_p]
+        L.gxop_groupjoin_create.restype = C.c_void_p
+        L.gxop_groupjoin_create.argtypes = [C.POINTER(GxGroupJoinCfg)]
+        L.gxop_groupjoin_consume.argtypes = [C.c_void_p, C.POINTER(GxChunk)]
+        L.gxop_groupjoin_build.argtypes = [C.c_void_p]
+        L.gxop_groupjoin_probe.argtypes = [C.c_void_p, C.POINTER(GxChunk)]
+        L.gxop_groupjoin_next.argtypes = [C.c_void_p,
+                                          C.POINTER(C.POINTER(GxResult))]
+        L.gxop_groupjoin_close.argtypes = [C.c_void_p]
         L.gxop_part_create.restype = C.c_void_p
         L.gxop_part_create.argtypes = [C.POINTER(GxPartCfg)]
         L.gxop_part_consume.argtypes = [C.c_void_p, C.POINTER(GxChunk),

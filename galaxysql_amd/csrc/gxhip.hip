@@ -33,6 +33,7 @@
 #include <algorithm>
 #include <cstdlib>
 #include <deque>
+#include <functional>
 #include <hipcub/hipcub.hpp>
 
 thread_local std::string gx_err;
@@ -176,9 +177,15 @@ __device__ static inline bool rows_key_equal(const KeyViews &a, int64_t i,
 /* ======================= generic small kernels ========================= */
 
 __global__ void k_hash_rows(KeyViews keys, int64_t n, int32_t *hashes,
-                            uint8_t *keynull) {
+                            uint8_t *keynull, int null_safe) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += (int64_t)gridDim.x * blockDim.x) {
+        if (null_safe) {
+            /* group-join: Chunk.equals matching -- null keys participate */
+            keynull[i] = 0;
+            hashes[i] = row_hash(keys, i);
+            continue;
+        }
         bool isnull = row_has_null_key(keys, i);
         keynull[i] = isnull;
         hashes[i] = isnull ? 0 : row_hash(keys, i);
@@ -640,7 +647,8 @@ static inline int gx_grid(int64_t n, int block = 256) {
 
 /* ========================= operator base =============================== */
 
-enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4 };
+enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4,
+              OP_GROUPJOIN = 5 };
 
 struct gx_op {
     int kind;
@@ -834,6 +842,11 @@ struct JoinOp : gx_op {
     bool built = false;
     bool pass_nothing = false, pass_through = false;
     bool tail_done = false;
+    /* group-join hooks: null-safe key matching, and a pair sink that
+     * replaces materialize() (receives staged probe chunk + pair lists) */
+    bool null_safe_keys = false;
+    std::function<int(const StagedChunk &, const uint32_t *,
+                      const uint32_t *, int64_t)> pair_sink;
 
     /* live probe-kernel stats for bench.py's roofline leg */
     hipEvent_t ev0 = nullptr, ev1 = nullptr;
@@ -887,7 +900,8 @@ struct JoinOp : gx_op {
     int do_build() {
         if (ensure_device(device)) return -1;
         const int64_t n = build.n_rows;
-        fast_i64 = (keys.size() == 1 && keys[0].unified_type == GX_I64);
+        fast_i64 = (keys.size() == 1 && keys[0].unified_type == GX_I64)
+                   && !null_safe_keys;
 
         bool semi_join = (cfg.join_type == GX_JOIN_SEMI || cfg.join_type == GX_JOIN_ANTI)
                          && !cfg.single_join;
@@ -910,7 +924,8 @@ struct JoinOp : gx_op {
                 d_keynull.grow((size_t)n, stream)) return -1;
             KeyViews bk = key_views(build, build_key_cols);
             hipLaunchKernelGGL(k_hash_rows, dim3(gx_grid(n)), dim3(256), 0, stream,
-                               bk, n, (int32_t *)d_hashes.p, (uint8_t *)d_keynull.p);
+                               bk, n, (int32_t *)d_hashes.p, (uint8_t *)d_keynull.p,
+                               (int)null_safe_keys);
 
             n_buckets = gx_pow2(n * 2); /* avg load 0.5 */
             mask = (uint32_t)(n_buckets - 1);
@@ -1089,7 +1104,8 @@ struct JoinOp : gx_op {
             return -1;
         KeyViews pk = key_views_staged(probe_st, probe_key_cols);
         hipLaunchKernelGGL(k_hash_rows, dim3(gx_grid(n)), dim3(256), 0, stream,
-                           pk, n, (int32_t *)d_ph.p, (uint8_t *)d_pn.p);
+                           pk, n, (int32_t *)d_ph.p, (uint8_t *)d_pn.p,
+                           (int)null_safe_keys);
 
         int rc = -1;
         do {
@@ -1236,6 +1252,14 @@ struct JoinOp : gx_op {
                 if (meta[0] <= cap) {
                     probe_rows_total += n;
                     matches_total += meta[0];
+                    if (pair_sink) {
+                        if (pair_sink(probe_st, (const uint32_t *)d_pidx.p,
+                                      (const uint32_t *)d_bpos.p,
+                                      (int64_t)meta[0])) { attempt = 99; break; }
+                        HIP_OK(hipStreamSynchronize(stream));
+                        rc = 0;
+                        break;
+                    }
                     HipResult *h = materialize(probe_st, (uint32_t *)d_pidx.p,
                                                (uint32_t *)d_bpos.p, meta[0]);
                     if (!h) { attempt = 99; break; }
@@ -1309,6 +1333,7 @@ struct JoinOp : gx_op {
 #include "gxhip_agg.inc"
 #include "gxhip_part.inc"
 #include "gxhip_hybrid.inc"
+#include "gxhip_groupjoin.inc"
 #include "gxhip_scan.inc"
 
 /* ========================= C ABI ======================================= */

@@ -173,6 +173,96 @@ class HashAggExec:
             pass
 
 
+class HashGroupJoinExec:
+    """Fused join+agg (HashGroupJoinExec.java:186-447): consume the GROUP
+    side, then feed probe chunks; one output row per (matched) consumed
+    row: group_cols then agg values. join_keys: EquiJoinKey with
+    outer_index = consumed-side col, inner_index = probe-side col."""
+
+    def __init__(self, lib, join_type, join_keys, build_types, probe_types,
+                 group_cols, aggs, device=-1, stream=0,
+                 expected_build_rows=0):
+        from .abi import GxGroupJoinCfg
+        self._lib = lib
+        self._keep = []
+        keys = (GxEquiKey * len(join_keys))()
+        for i, k in enumerate(join_keys):
+            keys[i] = GxEquiKey(k.outer_index, k.inner_index, k.unified_type, 0)
+        bt = (C.c_int32 * len(build_types))(*build_types)
+        pt = (C.c_int32 * len(probe_types))(*probe_types)
+        gc = (C.c_int32 * max(1, len(group_cols)))(*(group_cols or [0]))
+        sp = (GxAggSpec * max(1, len(aggs)))()
+        for i, (f, col) in enumerate(aggs):
+            sp[i] = GxAggSpec(f, col)
+        cfg = GxGroupJoinCfg(
+            join_type=join_type, n_keys=len(join_keys), keys=keys,
+            n_build_cols=len(build_types), build_types=bt,
+            n_probe_cols=len(probe_types), probe_types=pt,
+            n_group_cols=len(group_cols), group_cols=gc,
+            n_aggs=len(aggs), aggs=sp, device=device, stream=stream,
+            expected_build_rows=expected_build_rows)
+        self._keep += [keys, bt, pt, gc, sp, cfg]
+        self._op = lib.lib.gxop_groupjoin_create(C.byref(cfg))
+        if not self._op:
+            raise RuntimeError(f"gxop_groupjoin_create: {lib.error()}")
+
+    def consume_chunk(self, chunk: Chunk):
+        ka = []
+        gc = self._lib.to_gx_chunk(chunk, ka)
+        self._lib.check(
+            self._lib.lib.gxop_groupjoin_consume(self._op, C.byref(gc)),
+            "groupjoin_consume")
+
+    def build_consume(self):
+        self._lib.check(self._lib.lib.gxop_groupjoin_build(self._op),
+                        "groupjoin_build")
+
+    def probe_chunk(self, chunk: Chunk):
+        ka = []
+        gc = self._lib.to_gx_chunk(chunk, ka)
+        self._lib.check(
+            self._lib.lib.gxop_groupjoin_probe(self._op, C.byref(gc)),
+            "groupjoin_probe")
+
+    def result_chunks(self):
+        chunks = []
+        while True:
+            out = C.POINTER(GxResult)()
+            self._lib.check(
+                self._lib.lib.gxop_groupjoin_next(self._op, C.byref(out)),
+                "groupjoin_next")
+            if not out:
+                break
+            chunks.append(self._lib.result_to_chunk(out))
+        return chunks
+
+    def close(self):
+        if self._op:
+            self._lib.lib.gxop_groupjoin_close(self._op)
+            self._op = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+def run_groupjoin(lib, join_type, join_keys, build_chunks, probe_chunks,
+                  build_types, probe_types, group_cols, aggs, **kw):
+    op = HashGroupJoinExec(lib, join_type, join_keys, build_types,
+                           probe_types, group_cols, aggs, **kw)
+    try:
+        for ch in build_chunks:
+            op.consume_chunk(ch)
+        op.build_consume()
+        for ch in probe_chunks:
+            op.probe_chunk(ch)
+        return op.result_chunks()
+    finally:
+        op.close()
+
+
 class PartitioningExchanger:
     """Mirrors mpp/operator/PartitioningExchanger.java:71-134."""
 

@@ -206,6 +206,53 @@ int gxop_agg_build(gx_op *op);
 int gxop_agg_next(gx_op *op, gx_result **out);
 int gxop_agg_close(gx_op *op);
 
+/* ---- fused group-join -------------------------------------------------- */
+
+/* HashGroupJoinExec (operator/HashGroupJoinExec.java:186-447): join + agg
+ * in one operator. The CONSUMED side's rows are the groups — one group per
+ * consumed row position (buildOneChunk:296-311 appendInitValue per row),
+ * keyed by the equi-join keys; every matching probe row accumulates into
+ * that position's aggregators (buildJoinRow:469-490). Output: group-key
+ * columns picked from the consumed row (groups[]) then one column per
+ * aggregator; INNER emits matched positions only, LEFT emits every
+ * position with one null-row accumulation for unmatched ones
+ * (buildNullRow via doNextChunk:324-330 -> COUNT(*)=1, COUNT(col)=0,
+ * SUM/MIN/MAX=NULL). single_join is not supported here (the reference
+ * only uses it for scalar-subquery plans, not the group-join path).
+ * NOTE DESIGN.md "group-join emission pairing": the reference's
+ * buildValueChunks pairs value slots with group keys through two counters
+ * that only agree when first-match order is position order
+ * (HashGroupJoinExec.java:410-451); we emit the self-consistent pairing
+ * (each group's own values) — identical whenever the reference is. */
+typedef struct gx_groupjoin_cfg {
+    int32_t join_type;          /* GX_JOIN_INNER or GX_JOIN_LEFT */
+    int32_t n_keys;
+    const gx_equi_key *keys;    /* outer_index -> consumed/group side cols,
+                                   inner_index -> probe side cols */
+    int32_t n_build_cols;       /* the consumed ("outer"/group) side */
+    const int32_t *build_types;
+    int32_t n_probe_cols;
+    const int32_t *probe_types;
+    int32_t n_group_cols;       /* groups[]: output key cols, indexes into
+                                   the consumed side */
+    const int32_t *group_cols;
+    int32_t n_aggs;
+    const gx_agg_spec *aggs;    /* input_col indexes into the PROBE side */
+    int32_t device;
+    uint64_t stream;
+    int64_t expected_build_rows;
+} gx_groupjoin_cfg;
+
+gx_op *gxop_groupjoin_create(const gx_groupjoin_cfg *cfg);
+/* consume one GROUP-side chunk */
+int gxop_groupjoin_consume(gx_op *op, const gx_chunk *chunk);
+int gxop_groupjoin_build(gx_op *op);
+/* feed one probe chunk (accumulates; no output) */
+int gxop_groupjoin_probe(gx_op *op, const gx_chunk *chunk);
+/* emit result batches after the last probe chunk; *out=NULL when done */
+int gxop_groupjoin_next(gx_op *op, gx_result **out);
+int gxop_groupjoin_close(gx_op *op);
+
 /* ---- partition exchange ------------------------------------------------ */
 
 typedef struct gx_part_cfg {

@@ -365,7 +365,8 @@ static gx_result *make_result(std::vector<OutCol> &&cols, int32_t n_rows) {
 
 /* ---- operator base ----------------------------------------------------- */
 
-enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4 };
+enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4,
+              OP_GROUPJOIN = 5 };
 
 } // anonymous namespace
 
@@ -880,6 +881,195 @@ struct AggOp : gx_op {
     }
 };
 
+
+/* ---- fused group-join ---------------------------------------------------
+ * Restates HashGroupJoinExec (operator/HashGroupJoinExec.java): the
+ * CONSUMED side's rows are the groups (one per position, buildOneChunk
+ * :296-311 puts EVERY row — no null-key skip, unlike ExecUtils
+ * .buildOneChunk — and matching goes through Chunk.equals, so NULL keys
+ * match NULL keys, null-safe); each matching probe row accumulates into
+ * that position's aggregators (buildJoinRow:469-490 -> doAggregate).
+ * INNER emits matched positions; LEFT emits all, unmatched ones after one
+ * null-row accumulation (buildNullRow via doNextChunk:324-330): COUNT(*)
+ * counts the null row, COUNT(col)/SUM/MIN/MAX see NULL. Emission pairing
+ * is each group's own values (see include/gxop.h note on the reference's
+ * buildValueChunks counter mismatch, HashGroupJoinExec.java:410-451). */
+struct GroupJoinOp : gx_op {
+    gx_groupjoin_cfg cfg;
+    std::vector<gx_equi_key> keys;
+    std::vector<int32_t> build_types, probe_types, group_cols_;
+    std::vector<gx_agg_spec> aggs;
+    Store build;
+    std::vector<int> build_key_cols, probe_key_cols;
+    std::vector<int32_t> table, links;
+    int32_t mask = 0;
+    std::vector<uint8_t> used;
+    struct AggState { std::vector<int64_t> i64; std::vector<double> f64;
+                      std::vector<uint8_t> isnull; };
+    std::vector<AggState> states;
+    bool built = false, pass_nothing = false;
+    size_t emit_cursor = 0;
+    static constexpr int32_t CHUNK_SIZE = 1000;
+
+    GroupJoinOp(const gx_groupjoin_cfg *c) : gx_op(OP_GROUPJOIN), cfg(*c) {
+        keys.assign(c->keys, c->keys + c->n_keys);
+        build_types.assign(c->build_types, c->build_types + c->n_build_cols);
+        probe_types.assign(c->probe_types, c->probe_types + c->n_probe_cols);
+        group_cols_.assign(c->group_cols, c->group_cols + c->n_group_cols);
+        aggs.assign(c->aggs, c->aggs + c->n_aggs);
+        for (auto &k : keys) {
+            build_key_cols.push_back(k.outer_index);  /* consumed = "outer" */
+            probe_key_cols.push_back(k.inner_index);
+        }
+        build.init((int32_t)build_types.size(), build_types.data());
+    }
+
+    int consume(const gx_chunk *ch) { return build.append(ch); }
+
+    int do_build() {
+        const int64_t size = (int64_t)build.n_rows;
+        if (size == 0 && cfg.join_type == GX_JOIN_INNER) pass_nothing = true;
+        int64_t n = hc_array_size(std::max<int64_t>(size, 1),
+                                  join_load_factor(size));
+        mask = (int32_t)(n - 1);
+        table.assign((size_t)n, -1);
+        links.assign((size_t)size, -1);
+        for (size_t pos = 0; pos < build.n_rows; pos++) {
+            /* every position inserted; nulls hash per Block.hashCode */
+            int32_t h = build.row_hash(pos, build_key_cols);
+            int32_t slot = hc_mix(h) & mask;
+            links[pos] = table[slot];
+            table[slot] = (int32_t)pos;
+        }
+        used.assign(build.n_rows, 0);
+        states.resize(aggs.size());
+        for (size_t a = 0; a < aggs.size(); a++) {
+            AggState &s = states[a];
+            s.i64.assign(build.n_rows, 0);
+            s.f64.assign(build.n_rows, 0);
+            bool init_null = !(aggs[a].func == GX_AGG_COUNT_ROW ||
+                               aggs[a].func == GX_AGG_COUNT_COL ||
+                               aggs[a].func == GX_AGG_SUM_I64);
+            s.isnull.assign(build.n_rows, init_null ? 1 : 0);
+            if (aggs[a].func == GX_AGG_MIN_I64)
+                s.i64.assign(build.n_rows, INT64_MAX);
+            if (aggs[a].func == GX_AGG_MAX_I64)
+                s.i64.assign(build.n_rows, INT64_MIN);
+        }
+        built = true;
+        return 0;
+    }
+
+    void accumulate(size_t a, int32_t gid, const Store &in, size_t row) {
+        const gx_agg_spec &sp = aggs[a];
+        AggState &s = states[a];
+        const Column *c = sp.input_col >= 0 ? &in.cols[sp.input_col] : nullptr;
+        switch (sp.func) {
+        case GX_AGG_COUNT_ROW: s.i64[gid]++; break;
+        case GX_AGG_COUNT_COL: if (!c->is_null(row)) s.i64[gid]++; break;
+        case GX_AGG_SUM_I64:
+            if (!c->is_null(row))
+                s.i64[gid] = (int64_t)((uint64_t)s.i64[gid] +
+                    (uint64_t)(c->type == GX_I32 ? (int64_t)c->i32v[row]
+                                                 : c->i64v[row]));
+            break;
+        case GX_AGG_SUM_F64:
+            if (!c->is_null(row)) {
+                double v = c->type == GX_F64 ? c->f64v[row]
+                         : c->type == GX_I32 ? (double)c->i32v[row]
+                                             : (double)c->i64v[row];
+                if (s.isnull[gid]) { s.f64[gid] = v; s.isnull[gid] = 0; }
+                else s.f64[gid] += v;
+            }
+            break;
+        case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            if (!c->is_null(row)) {
+                int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[row]
+                                              : c->i64v[row];
+                if (s.isnull[gid]) { s.i64[gid] = v; s.isnull[gid] = 0; }
+                else s.i64[gid] = sp.func == GX_AGG_MIN_I64
+                        ? std::min(s.i64[gid], v) : std::max(s.i64[gid], v);
+            }
+            break;
+        case GX_AGG_MIN_F64: case GX_AGG_MAX_F64:
+            if (!c->is_null(row)) {
+                double v = c->f64v[row];
+                if (s.isnull[gid]) { s.f64[gid] = v; s.isnull[gid] = 0; }
+                else s.f64[gid] = sp.func == GX_AGG_MIN_F64
+                        ? std::min(s.f64[gid], v) : std::max(s.f64[gid], v);
+            }
+            break;
+        }
+    }
+
+    int probe(const gx_chunk *ch) {
+        if (!built) { set_err("probe before build"); return -1; }
+        if (pass_nothing) return 0;
+        Store in;
+        in.init((int32_t)probe_types.size(), probe_types.data());
+        if (in.append(ch) != 0) return -1;
+        for (size_t r = 0; r < in.n_rows; r++) {
+            int32_t h = in.row_hash(r, probe_key_cols);
+            for (int32_t m = table[(size_t)(hc_mix(h) & mask)]; m != -1;
+                 m = links[(size_t)m]) {
+                if (!build.keys_equal((size_t)m, in, r, build_key_cols,
+                                      probe_key_cols))
+                    continue;
+                used[(size_t)m] = 1;
+                for (size_t a = 0; a < aggs.size(); a++)
+                    accumulate(a, m, in, r);
+            }
+        }
+        return 0;
+    }
+
+    int next(gx_result **out) {
+        *out = nullptr;
+        if (pass_nothing) return 0;
+        std::vector<int32_t> otypes;
+        for (int32_t gc : group_cols_) otypes.push_back(build_types[gc]);
+        for (auto &sp : aggs) {
+            switch (sp.func) {
+            case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
+            case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+                otypes.push_back(GX_I64); break;
+            default: otypes.push_back(GX_F64); break;
+            }
+        }
+        std::vector<OutCol> cols(otypes.size());
+        for (size_t c = 0; c < otypes.size(); c++) cols[c].type = otypes[c];
+        int32_t n = 0;
+        const bool emit_all = cfg.join_type == GX_JOIN_LEFT;
+        while (emit_cursor < build.n_rows && n < CHUNK_SIZE) {
+            size_t g = emit_cursor++;
+            bool matched = used[g] != 0;
+            if (!matched && !emit_all) continue;
+            size_t col = 0;
+            for (int32_t gc : group_cols_)
+                cols[col++].append_from(build.cols[gc], g);
+            for (size_t a = 0; a < aggs.size(); a++) {
+                AggState &s = states[a];
+                int64_t i64v = s.i64[g];
+                bool nullv = s.isnull[g] != 0;
+                if (!matched && aggs[a].func == GX_AGG_COUNT_ROW)
+                    i64v += 1;  /* buildNullRow: null row counts */
+                if (nullv && !(aggs[a].func == GX_AGG_COUNT_ROW ||
+                               aggs[a].func == GX_AGG_COUNT_COL ||
+                               aggs[a].func == GX_AGG_SUM_I64))
+                    cols[col++].append_null();
+                else if (otypes[group_cols_.size() + a] == GX_I64)
+                    cols[col++].append_i64(i64v);
+                else
+                    cols[col++].append_f64(s.f64[g]);
+            }
+            n++;
+        }
+        if (n == 0) return 0;
+        *out = make_result(std::move(cols), n);
+        return 0;
+    }
+};
+
 /* ---- partition operator -------------------------------------------------
  * Restates PartitioningExchanger.consumeChunk (mpp/operator/
  * PartitioningExchanger.java:71-134): row hash over key cols (HashBucketFunction
@@ -1180,6 +1370,33 @@ int gxop_agg_next(gx_op *op, gx_result **out) {
     return static_cast<AggOp *>(op)->next(out);
 }
 int gxop_agg_close(gx_op *op) { delete op; return 0; }
+
+
+gx_op *gxop_groupjoin_create(const gx_groupjoin_cfg *cfg) {
+    if (!cfg || cfg->n_keys <= 0 ||
+        (cfg->join_type != GX_JOIN_INNER && cfg->join_type != GX_JOIN_LEFT)) {
+        set_err("bad groupjoin cfg");
+        return nullptr;
+    }
+    return new GroupJoinOp(cfg);
+}
+int gxop_groupjoin_consume(gx_op *op, const gx_chunk *c) {
+    if (!op || op->kind != OP_GROUPJOIN) { set_err("not a groupjoin op"); return -1; }
+    return static_cast<GroupJoinOp *>(op)->consume(c);
+}
+int gxop_groupjoin_build(gx_op *op) {
+    if (!op || op->kind != OP_GROUPJOIN) { set_err("not a groupjoin op"); return -1; }
+    return static_cast<GroupJoinOp *>(op)->do_build();
+}
+int gxop_groupjoin_probe(gx_op *op, const gx_chunk *c) {
+    if (!op || op->kind != OP_GROUPJOIN) { set_err("not a groupjoin op"); return -1; }
+    return static_cast<GroupJoinOp *>(op)->probe(c);
+}
+int gxop_groupjoin_next(gx_op *op, gx_result **out) {
+    if (!op || op->kind != OP_GROUPJOIN) { set_err("not a groupjoin op"); return -1; }
+    return static_cast<GroupJoinOp *>(op)->next(out);
+}
+int gxop_groupjoin_close(gx_op *op) { delete op; return 0; }
 
 gx_op *gxop_part_create(const gx_part_cfg *cfg) {
     if (!cfg || cfg->n_parts <= 0) { set_err("bad part cfg"); return nullptr; }
