@@ -278,6 +278,39 @@ int gxop_part_consume_concat(gx_op *op, const gx_chunk *chunk,
                              gx_result **out, int64_t *counts);
 int gxop_part_close(gx_op *op);
 
+/* ---- window: running aggregates over partition-sorted input ------------ */
+
+/* NonFrameOverWindowExec (operator/NonFrameOverWindowExec.java:34-160),
+ * the reference's stand-in for the north_star's HashWindowExec (absent in
+ * this snapshot — SURVEY.md §8f row 4): input arrives sorted by the
+ * PARTITION BY columns; each window function emits, per row, its running
+ * value from the partition start through the current row (ROWS UNBOUNDED
+ * PRECEDING .. CURRENT ROW), resetting at every partition change
+ * (isDifferentPartition:136-145, null-safe equality). reset[a]=1 gives
+ * the CURRENT ROW .. CURRENT ROW mode (resetAccumulators). ROW_NUMBER()
+ * is GX_AGG_COUNT_ROW cumulative (identical by definition). Output chunk
+ * = the input columns followed by one column per window function
+ * (buildResultChunk:123-134). The operator is STREAMING: each consume
+ * returns that chunk's rows; partition carry state (last partition key +
+ * running accumulators) lives inside the op across chunks. Frame-based
+ * windows (OverWindowFramesExec) are out of scope this round. */
+typedef struct gx_window_cfg {
+    int32_t n_part_cols;
+    const int32_t *part_cols;
+    int32_t n_aggs;
+    const gx_agg_spec *aggs;   /* window functions over the agg subset */
+    const uint8_t *reset;      /* per agg: 1 = CURRENT ROW..CURRENT ROW */
+    int32_t n_input_cols;
+    const int32_t *input_types;
+    int32_t device;
+    uint64_t stream;
+} gx_window_cfg;
+
+gx_op *gxop_window_create(const gx_window_cfg *cfg);
+/* process one partition-sorted chunk; *out = input cols + window cols */
+int gxop_window_consume(gx_op *op, const gx_chunk *chunk, gx_result **out);
+int gxop_window_close(gx_op *op);
+
 /* ---- scan: vectorized filter + project --------------------------------- */
 
 /* Mirrors the vectorized filter/projection stage (executor/vectorized/,

@@ -366,7 +366,7 @@ static gx_result *make_result(std::vector<OutCol> &&cols, int32_t n_rows) {
 /* ---- operator base ----------------------------------------------------- */
 
 enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4,
-              OP_GROUPJOIN = 5 };
+              OP_GROUPJOIN = 5, OP_WINDOW = 6 };
 
 } // anonymous namespace
 
@@ -1070,6 +1070,157 @@ struct GroupJoinOp : gx_op {
     }
 };
 
+
+/* ---- window: running aggregates over partition-sorted input -------------
+ * Restates NonFrameOverWindowExec (operator/NonFrameOverWindowExec.java:
+ * 34-160): per row, accumulate into ONE running aggregator slot, resetting
+ * when the partition-key row differs from the previous row
+ * (isDifferentPartition:136-145, null-safe) or when reset[a] is set
+ * (resetAccumulators / CURRENT ROW mode); emit the running value after
+ * each accumulate (processFirstLine/doNextChunk:80-120). Streaming: carry
+ * (last partition key + running state) survives across chunks. */
+struct WindowOp : gx_op {
+    gx_window_cfg cfg;
+    std::vector<int32_t> part_cols_, input_types;
+    std::vector<gx_agg_spec> aggs;
+    std::vector<uint8_t> reset_;
+
+    bool have_carry = false;
+    Store carry_key;                 /* 1 row: last partition key cols */
+    struct RunState { int64_t i64 = 0; double f64 = 0; uint8_t isnull = 0; };
+    std::vector<RunState> run;
+
+    WindowOp(const gx_window_cfg *c) : gx_op(OP_WINDOW), cfg(*c) {
+        part_cols_.assign(c->part_cols, c->part_cols + c->n_part_cols);
+        input_types.assign(c->input_types, c->input_types + c->n_input_cols);
+        aggs.assign(c->aggs, c->aggs + c->n_aggs);
+        reset_.assign(c->reset, c->reset + c->n_aggs);
+        std::vector<int32_t> kt;
+        for (int32_t pc : part_cols_) kt.push_back(input_types[pc]);
+        carry_key.init((int32_t)kt.size(), kt.data());
+        run.resize(aggs.size());
+    }
+
+    void reset_state(size_t a) {
+        RunState &s = run[a];
+        switch (aggs[a].func) {
+        case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL: case GX_AGG_SUM_I64:
+            s.i64 = 0; s.isnull = 0; break;
+        default: s.i64 = 0; s.f64 = 0; s.isnull = 1; break;
+        }
+    }
+
+    void acc(size_t a, const Store &in, size_t r) {
+        const gx_agg_spec &sp = aggs[a];
+        RunState &s = run[a];
+        const Column *c = sp.input_col >= 0 ? &in.cols[sp.input_col] : nullptr;
+        switch (sp.func) {
+        case GX_AGG_COUNT_ROW: s.i64++; break;
+        case GX_AGG_COUNT_COL: if (!c->is_null(r)) s.i64++; break;
+        case GX_AGG_SUM_I64:
+            if (!c->is_null(r))
+                s.i64 = (int64_t)((uint64_t)s.i64 + (uint64_t)(
+                    c->type == GX_I32 ? (int64_t)c->i32v[r] : c->i64v[r]));
+            break;
+        case GX_AGG_SUM_F64:
+            if (!c->is_null(r)) {
+                double v = c->type == GX_F64 ? c->f64v[r]
+                         : c->type == GX_I32 ? (double)c->i32v[r]
+                                             : (double)c->i64v[r];
+                if (s.isnull) { s.f64 = v; s.isnull = 0; } else s.f64 += v;
+            }
+            break;
+        case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            if (!c->is_null(r)) {
+                int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[r]
+                                              : c->i64v[r];
+                if (s.isnull) { s.i64 = v; s.isnull = 0; }
+                else s.i64 = sp.func == GX_AGG_MIN_I64 ? std::min(s.i64, v)
+                                                       : std::max(s.i64, v);
+            }
+            break;
+        case GX_AGG_MIN_F64: case GX_AGG_MAX_F64:
+            if (!c->is_null(r)) {
+                double v = c->f64v[r];
+                if (s.isnull) { s.f64 = v; s.isnull = 0; }
+                else s.f64 = sp.func == GX_AGG_MIN_F64 ? std::min(s.f64, v)
+                                                       : std::max(s.f64, v);
+            }
+            break;
+        }
+    }
+
+    bool same_partition(const Store &in, size_t r) {
+        if (!have_carry) return false;
+        std::vector<int> kidx;
+        for (size_t k = 0; k < part_cols_.size(); k++) kidx.push_back((int)k);
+        std::vector<int> pcols(part_cols_.begin(), part_cols_.end());
+        return carry_key.keys_equal(0, in, r, kidx, pcols);
+    }
+
+    void save_carry(const Store &in, size_t r) {
+        for (size_t k = 0; k < part_cols_.size(); k++) {
+            Column &dst = carry_key.cols[k];
+            const Column &src = in.cols[part_cols_[k]];
+            dst.i64v.clear(); dst.i32v.clear(); dst.f64v.clear();
+            dst.off.clear(); dst.bytes.clear(); dst.null_.clear();
+            dst.null_.push_back(src.is_null(r) ? 1 : 0);
+            switch (src.type) {
+            case GX_I64: dst.i64v.push_back(src.is_null(r) ? 0 : src.i64v[r]); break;
+            case GX_I32: dst.i32v.push_back(src.is_null(r) ? 0 : src.i32v[r]); break;
+            case GX_F64: dst.f64v.push_back(src.is_null(r) ? 0 : src.f64v[r]); break;
+            case GX_SLICE: {
+                if (!src.is_null(r)) {
+                    int32_t b = src.begin_off(r), e = src.off[r];
+                    dst.bytes.assign(src.bytes.data() + b, src.bytes.data() + e);
+                }
+                dst.off.push_back((int32_t)dst.bytes.size());
+                break; }
+            }
+        }
+        carry_key.n_rows = 1;
+        have_carry = true;
+    }
+
+    int consume(const gx_chunk *ch, gx_result **out) {
+        *out = nullptr;
+        Store in;
+        in.init((int32_t)input_types.size(), input_types.data());
+        if (in.append(ch) != 0) return -1;
+
+        std::vector<int32_t> otypes = input_types;
+        for (auto &sp : aggs) {
+            switch (sp.func) {
+            case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
+            case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+                otypes.push_back(GX_I64); break;
+            default: otypes.push_back(GX_F64); break;
+            }
+        }
+        std::vector<OutCol> cols(otypes.size());
+        for (size_t c = 0; c < otypes.size(); c++) cols[c].type = otypes[c];
+
+        for (size_t r = 0; r < in.n_rows; r++) {
+            bool change = !same_partition(in, r);
+            if (change) save_carry(in, r);
+            for (size_t c = 0; c < input_types.size(); c++)
+                cols[c].append_from(in.cols[c], r);
+            for (size_t a = 0; a < aggs.size(); a++) {
+                if (reset_[a] || change) reset_state(a);
+                acc(a, in, r);
+                size_t col = input_types.size() + a;
+                RunState &s = run[a];
+                bool i64out = otypes[col] == GX_I64;
+                if (s.isnull) cols[col].append_null();
+                else if (i64out) cols[col].append_i64(s.i64);
+                else cols[col].append_f64(s.f64);
+            }
+        }
+        *out = make_result(std::move(cols), (int32_t)in.n_rows);
+        return 0;
+    }
+};
+
 /* ---- partition operator -------------------------------------------------
  * Restates PartitioningExchanger.consumeChunk (mpp/operator/
  * PartitioningExchanger.java:71-134): row hash over key cols (HashBucketFunction
@@ -1397,6 +1548,17 @@ int gxop_groupjoin_next(gx_op *op, gx_result **out) {
     return static_cast<GroupJoinOp *>(op)->next(out);
 }
 int gxop_groupjoin_close(gx_op *op) { delete op; return 0; }
+
+
+gx_op *gxop_window_create(const gx_window_cfg *cfg) {
+    if (!cfg || cfg->n_aggs <= 0) { set_err("bad window cfg"); return nullptr; }
+    return new WindowOp(cfg);
+}
+int gxop_window_consume(gx_op *op, const gx_chunk *c, gx_result **out) {
+    if (!op || op->kind != OP_WINDOW) { set_err("not a window op"); return -1; }
+    return static_cast<WindowOp *>(op)->consume(c, out);
+}
+int gxop_window_close(gx_op *op) { delete op; return 0; }
 
 gx_op *gxop_part_create(const gx_part_cfg *cfg) {
     if (!cfg || cfg->n_parts <= 0) { set_err("bad part cfg"); return nullptr; }
