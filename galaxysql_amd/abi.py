@@ -129,6 +129,20 @@ class GxGroupJoinCfg(C.Structure):
     ]
 
 
+class GxWindowCfg(C.Structure):
+    _fields_ = [
+        ("n_part_cols", C.c_int32),
+        ("part_cols", C.POINTER(C.c_int32)),
+        ("n_aggs", C.c_int32),
+        ("aggs", C.POINTER(GxAggSpec)),
+        ("reset", C.POINTER(C.c_uint8)),
+        ("n_input_cols", C.c_int32),
+        ("input_types", C.POINTER(C.c_int32)),
+        ("device", C.c_int32),
+        ("stream", C.c_uint64),
+    ]
+
+
 class GxPartCfg(C.Structure):
     _fields_ = [
         ("n_parts", C.c_int32),
@@ -211,6 +225,11 @@ class GxLib:
         L.gxop_groupjoin_next.argtypes = [C.c_void_p,
                                           C.POINTER(C.POINTER(GxResult))]
         L.gxop_groupjoin_close.argtypes = [C.c_void_p]
+        L.gxop_window_create.restype = C.c_void_p
+        L.gxop_window_create.argtypes = [C.POINTER(GxWindowCfg)]
+        L.gxop_window_consume.argtypes = [C.c_void_p, C.POINTER(GxChunk),
+                                          C.POINTER(C.POINTER(GxResult))]
+        L.gxop_window_close.argtypes = [C.c_void_p]
         L.gxop_part_create.restype = C.c_void_p
         L.gxop_part_create.argtypes = [C.POINTER(GxPartCfg)]
         L.gxop_part_consume.argtypes = [C.c_void_p, C.POINTER(GxChunk),

@@ -263,6 +263,65 @@ def run_groupjoin(lib, join_type, join_keys, build_chunks, probe_chunks,
         op.close()
 
 
+class NonFrameOverWindowExec:
+    """Running window aggregates over partition-sorted input
+    (NonFrameOverWindowExec.java:34-160). aggs: (abi.COUNT_ROW.., col);
+    reset[a]=True = CURRENT ROW..CURRENT ROW mode. ROW_NUMBER() =
+    (abi.COUNT_ROW, -1) cumulative."""
+
+    def __init__(self, lib, part_cols, aggs, input_types, reset=None,
+                 device=-1, stream=0):
+        from .abi import GxWindowCfg
+        self._lib = lib
+        self._keep = []
+        pc = (C.c_int32 * max(1, len(part_cols)))(*(part_cols or [0]))
+        sp = (GxAggSpec * max(1, len(aggs)))()
+        for i, (f, col) in enumerate(aggs):
+            sp[i] = GxAggSpec(f, col)
+        rs = (C.c_uint8 * max(1, len(aggs)))(
+            *[1 if (reset and reset[i]) else 0 for i in range(len(aggs))])
+        it = (C.c_int32 * len(input_types))(*input_types)
+        cfg = GxWindowCfg(n_part_cols=len(part_cols), part_cols=pc,
+                          n_aggs=len(aggs), aggs=sp, reset=rs,
+                          n_input_cols=len(input_types), input_types=it,
+                          device=device, stream=stream)
+        self._keep += [pc, sp, rs, it, cfg]
+        self._op = lib.lib.gxop_window_create(C.byref(cfg))
+        if not self._op:
+            raise RuntimeError(f"gxop_window_create: {lib.error()}")
+
+    def consume_chunk(self, chunk: Chunk) -> Chunk:
+        ka = []
+        gc = self._lib.to_gx_chunk(chunk, ka)
+        out = C.POINTER(GxResult)()
+        self._lib.check(
+            self._lib.lib.gxop_window_consume(self._op, C.byref(gc),
+                                              C.byref(out)),
+            "window_consume")
+        return self._lib.result_to_chunk(out)
+
+    def close(self):
+        if self._op:
+            self._lib.lib.gxop_window_close(self._op)
+            self._op = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+def run_window(lib, part_cols, aggs, input_types, input_chunks, reset=None,
+               **kw):
+    op = NonFrameOverWindowExec(lib, part_cols, aggs, input_types,
+                                reset=reset, **kw)
+    try:
+        return [op.consume_chunk(ch) for ch in input_chunks]
+    finally:
+        op.close()
+
+
 class PartitioningExchanger:
     """Mirrors mpp/operator/PartitioningExchanger.java:71-134."""
 

@@ -47,6 +47,15 @@ struct GxApi {
     decltype(&gxop_part_consume)  part_consume;
     decltype(&gxop_part_consume_concat) part_consume_concat;
     decltype(&gxop_part_close)    part_close;
+    decltype(&gxop_groupjoin_create)  gj_create;
+    decltype(&gxop_groupjoin_consume) gj_consume;
+    decltype(&gxop_groupjoin_build)   gj_build;
+    decltype(&gxop_groupjoin_probe)   gj_probe;
+    decltype(&gxop_groupjoin_next)    gj_next;
+    decltype(&gxop_groupjoin_close)   gj_close;
+    decltype(&gxop_window_create)     win_create;
+    decltype(&gxop_window_consume)    win_consume;
+    decltype(&gxop_window_close)      win_close;
     decltype(&gxop_scan_create)   scan_create;
     decltype(&gxop_scan_consume)  scan_consume;
     decltype(&gxop_scan_close)    scan_close;
@@ -83,6 +92,15 @@ struct GxApi {
         BIND(part_consume, gxop_part_consume)
         BIND(part_consume_concat, gxop_part_consume_concat)
         BIND(part_close, gxop_part_close)
+        BIND(gj_create, gxop_groupjoin_create)
+        BIND(gj_consume, gxop_groupjoin_consume)
+        BIND(gj_build, gxop_groupjoin_build)
+        BIND(gj_probe, gxop_groupjoin_probe)
+        BIND(gj_next, gxop_groupjoin_next)
+        BIND(gj_close, gxop_groupjoin_close)
+        BIND(win_create, gxop_window_create)
+        BIND(win_consume, gxop_window_consume)
+        BIND(win_close, gxop_window_close)
         BIND(scan_create, gxop_scan_create)
         BIND(scan_consume, gxop_scan_consume)
         BIND(scan_close, gxop_scan_close)
@@ -422,6 +440,133 @@ static void t_scan(int device) {
                 (long long)kept);
 }
 
+/* groupjoin: G build groups cycled by probe keys -> closed-form counts */
+static void t_groupjoin(int device) {
+    const int B = 2000, P = 30000;
+    std::vector<int64_t> bk(B), bp(B), pk(P), pv(P);
+    for (int i = 0; i < B; i++) { bk[i] = i; bp[i] = i * 100; }
+    for (int i = 0; i < P; i++) {
+        pk[i] = i % (2 * B);  /* half the probe keys miss */
+        pv[i] = i % 7;
+    }
+    gx_equi_key key{0, 0, GX_I64, 0};  /* outer=consumed col0, inner=probe col0 */
+    int32_t bt[2] = {GX_I64, GX_I64};
+    int32_t pt[2] = {GX_I64, GX_I64};
+    int32_t gcols[2] = {0, 1};
+    gx_agg_spec sp[2] = {{GX_AGG_COUNT_ROW, -1}, {GX_AGG_SUM_I64, 1}};
+    gx_groupjoin_cfg cfg{};
+    cfg.join_type = GX_JOIN_INNER;
+    cfg.n_keys = 1; cfg.keys = &key;
+    cfg.n_build_cols = 2; cfg.build_types = bt;
+    cfg.n_probe_cols = 2; cfg.probe_types = pt;
+    cfg.n_group_cols = 2; cfg.group_cols = gcols;
+    cfg.n_aggs = 2; cfg.aggs = sp;
+    cfg.device = device;
+    gx_op *op = api.gj_create(&cfg);
+    CHECK(op, "groupjoin_create");
+    if (!op) return;
+    gx_block bb[2] = {mk_i64(bk.data()), mk_i64(bp.data())};
+    gx_chunk bc{B, 2, bb};
+    CHECK(api.gj_consume(op, &bc) == 0, "gj consume");
+    CHECK(api.gj_build(op) == 0, "gj build");
+    gx_block pb[2] = {mk_i64(pk.data()), mk_i64(pv.data())};
+    gx_chunk pc{P, 2, pb};
+    CHECK(api.gj_probe(op, &pc) == 0, "gj probe");
+    int64_t groups = 0, cnt_total = 0;
+    for (;;) {
+        gx_result *res = nullptr;
+        CHECK(api.gj_next(op, &res) == 0, "gj next");
+        if (!res) break;
+        api.result_to_host(res);
+        for (int r = 0; r < res->chunk.n_rows; r++) {
+            int64_t k = cell_i64(res, 0, r);
+            if (cell_i64(res, 1, r) != k * 100) {
+                CHECK(false, "gj group payload row %d", r);
+                break;
+            }
+            /* every build key 0..B-1 is hit by P/(2B) probe rows + spill */
+            int64_t exp = P / (2 * B) + ((int64_t)k < (P % (2 * B)) ? 1 : 0);
+            if (cell_i64(res, 2, r) != exp) {
+                CHECK(false, "gj count group %lld: %lld != %lld",
+                      (long long)k, (long long)cell_i64(res, 2, r),
+                      (long long)exp);
+                break;
+            }
+            groups++;
+            cnt_total += cell_i64(res, 2, r);
+        }
+        api.result_release(res);
+    }
+    CHECK(groups == B, "gj groups %lld != %d", (long long)groups, B);
+    CHECK(cnt_total == P / 2, "gj total matches");
+    api.gj_close(op);
+    std::printf("  groupjoin: %lld groups, counts closed-form ok
+",
+                (long long)groups);
+}
+
+/* window: running COUNT/SUM over sorted partitions, checked closed-form */
+static void t_window(int device) {
+    const int N = 40000, PARTLEN = 37;
+    std::vector<int64_t> part(N), val(N);
+    for (int i = 0; i < N; i++) { part[i] = i / PARTLEN; val[i] = i % 5; }
+    int32_t pcols[1] = {0};
+    gx_agg_spec sp[2] = {{GX_AGG_COUNT_ROW, -1}, {GX_AGG_SUM_I64, 1}};
+    uint8_t rs[2] = {0, 0};
+    int32_t itypes[2] = {GX_I64, GX_I64};
+    gx_window_cfg cfg{};
+    cfg.n_part_cols = 1; cfg.part_cols = pcols;
+    cfg.n_aggs = 2; cfg.aggs = sp; cfg.reset = rs;
+    cfg.n_input_cols = 2; cfg.input_types = itypes;
+    cfg.device = device;
+    gx_op *op = api.win_create(&cfg);
+    CHECK(op, "window_create");
+    if (!op) return;
+    /* feed in two chunks so a partition straddles the boundary */
+    int64_t checked = 0;
+    const int SPLIT = N / 2 + 11;
+    for (int c = 0; c < 2; c++) {
+        int from = c == 0 ? 0 : SPLIT;
+        int to = c == 0 ? SPLIT : N;
+        gx_block bb[2] = {mk_i64(part.data() + from),
+                          mk_i64(val.data() + from)};
+        gx_chunk ch{to - from, 2, bb};
+        gx_result *res = nullptr;
+        CHECK(api.win_consume(op, &ch, &res) == 0, "win consume");
+        if (!res) continue;
+        api.result_to_host(res);
+        int64_t run_cnt = 0, run_sum = 0;
+        for (int r = 0; r < res->chunk.n_rows; r++) {
+            int64_t gi = from + r;
+            if (gi % PARTLEN == 0) { run_cnt = 0; run_sum = 0; }
+            /* recompute running from partition start (cross-chunk!) */
+            if (gi % PARTLEN == 0 || r == 0) {
+                run_cnt = 0; run_sum = 0;
+                for (int64_t j = gi - gi % PARTLEN; j <= gi; j++) {
+                    run_cnt++; run_sum += val[(size_t)j];
+                }
+            } else {
+                run_cnt++; run_sum += val[(size_t)gi];
+            }
+            if (cell_i64(res, 2, r) != run_cnt ||
+                cell_i64(res, 3, r) != run_sum) {
+                CHECK(false, "window row %lld: (%lld,%lld) != (%lld,%lld)",
+                      (long long)gi, (long long)cell_i64(res, 2, r),
+                      (long long)cell_i64(res, 3, r), (long long)run_cnt,
+                      (long long)run_sum);
+                break;
+            }
+            checked++;
+        }
+        api.result_release(res);
+    }
+    CHECK(checked == N, "window rows %lld != %d", (long long)checked, N);
+    api.win_close(op);
+    std::printf("  window: %lld running values ok (cross-chunk carry)
+",
+                (long long)checked);
+}
+
 /* ---- bench: join probe throughput through the pure C ABI --------------- */
 
 static void bench_join(int device, int64_t build_rows, int64_t probe_rows,
@@ -515,6 +660,8 @@ int main(int argc, char **argv) {
         t_agg(device);
         t_part(device);
         t_scan(device);
+        t_groupjoin(device);
+        t_window(device);
         if (g_fail) {
             std::printf("SELFTEST FAILED: %d check(s)\n", g_fail);
             return 1;
