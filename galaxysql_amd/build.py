@@ -35,9 +35,16 @@ def build_oracle(verbose=True):
     return os.path.join(REPO, "oracle", "libgxoracle.so")
 
 
+def build_driver(verbose=True):
+    subprocess.run(["make", "-C", os.path.join(REPO, "tools")], check=True,
+                   capture_output=not verbose)
+    return os.path.join(REPO, "tools", "gx_driver")
+
+
 def build_all():
     build_oracle()
     build_hip()
+    build_driver()
 
 
 if __name__ == "__main__":
