@@ -406,10 +406,12 @@ class C4:
                 "roofline_kernel": "agg insert+gid+accumulate chain"}
 
     def roofline_bytes_per_row(self, acc):
-        # agg chain algorithmic bytes per input row: pack 16 wr
-        # + 4 hash + insert slot 16 (+compare 16 for dup rows ~75%) +
-        # accumulate slot 16 + compare 16 + record RMW 16
-        return 16 + 4 + 16 + 0.75 * 16 + 16 + 16 + 16
+        # agg chain algorithmic bytes per input row (slot-indexed
+        # records): keystore append 8 wr + hash 4 wr + 4 rd + insert slot 16
+        # (+ keystore key compare 8 for the ~75% tie rows) + slot_of_row
+        # 4 wr + 4 rd + record RMW 8 (records[slot] directly — no gid
+        # indirection) + l_quantity 8 rd
+        return 8 + 4 + 4 + 16 + 0.75 * 8 + 4 + 4 + 8 + 8
 
     def cpu_baseline(self, sample_rows):
         return c4_cpu_baseline(sample_scale=1.0 / 64)
