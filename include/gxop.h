@@ -311,6 +311,58 @@ gx_op *gxop_window_create(const gx_window_cfg *cfg);
 int gxop_window_consume(gx_op *op, const gx_chunk *chunk, gx_result **out);
 int gxop_window_close(gx_op *op);
 
+/* ---- frame windows: whole-partition / sliding-rows frames -------------- */
+
+/* OverWindowFramesExec (operator/OverWindowFramesExec.java:38-200 + the
+ * frame classes under operator/frame/): windows whose value needs rows
+ * AFTER the current one. The reference buffers chunks per partition
+ * (ChunksIndex) and emits as partitions complete; this op buffers the
+ * whole input (agg-style consume -> finish barrier -> next emission) —
+ * identical results, a simpler barrier contract, and one big segmented
+ * GPU computation instead of per-partition replays. Frames supported:
+ *   GX_FRAME_WHOLE_PARTITION — UnboundedOverFrame: every row gets the
+ *     partition total (all 8 agg funcs);
+ *   GX_FRAME_ROWS_SLIDING — RowSlidingOverFrame: ROWS BETWEEN p PRECEDING
+ *     AND f FOLLOWING, exact-additive funcs only (COUNT_ROW / COUNT_COL /
+ *     SUM_I64) via segmented prefix differences;
+ *   GX_FRAME_ROWS_UNBOUNDED_FOLLOWING — RowUnboundedFollowingOverFrame:
+ *     CURRENT ROW .. UNBOUNDED FOLLOWING, same additive funcs.
+ * RANGE frames, sliding MIN/MAX (needs a deque/sparse-table pass) and
+ * sliding SUM(DOUBLE) (prefix differences change fp rounding order) are
+ * round-2; the create call rejects those combinations loudly. */
+typedef enum gx_frame_kind {
+    GX_FRAME_WHOLE_PARTITION = 0,
+    GX_FRAME_ROWS_SLIDING = 1,
+    GX_FRAME_ROWS_UNBOUNDED_FOLLOWING = 2
+} gx_frame_kind;
+
+typedef struct gx_frame_spec {
+    int32_t func;        /* gx_agg_func */
+    int32_t input_col;
+    int32_t kind;        /* gx_frame_kind */
+    int64_t preceding;   /* ROWS_SLIDING bounds (>=0) */
+    int64_t following;
+} gx_frame_spec;
+
+typedef struct gx_fwindow_cfg {
+    int32_t n_part_cols;
+    const int32_t *part_cols;
+    int32_t n_frames;
+    const gx_frame_spec *frames;
+    int32_t n_input_cols;
+    const int32_t *input_types;
+    int32_t device;
+    uint64_t stream;
+} gx_fwindow_cfg;
+
+gx_op *gxop_fwindow_create(const gx_fwindow_cfg *cfg);
+int gxop_fwindow_consume(gx_op *op, const gx_chunk *chunk); /* buffer */
+int gxop_fwindow_finish(gx_op *op);                         /* barrier */
+/* emit result batches (input cols + one col per frame, input order);
+ * *out=NULL when exhausted */
+int gxop_fwindow_next(gx_op *op, gx_result **out);
+int gxop_fwindow_close(gx_op *op);
+
 /* ---- scan: vectorized filter + project --------------------------------- */
 
 /* Mirrors the vectorized filter/projection stage (executor/vectorized/,

@@ -366,7 +366,7 @@ static gx_result *make_result(std::vector<OutCol> &&cols, int32_t n_rows) {
 /* ---- operator base ----------------------------------------------------- */
 
 enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4,
-              OP_GROUPJOIN = 5, OP_WINDOW = 6 };
+              OP_GROUPJOIN = 5, OP_WINDOW = 6, OP_FWINDOW = 7 };
 
 } // anonymous namespace
 
@@ -1221,6 +1221,147 @@ struct WindowOp : gx_op {
     }
 };
 
+
+/* ---- frame windows ------------------------------------------------------
+ * Restates OverWindowFramesExec + operator/frame/ UnboundedOverFrame /
+ * RowSlidingOverFrame / RowUnboundedFollowingOverFrame: buffer input,
+ * then per partition segment compute each frame's value per row.
+ * Emission preserves input order (the operator appends window columns). */
+struct FWindowOp : gx_op {
+    gx_fwindow_cfg cfg;
+    std::vector<int32_t> part_cols_, input_types;
+    std::vector<gx_frame_spec> frames;
+    Store input;
+    bool finished = false;
+    size_t emit_cursor = 0;
+    std::vector<size_t> seg_start_of_row;   /* first row of my segment */
+    std::vector<size_t> seg_end_of_row;     /* one past last row */
+    static constexpr int32_t CHUNK_SIZE = 1000;
+
+    FWindowOp(const gx_fwindow_cfg *c) : gx_op(OP_FWINDOW), cfg(*c) {
+        part_cols_.assign(c->part_cols, c->part_cols + c->n_part_cols);
+        input_types.assign(c->input_types, c->input_types + c->n_input_cols);
+        frames.assign(c->frames, c->frames + c->n_frames);
+        input.init((int32_t)input_types.size(), input_types.data());
+    }
+
+    int consume(const gx_chunk *ch) { return input.append(ch); }
+
+    bool same_part(size_t a, size_t b) const {
+        std::vector<int> pc(part_cols_.begin(), part_cols_.end());
+        return input.keys_equal(a, input, b, pc, pc);
+    }
+
+    int finish() {
+        const size_t n = input.n_rows;
+        seg_start_of_row.resize(n);
+        seg_end_of_row.resize(n);
+        size_t start = 0;
+        for (size_t i = 0; i < n; i++) {
+            if (i > 0 && !same_part(i - 1, i)) {
+                for (size_t j = start; j < i; j++) seg_end_of_row[j] = i;
+                start = i;
+            }
+            seg_start_of_row[i] = start;
+        }
+        for (size_t j = start; j < n; j++) seg_end_of_row[j] = n;
+        finished = true;
+        return 0;
+    }
+
+    /* accumulate rows [lo, hi) of frame f into (i64 or f64, isnull) */
+    void range_agg(const gx_frame_spec &f, size_t lo, size_t hi,
+                   int64_t &iv, double &dv, bool &isnull) const {
+        bool never_null = f.func == GX_AGG_COUNT_ROW ||
+                          f.func == GX_AGG_COUNT_COL ||
+                          f.func == GX_AGG_SUM_I64;
+        iv = 0; dv = 0; isnull = !never_null;
+        const Column *c = f.input_col >= 0 ? &input.cols[f.input_col] : nullptr;
+        if (f.func == GX_AGG_MIN_I64) iv = INT64_MAX;
+        if (f.func == GX_AGG_MAX_I64) iv = INT64_MIN;
+        for (size_t r = lo; r < hi; r++) {
+            switch (f.func) {
+            case GX_AGG_COUNT_ROW: iv++; break;
+            case GX_AGG_COUNT_COL: if (!c->is_null(r)) iv++; break;
+            case GX_AGG_SUM_I64:
+                if (!c->is_null(r))
+                    iv = (int64_t)((uint64_t)iv + (uint64_t)(
+                        c->type == GX_I32 ? (int64_t)c->i32v[r] : c->i64v[r]));
+                break;
+            case GX_AGG_SUM_F64:
+                if (!c->is_null(r)) {
+                    double v = c->type == GX_F64 ? c->f64v[r]
+                             : c->type == GX_I32 ? (double)c->i32v[r]
+                                                 : (double)c->i64v[r];
+                    if (isnull) { dv = v; isnull = false; } else dv += v;
+                }
+                break;
+            case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+                if (!c->is_null(r)) {
+                    int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[r]
+                                                  : c->i64v[r];
+                    iv = f.func == GX_AGG_MIN_I64 ? std::min(iv, v)
+                                                  : std::max(iv, v);
+                    isnull = false;
+                }
+                break;
+            case GX_AGG_MIN_F64: case GX_AGG_MAX_F64:
+                if (!c->is_null(r)) {
+                    double v = c->f64v[r];
+                    if (isnull) { dv = v; isnull = false; }
+                    else dv = f.func == GX_AGG_MIN_F64 ? std::min(dv, v)
+                                                       : std::max(dv, v);
+                }
+                break;
+            }
+        }
+    }
+
+    int next(gx_result **out) {
+        *out = nullptr;
+        if (!finished) { set_err("fwindow next before finish"); return -1; }
+        const size_t n = input.n_rows;
+        if (emit_cursor >= n) return 0;
+        size_t m = std::min<size_t>(CHUNK_SIZE, n - emit_cursor);
+        std::vector<int32_t> otypes = input_types;
+        for (auto &f : frames) {
+            switch (f.func) {
+            case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
+            case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+                otypes.push_back(GX_I64); break;
+            default: otypes.push_back(GX_F64); break;
+            }
+        }
+        std::vector<OutCol> cols(otypes.size());
+        for (size_t c = 0; c < otypes.size(); c++) cols[c].type = otypes[c];
+        for (size_t k = 0; k < m; k++) {
+            size_t i = emit_cursor + k;
+            for (size_t c = 0; c < input_types.size(); c++)
+                cols[c].append_from(input.cols[c], i);
+            for (size_t a = 0; a < frames.size(); a++) {
+                const gx_frame_spec &f = frames[a];
+                size_t s = seg_start_of_row[i], e = seg_end_of_row[i];
+                size_t lo = s, hi = e;
+                if (f.kind == GX_FRAME_ROWS_SLIDING) {
+                    lo = i >= s + (size_t)f.preceding ? i - (size_t)f.preceding : s;
+                    hi = std::min<size_t>(e, i + (size_t)f.following + 1);
+                } else if (f.kind == GX_FRAME_ROWS_UNBOUNDED_FOLLOWING) {
+                    lo = i;
+                }
+                int64_t iv; double dv; bool isnull;
+                range_agg(f, lo, hi, iv, dv, isnull);
+                size_t col = input_types.size() + a;
+                if (isnull) cols[col].append_null();
+                else if (otypes[col] == GX_I64) cols[col].append_i64(iv);
+                else cols[col].append_f64(dv);
+            }
+        }
+        emit_cursor += m;
+        *out = make_result(std::move(cols), (int32_t)m);
+        return 0;
+    }
+};
+
 /* ---- partition operator -------------------------------------------------
  * Restates PartitioningExchanger.consumeChunk (mpp/operator/
  * PartitioningExchanger.java:71-134): row hash over key cols (HashBucketFunction
@@ -1559,6 +1700,41 @@ int gxop_window_consume(gx_op *op, const gx_chunk *c, gx_result **out) {
     return static_cast<WindowOp *>(op)->consume(c, out);
 }
 int gxop_window_close(gx_op *op) { delete op; return 0; }
+
+
+static int fwindow_cfg_ok(const gx_fwindow_cfg *c) {
+    if (!c || c->n_frames <= 0) return 0;
+    for (int32_t i = 0; i < c->n_frames; i++) {
+        const gx_frame_spec &f = c->frames[i];
+        if (f.kind == GX_FRAME_WHOLE_PARTITION) continue;
+        /* sliding / following: exact-additive funcs only (see gxop.h) */
+        if (f.func != GX_AGG_COUNT_ROW && f.func != GX_AGG_COUNT_COL &&
+            f.func != GX_AGG_SUM_I64)
+            return 0;
+        if (f.kind == GX_FRAME_ROWS_SLIDING &&
+            (f.preceding < 0 || f.following < 0))
+            return 0;
+    }
+    return 1;
+}
+
+gx_op *gxop_fwindow_create(const gx_fwindow_cfg *cfg) {
+    if (!fwindow_cfg_ok(cfg)) { set_err("bad fwindow cfg (unsupported frame/func)"); return nullptr; }
+    return new FWindowOp(cfg);
+}
+int gxop_fwindow_consume(gx_op *op, const gx_chunk *c) {
+    if (!op || op->kind != OP_FWINDOW) { set_err("not an fwindow op"); return -1; }
+    return static_cast<FWindowOp *>(op)->consume(c);
+}
+int gxop_fwindow_finish(gx_op *op) {
+    if (!op || op->kind != OP_FWINDOW) { set_err("not an fwindow op"); return -1; }
+    return static_cast<FWindowOp *>(op)->finish();
+}
+int gxop_fwindow_next(gx_op *op, gx_result **out) {
+    if (!op || op->kind != OP_FWINDOW) { set_err("not an fwindow op"); return -1; }
+    return static_cast<FWindowOp *>(op)->next(out);
+}
+int gxop_fwindow_close(gx_op *op) { delete op; return 0; }
 
 gx_op *gxop_part_create(const gx_part_cfg *cfg) {
     if (!cfg || cfg->n_parts <= 0) { set_err("bad part cfg"); return nullptr; }
