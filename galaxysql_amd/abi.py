@@ -143,6 +143,29 @@ class GxWindowCfg(C.Structure):
     ]
 
 
+class GxFrameSpec(C.Structure):
+    _fields_ = [
+        ("func", C.c_int32),
+        ("input_col", C.c_int32),
+        ("kind", C.c_int32),
+        ("preceding", C.c_int64),
+        ("following", C.c_int64),
+    ]
+
+
+class GxFWindowCfg(C.Structure):
+    _fields_ = [
+        ("n_part_cols", C.c_int32),
+        ("part_cols", C.POINTER(C.c_int32)),
+        ("n_frames", C.c_int32),
+        ("frames", C.POINTER(GxFrameSpec)),
+        ("n_input_cols", C.c_int32),
+        ("input_types", C.POINTER(C.c_int32)),
+        ("device", C.c_int32),
+        ("stream", C.c_uint64),
+    ]
+
+
 class GxPartCfg(C.Structure):
     _fields_ = [
         ("n_parts", C.c_int32),
@@ -183,6 +206,8 @@ CONTAINS = 6  # SLICE LIKE '%pat%'
 # Projections (gx_proj_op)
 PROJ_COPY, PROJ_REV_F64, PROJ_REV_SCALED4, PROJ_Q9_AMOUNT4 = 0, 1, 2, 3
 PROJ_DEC_TO_SCALED, PROJ_SCALED_TO_DEC = 4, 5  # gx_proj.c = decimal scale
+FRAME_WHOLE_PARTITION, FRAME_ROWS_SLIDING, FRAME_ROWS_UNBOUNDED_FOLLOWING = \
+    0, 1, 2
 # Agg funcs (gx_agg_func)
 COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64, MAX_F64 = range(8)
 
@@ -230,6 +255,13 @@ class GxLib:
         L.gxop_window_consume.argtypes = [C.c_void_p, C.POINTER(GxChunk),
                                           C.POINTER(C.POINTER(GxResult))]
         L.gxop_window_close.argtypes = [C.c_void_p]
+        L.gxop_fwindow_create.restype = C.c_void_p
+        L.gxop_fwindow_create.argtypes = [C.POINTER(GxFWindowCfg)]
+        L.gxop_fwindow_consume.argtypes = [C.c_void_p, C.POINTER(GxChunk)]
+        L.gxop_fwindow_finish.argtypes = [C.c_void_p]
+        L.gxop_fwindow_next.argtypes = [C.c_void_p,
+                                        C.POINTER(C.POINTER(GxResult))]
+        L.gxop_fwindow_close.argtypes = [C.c_void_p]
         L.gxop_part_create.restype = C.c_void_p
         L.gxop_part_create.argtypes = [C.POINTER(GxPartCfg)]
         L.gxop_part_consume.argtypes = [C.c_void_p, C.POINTER(GxChunk),

@@ -648,7 +648,7 @@ static inline int gx_grid(int64_t n, int block = 256) {
 /* ========================= operator base =============================== */
 
 enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4,
-              OP_GROUPJOIN = 5, OP_WINDOW = 6 };
+              OP_GROUPJOIN = 5, OP_WINDOW = 6, OP_FWINDOW = 7 };
 
 struct gx_op {
     int kind;
@@ -1335,6 +1335,7 @@ struct JoinOp : gx_op {
 #include "gxhip_hybrid.inc"
 #include "gxhip_groupjoin.inc"
 #include "gxhip_window.inc"
+#include "gxhip_fwindow.inc"
 #include "gxhip_scan.inc"
 
 /* ========================= C ABI ======================================= */

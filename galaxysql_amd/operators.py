@@ -322,6 +322,77 @@ def run_window(lib, part_cols, aggs, input_types, input_chunks, reset=None,
         op.close()
 
 
+class OverWindowFramesExec:
+    """Frame windows (OverWindowFramesExec.java + operator/frame/):
+    consume* -> finish barrier -> result_chunks. frames: (func, input_col,
+    kind, preceding, following)."""
+
+    def __init__(self, lib, part_cols, frames, input_types, device=-1,
+                 stream=0):
+        from .abi import GxFrameSpec, GxFWindowCfg
+        self._lib = lib
+        self._keep = []
+        pc = (C.c_int32 * max(1, len(part_cols)))(*(part_cols or [0]))
+        fr = (GxFrameSpec * len(frames))()
+        for i, spec in enumerate(frames):
+            spec = tuple(spec) + (0,) * (5 - len(spec))
+            fr[i] = GxFrameSpec(*spec[:5])
+        it = (C.c_int32 * len(input_types))(*input_types)
+        cfg = GxFWindowCfg(n_part_cols=len(part_cols), part_cols=pc,
+                           n_frames=len(frames), frames=fr,
+                           n_input_cols=len(input_types), input_types=it,
+                           device=device, stream=stream)
+        self._keep += [pc, fr, it, cfg]
+        self._op = lib.lib.gxop_fwindow_create(C.byref(cfg))
+        if not self._op:
+            raise RuntimeError(f"gxop_fwindow_create: {lib.error()}")
+
+    def consume_chunk(self, chunk: Chunk):
+        ka = []
+        gc = self._lib.to_gx_chunk(chunk, ka)
+        self._lib.check(
+            self._lib.lib.gxop_fwindow_consume(self._op, C.byref(gc)),
+            "fwindow_consume")
+
+    def finish(self):
+        self._lib.check(self._lib.lib.gxop_fwindow_finish(self._op),
+                        "fwindow_finish")
+
+    def result_chunks(self):
+        chunks = []
+        while True:
+            out = C.POINTER(GxResult)()
+            self._lib.check(
+                self._lib.lib.gxop_fwindow_next(self._op, C.byref(out)),
+                "fwindow_next")
+            if not out:
+                break
+            chunks.append(self._lib.result_to_chunk(out))
+        return chunks
+
+    def close(self):
+        if self._op:
+            self._lib.lib.gxop_fwindow_close(self._op)
+            self._op = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+def run_fwindow(lib, part_cols, frames, input_types, input_chunks, **kw):
+    op = OverWindowFramesExec(lib, part_cols, frames, input_types, **kw)
+    try:
+        for ch in input_chunks:
+            op.consume_chunk(ch)
+        op.finish()
+        return op.result_chunks()
+    finally:
+        op.close()
+
+
 class PartitioningExchanger:
     """Mirrors mpp/operator/PartitioningExchanger.java:71-134."""
 

@@ -1,0 +1,104 @@
+"""Frame windows (OverWindowFramesExec subset) — hand cases + numpy
+cross-check on the oracle, HIP-vs-oracle parity on GPU.
+
+Semantics pinned: whole-partition totals on every row (UnboundedOverFrame),
+ROWS BETWEEN p PRECEDING AND f FOLLOWING clamped to the partition
+(RowSlidingOverFrame), CURRENT ROW..UNBOUNDED FOLLOWING
+(RowUnboundedFollowingOverFrame). Emission preserves input order."""
+import numpy as np
+import pytest
+
+from galaxysql_amd import abi
+from galaxysql_amd.chunk import Block, Chunk, I64, F64, chunks_from_columns, \
+    rows_of
+from galaxysql_amd.operators import run_fwindow
+
+
+def test_fwindow_hand_case():
+    lib = abi.load_oracle()
+    part = [1, 1, 1, 2, 2]
+    val = [10, None, 5, 7, 1]
+    chunks = [Chunk([Block.of(I64, part), Block.of(I64, val)])]
+    out = run_fwindow(
+        lib, [0],
+        [(abi.SUM_I64, 1, abi.FRAME_WHOLE_PARTITION),
+         (abi.MIN_I64, 1, abi.FRAME_WHOLE_PARTITION),
+         (abi.SUM_I64, 1, abi.FRAME_ROWS_SLIDING, 1, 1),
+         (abi.COUNT_ROW, -1, abi.FRAME_ROWS_SLIDING, 0, 1),
+         (abi.SUM_I64, 1, abi.FRAME_ROWS_UNBOUNDED_FOLLOWING)],
+        [I64, I64], chunks)
+    rows = rows_of(out)
+    assert rows == [
+        # part, val, whole_sum, whole_min, slide+-1, cnt_0_1, to_end
+        (1, 10, 15, 5, 10, 2, 15),
+        (1, None, 15, 5, 15, 2, 5),
+        (1, 5, 15, 5, 5, 1, 5),
+        (2, 7, 8, 1, 8, 2, 8),
+        (2, 1, 8, 1, 8, 1, 1),
+    ]
+
+
+def test_fwindow_rejects_unsupported():
+    lib = abi.load_oracle()
+    with pytest.raises(RuntimeError):
+        run_fwindow(lib, [0],
+                    [(abi.MIN_I64, 1, abi.FRAME_ROWS_SLIDING, 1, 1)],
+                    [I64, I64], [])
+
+
+def _gen(rng, n):
+    parts = np.sort(rng.integers(0, n // 23 + 1, n)).astype(np.int64)
+    vals = rng.integers(-50, 50, n).astype(np.int64)
+    nulls = (rng.random(n) < 0.1).astype(np.uint8)
+    fvals = rng.random(n) * 7
+    return chunks_from_columns([I64, I64, F64],
+                               [(parts, None), (vals, nulls),
+                                (fvals, None)], chunk_size=997), parts, \
+        vals, nulls
+
+
+def test_fwindow_oracle_vs_numpy():
+    lib = abi.load_oracle()
+    rng = np.random.default_rng(51)
+    chunks, parts, vals, nulls = _gen(rng, 4000)
+    out = run_fwindow(lib, [0],
+                      [(abi.SUM_I64, 1, abi.FRAME_WHOLE_PARTITION),
+                       (abi.SUM_I64, 1, abi.FRAME_ROWS_SLIDING, 2, 3)],
+                      [I64, I64, F64], chunks)
+    rows = rows_of(out)
+    v = np.where(nulls == 1, 0, vals)
+    for i, r in enumerate(rows):
+        seg = parts == parts[i]
+        idx = np.nonzero(seg)[0]
+        assert r[3] == int(v[seg].sum()), i
+        lo = max(idx[0], i - 2)
+        hi = min(idx[-1], i + 3)
+        assert r[4] == int(v[lo:hi + 1].sum()), i
+
+
+FRAMES = [(abi.SUM_I64, 1, abi.FRAME_WHOLE_PARTITION),
+          (abi.COUNT_COL, 1, abi.FRAME_WHOLE_PARTITION),
+          (abi.SUM_F64, 2, abi.FRAME_WHOLE_PARTITION),
+          (abi.MAX_I64, 1, abi.FRAME_WHOLE_PARTITION),
+          (abi.SUM_I64, 1, abi.FRAME_ROWS_SLIDING, 3, 2),
+          (abi.COUNT_ROW, -1, abi.FRAME_ROWS_SLIDING, 0, 0),
+          (abi.SUM_I64, 1, abi.FRAME_ROWS_UNBOUNDED_FOLLOWING)]
+
+
+@pytest.mark.gpu
+def test_gpu_fwindow_matches_oracle():
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    rng = np.random.default_rng(52)
+    chunks, *_ = _gen(rng, 30000)
+    got = run_fwindow(hip, [0], FRAMES, [I64, I64, F64], chunks, device=0)
+    want = run_fwindow(ora, [0], FRAMES, [I64, I64, F64], chunks, device=-1)
+    grows = rows_of(got)
+    wrows = rows_of(want)
+    assert len(grows) == len(wrows)
+    for i, (g, w) in enumerate(zip(grows, wrows)):
+        for a, b in zip(g, w):
+            if isinstance(a, float) and b is not None:
+                assert abs(a - b) < 1e-9, (i, g, w)
+            else:
+                assert a == b, (i, g, w)

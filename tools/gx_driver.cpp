@@ -54,6 +54,11 @@ struct GxApi {
     decltype(&gxop_groupjoin_next)    gj_next;
     decltype(&gxop_groupjoin_close)   gj_close;
     decltype(&gxop_window_create)     win_create;
+    decltype(&gxop_fwindow_create)    fwin_create;
+    decltype(&gxop_fwindow_consume)   fwin_consume;
+    decltype(&gxop_fwindow_finish)    fwin_finish;
+    decltype(&gxop_fwindow_next)      fwin_next;
+    decltype(&gxop_fwindow_close)     fwin_close;
     decltype(&gxop_window_consume)    win_consume;
     decltype(&gxop_window_close)      win_close;
     decltype(&gxop_scan_create)   scan_create;
@@ -99,6 +104,11 @@ struct GxApi {
         BIND(gj_next, gxop_groupjoin_next)
         BIND(gj_close, gxop_groupjoin_close)
         BIND(win_create, gxop_window_create)
+        BIND(fwin_create, gxop_fwindow_create)
+        BIND(fwin_consume, gxop_fwindow_consume)
+        BIND(fwin_finish, gxop_fwindow_finish)
+        BIND(fwin_next, gxop_fwindow_next)
+        BIND(fwin_close, gxop_fwindow_close)
         BIND(win_consume, gxop_window_consume)
         BIND(win_close, gxop_window_close)
         BIND(scan_create, gxop_scan_create)
@@ -567,6 +577,56 @@ static void t_window(int device) {
                 (long long)checked);
 }
 
+/* frame window: whole-partition totals + sliding counts, closed form */
+static void t_fwindow(int device) {
+    const int N = 30000, PARTLEN = 29;
+    std::vector<int64_t> part(N), val(N);
+    for (int i = 0; i < N; i++) { part[i] = i / PARTLEN; val[i] = i % 3; }
+    int32_t pcols[1] = {0};
+    gx_frame_spec fs[2] = {};
+    fs[0] = {GX_AGG_SUM_I64, 1, GX_FRAME_WHOLE_PARTITION, 0, 0};
+    fs[1] = {GX_AGG_COUNT_ROW, -1, GX_FRAME_ROWS_SLIDING, 2, 2};
+    int32_t itypes[2] = {GX_I64, GX_I64};
+    gx_fwindow_cfg cfg{};
+    cfg.n_part_cols = 1; cfg.part_cols = pcols;
+    cfg.n_frames = 2; cfg.frames = fs;
+    cfg.n_input_cols = 2; cfg.input_types = itypes;
+    cfg.device = device;
+    gx_op *op = api.fwin_create(&cfg);
+    CHECK(op, "fwindow_create");
+    if (!op) return;
+    gx_block bb[2] = {mk_i64(part.data()), mk_i64(val.data())};
+    gx_chunk c{N, 2, bb};
+    CHECK(api.fwin_consume(op, &c) == 0, "fwin consume");
+    CHECK(api.fwin_finish(op) == 0, "fwin finish");
+    int64_t checked = 0, at = 0;
+    for (;;) {
+        gx_result *res = nullptr;
+        CHECK(api.fwin_next(op, &res) == 0, "fwin next");
+        if (!res) break;
+        api.result_to_host(res);
+        for (int r = 0; r < res->chunk.n_rows; r++, at++) {
+            int64_t s = at - at % PARTLEN;
+            int64_t e = std::min<int64_t>(s + PARTLEN, N);
+            int64_t tot = 0;
+            for (int64_t j = s; j < e; j++) tot += val[(size_t)j];
+            int64_t lo = std::max<int64_t>(s, at - 2);
+            int64_t hi = std::min<int64_t>(e - 1, at + 2);
+            if (cell_i64(res, 2, r) != tot ||
+                cell_i64(res, 3, r) != hi - lo + 1) {
+                CHECK(false, "fwindow row %lld", (long long)at);
+                break;
+            }
+            checked++;
+        }
+        api.result_release(res);
+    }
+    CHECK(checked == N, "fwindow rows %lld != %d", (long long)checked, N);
+    api.fwin_close(op);
+    std::printf("  fwindow: %lld whole-partition + sliding values ok\n",
+                (long long)checked);
+}
+
 /* ---- bench: join probe throughput through the pure C ABI --------------- */
 
 static void bench_join(int device, int64_t build_rows, int64_t probe_rows,
@@ -662,6 +722,7 @@ int main(int argc, char **argv) {
         t_scan(device);
         t_groupjoin(device);
         t_window(device);
+        t_fwindow(device);
         if (g_fail) {
             std::printf("SELFTEST FAILED: %d check(s)\n", g_fail);
             return 1;
