@@ -1,0 +1,189 @@
+"""Randomized differential sweep: HIP vs oracle over random operator
+configurations (join type matrix x key shapes x null densities x hybrid
+spill x agg function subsets). Seeds are fixed — failures reproduce.
+
+This is breadth insurance on top of the targeted parity tests: every
+iteration builds a fresh random config, runs both implementations on the
+same data, and compares row multisets (f64 rounded to 1e-9)."""
+import numpy as np
+import pytest
+
+from galaxysql_amd import abi
+from galaxysql_amd.chunk import Block, Chunk, I64, I32, F64, SLICE, \
+    chunks_from_columns, multiset, rows_of
+from galaxysql_amd.operators import (EquiJoinKey, run_join, run_agg,
+                                     run_groupjoin, run_window)
+
+pytestmark = pytest.mark.gpu
+
+JOIN_TYPES = [abi.INNER, abi.LEFT, abi.RIGHT, abi.SEMI, abi.ANTI]
+KEY_TYPES = [I64, I32, SLICE]
+PAYLOADS = [I64, I32, F64, SLICE]
+AGG_FUNCS = [(abi.COUNT_ROW, None), (abi.COUNT_COL, I64),
+             (abi.SUM_I64, I64), (abi.SUM_F64, F64),
+             (abi.MIN_I64, I64), (abi.MAX_I64, I64),
+             (abi.MIN_F64, F64), (abi.MAX_F64, F64)]
+
+
+def _col(rng, t, n, card, null_frac):
+    nulls = (rng.random(n) < null_frac).astype(np.uint8) \
+        if null_frac > 0 else None
+    if t == SLICE:
+        vals = [None if nulls is not None and nulls[i]
+                else f"v{rng.integers(0, card)}" for i in range(n)]
+        return Block.of(SLICE, vals)
+    if t == I32:
+        v = rng.integers(-card, card, n).astype(np.int32)
+    elif t == F64:
+        v = np.round(rng.standard_normal(n) * card, 3)
+    else:
+        v = rng.integers(-card, card, n)
+    return Block(t, values=v, nulls=nulls)
+
+
+def _side(rng, key_types, n, card, null_frac, extra_payloads):
+    types = list(key_types)
+    blocks = [_col(rng, t, n, card, null_frac) for t in key_types]
+    for t in extra_payloads:
+        types.append(t)
+        blocks.append(_col(rng, t, n, 1000, null_frac / 2))
+    chunk_size = int(rng.integers(100, 2000))
+    cols = []
+    for b in blocks:
+        cols.append(b)
+    chunks = []
+    for start in range(0, n, chunk_size):
+        sub = []
+        for b in blocks:
+            if b.type == SLICE:
+                base = int(b.offsets[start - 1]) if start > 0 else 0
+                end = min(start + chunk_size, n)
+                off = (b.offsets[start:end] - base).astype(np.int32)
+                dend = int(b.offsets[end - 1]) if end > 0 else 0
+                sub.append(Block(SLICE,
+                                 nulls=None if b.nulls is None
+                                 else b.nulls[start:end],
+                                 offsets=off, data=b.data[base:dend]))
+            else:
+                end = min(start + chunk_size, n)
+                sub.append(Block(b.type, values=b.values[start:end],
+                                 nulls=None if b.nulls is None
+                                 else b.nulls[start:end]))
+        chunks.append(Chunk(sub))
+    return types, chunks
+
+
+@pytest.mark.parametrize("seed", range(12))
+def test_fuzz_join(seed):
+    rng = np.random.default_rng(1000 + seed)
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    n_keys = int(rng.integers(1, 3))
+    key_types = [KEY_TYPES[rng.integers(0, len(KEY_TYPES))]
+                 for _ in range(n_keys)]
+    jt = JOIN_TYPES[rng.integers(0, len(JOIN_TYPES))]
+    null_frac = float(rng.choice([0.0, 0.05, 0.3]))
+    card = int(rng.choice([5, 100, 5000]))
+    n_build = int(rng.integers(0, 4000))
+    n_probe = int(rng.integers(1, 8000))
+    bp = [PAYLOADS[rng.integers(0, len(PAYLOADS))]]
+    pp = [PAYLOADS[rng.integers(0, len(PAYLOADS))]]
+    btypes, build = _side(rng, key_types, n_build, card, null_frac, bp)
+    ptypes, probe = _side(rng, key_types, n_probe, card, null_frac, pp)
+    keys = [EquiJoinKey(i, i, key_types[i]) for i in range(n_keys)]
+    kw = {}
+    anti_ok = jt == abi.ANTI and n_keys == 1 and len(btypes) == 1
+    if anti_ok and rng.random() < 0.5:
+        kw["anti_null_col"] = 0
+    if anti_ok:
+        btypes, build = _side(rng, key_types, n_build, card, null_frac, [])
+    if rng.random() < 0.3 and jt in (abi.INNER, abi.LEFT):
+        kw["build_outer"] = True
+        # build side is the outer side: swap type roles
+        got = run_join(hip, jt, keys, build, probe, btypes, ptypes,
+                       device=0, expected_build_rows=n_build,
+                       memory_budget_bytes=4096 if rng.random() < 0.5 else 0,
+                       **kw)
+        want = run_join(ora, jt, keys, build, probe, btypes, ptypes,
+                        device=-1, **kw)
+    else:
+        got = run_join(hip, jt, keys, build, probe, ptypes, btypes,
+                       device=0, expected_build_rows=n_build,
+                       memory_budget_bytes=4096 if rng.random() < 0.5 else 0,
+                       **kw)
+        want = run_join(ora, jt, keys, build, probe, ptypes, btypes,
+                        device=-1, **kw)
+    assert multiset(rows_of(got), f64_round=9) == \
+        multiset(rows_of(want), f64_round=9), f"seed {seed}"
+
+
+@pytest.mark.parametrize("seed", range(8))
+def test_fuzz_agg(seed):
+    rng = np.random.default_rng(2000 + seed)
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    n = int(rng.integers(1, 50000))
+    n_group = int(rng.integers(0, 3))
+    gtypes = [KEY_TYPES[rng.integers(0, len(KEY_TYPES))]
+              for _ in range(n_group)]
+    null_frac = float(rng.choice([0.0, 0.2]))
+    card = int(rng.choice([3, 50, 2000]))
+    n_aggs = int(rng.integers(1, 5))
+    picks = [AGG_FUNCS[rng.integers(0, len(AGG_FUNCS))]
+             for _ in range(n_aggs)]
+    # input: group cols then one value col per agg that needs one
+    types = list(gtypes)
+    aggs = []
+    extra = []
+    for f, vt in picks:
+        if vt is None:
+            aggs.append((f, -1))
+        else:
+            aggs.append((f, len(gtypes) + len(extra)))
+            extra.append(vt)
+    types2, chunks = _side(rng, gtypes, n, card, null_frac, extra)
+    kw = dict(group_cols=list(range(n_group)), aggs=aggs,
+              expected_groups=int(rng.choice([0, 10, 100000])))
+    got = run_agg(hip, input_types=types2, input_chunks=chunks, device=0,
+                  **kw)
+    want = run_agg(ora, input_types=types2, input_chunks=chunks, device=-1,
+                   **kw)
+    assert multiset(rows_of(got), f64_round=9) == \
+        multiset(rows_of(want), f64_round=9), f"seed {seed}"
+
+
+@pytest.mark.parametrize("seed", range(4))
+def test_fuzz_groupjoin_window(seed):
+    rng = np.random.default_rng(3000 + seed)
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    # group-join
+    n_b, n_p = int(rng.integers(1, 3000)), int(rng.integers(1, 6000))
+    card = int(rng.choice([10, 500]))
+    btypes, build = _side(rng, [I64], n_b, card, 0.1, [I64])
+    ptypes, probe = _side(rng, [I64], n_p, card, 0.1, [I64, F64])
+    jt = abi.INNER if rng.random() < 0.5 else abi.LEFT
+    kw = dict(group_cols=[0, 1],
+              aggs=[(abi.COUNT_ROW, -1), (abi.SUM_I64, 1),
+                    (abi.SUM_F64, 2)])
+    got = run_groupjoin(hip, jt, [EquiJoinKey(0, 0, I64)], build, probe,
+                        btypes, ptypes, device=0, **kw)
+    want = run_groupjoin(ora, jt, [EquiJoinKey(0, 0, I64)], build, probe,
+                         btypes, ptypes, device=-1, **kw)
+    assert multiset(rows_of(got), f64_round=9) == \
+        multiset(rows_of(want), f64_round=9), f"seed {seed} gj"
+
+    # window (sorted partitions)
+    n = int(rng.integers(1, 20000))
+    parts = np.sort(rng.integers(0, max(n // 17, 1), n)).astype(np.int64)
+    vals = rng.integers(-100, 100, n)
+    nulls = (rng.random(n) < 0.15).astype(np.uint8)
+    chunks = chunks_from_columns([I64, I64], [(parts, None), (vals, nulls)],
+                                 chunk_size=int(rng.integers(50, 3000)))
+    aggs = [(abi.COUNT_ROW, -1), (abi.SUM_I64, 1), (abi.MIN_I64, 1)]
+    reset = [bool(rng.random() < 0.3) for _ in aggs]
+    got = run_window(hip, [0], aggs, [I64, I64], chunks, reset=reset,
+                     device=0)
+    want = run_window(ora, [0], aggs, [I64, I64], chunks, reset=reset,
+                      device=-1)
+    assert rows_of(got) == rows_of(want), f"seed {seed} win"
