@@ -1337,6 +1337,7 @@ struct JoinOp : gx_op {
 #include "gxhip_window.inc"
 #include "gxhip_fwindow.inc"
 #include "gxhip_scan.inc"
+#include "gx_serde.inc"
 
 /* ========================= C ABI ======================================= */
 

@@ -431,6 +431,19 @@ int gxop_result_copy_col(const gx_result *res, int32_t col,
                          void *dst_values, void *dst_nulls);
 const char *gx_last_error(void);
 
+/* ---- chunk wire format (PagesSerde-compatible; see galaxysql_amd/serde.py
+ * for the cited byte layout — gx_serde.inc is the compiled twin). Host
+ * chunks only. serialize mallocs *out (free with gxop_buf_free);
+ * deserialize returns a self-contained chunk (free with gxop_chunk_free)
+ * and the frame length consumed (for frame streams). */
+int gxop_chunk_serialize(const gx_chunk *chunk, uint8_t **out,
+                         int64_t *out_len);
+int gxop_chunk_deserialize(const uint8_t *buf, int64_t len,
+                           const int32_t *types, int32_t n_types,
+                           gx_chunk **out, int64_t *consumed);
+void gxop_chunk_free(gx_chunk *chunk);
+void gxop_buf_free(uint8_t *buf);
+
 /* perf introspection: cumulative probe-kernel time (HIP events around the
  * k_probe launches on the op's stream) + rows/matches, for bench.py's live
  * roofline leg. CPU oracle returns zeros. */

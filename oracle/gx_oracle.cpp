@@ -1797,3 +1797,5 @@ const char *gx_last_error(void) { return g_err.c_str(); }
 int gxop_abi_version(void) { return 0; /* CPU oracle */ }
 
 } /* extern "C" */
+
+#include "../galaxysql_amd/csrc/gx_serde.inc"
