@@ -73,7 +73,7 @@ def _side(rng, key_types, n, card, null_frac, extra_payloads):
     return types, chunks
 
 
-@pytest.mark.parametrize("seed", range(12))
+@pytest.mark.parametrize("seed", range(24))
 def test_fuzz_join(seed):
     rng = np.random.default_rng(1000 + seed)
     hip = abi.load_hip()
@@ -117,7 +117,7 @@ def test_fuzz_join(seed):
         multiset(rows_of(want), f64_round=9), f"seed {seed}"
 
 
-@pytest.mark.parametrize("seed", range(8))
+@pytest.mark.parametrize("seed", range(16))
 def test_fuzz_agg(seed):
     rng = np.random.default_rng(2000 + seed)
     hip = abi.load_hip()
