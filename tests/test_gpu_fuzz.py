@@ -16,6 +16,39 @@ from galaxysql_amd.operators import (EquiJoinKey, run_join, run_agg,
 
 pytestmark = pytest.mark.gpu
 
+
+def _close(a, b):
+    if a is None or b is None:
+        return a is b
+    return abs(a - b) <= 1e-9 * max(1.0, abs(a), abs(b))
+
+
+def assert_rows_match(got, want, ctx):
+    """Exact columns via multiset; float columns via RELATIVE tolerance
+    (atomic-order fp drift on a 50K-value SUM(double) exceeds any absolute
+    rounding — fuzz seed 2010 caught exactly that)."""
+    from collections import defaultdict
+
+    def split(rows):
+        g = defaultdict(list)
+        for r in rows:
+            exact = tuple(v for v in r if not isinstance(v, float))
+            floats = tuple(v for v in r if isinstance(v, float))
+            g[exact].append(floats)
+        return g
+
+    gg, ww = split(got), split(want)
+    assert sorted(gg.keys()) == sorted(ww.keys()), ctx
+    for k in gg:
+        a = sorted(gg[k], key=lambda t: [(-1e300 if x is None else x)
+                                         for x in t])
+        b = sorted(ww[k], key=lambda t: [(-1e300 if x is None else x)
+                                         for x in t])
+        assert len(a) == len(b), (ctx, k)
+        for fa, fb in zip(a, b):
+            for x, y in zip(fa, fb):
+                assert _close(x, y), (ctx, k, fa, fb)
+
 JOIN_TYPES = [abi.INNER, abi.LEFT, abi.RIGHT, abi.SEMI, abi.ANTI]
 KEY_TYPES = [I64, I32, SLICE]
 PAYLOADS = [I64, I32, F64, SLICE]
@@ -113,8 +146,7 @@ def test_fuzz_join(seed):
                        **kw)
         want = run_join(ora, jt, keys, build, probe, ptypes, btypes,
                         device=-1, **kw)
-    assert multiset(rows_of(got), f64_round=9) == \
-        multiset(rows_of(want), f64_round=9), f"seed {seed}"
+    assert_rows_match(rows_of(got), rows_of(want), f"join seed {seed}")
 
 
 @pytest.mark.parametrize("seed", range(16))
@@ -148,8 +180,7 @@ def test_fuzz_agg(seed):
                   **kw)
     want = run_agg(ora, input_types=types2, input_chunks=chunks, device=-1,
                    **kw)
-    assert multiset(rows_of(got), f64_round=9) == \
-        multiset(rows_of(want), f64_round=9), f"seed {seed}"
+    assert_rows_match(rows_of(got), rows_of(want), f"agg seed {seed}")
 
 
 @pytest.mark.parametrize("seed", range(4))
@@ -170,8 +201,7 @@ def test_fuzz_groupjoin_window(seed):
                         btypes, ptypes, device=0, **kw)
     want = run_groupjoin(ora, jt, [EquiJoinKey(0, 0, I64)], build, probe,
                          btypes, ptypes, device=-1, **kw)
-    assert multiset(rows_of(got), f64_round=9) == \
-        multiset(rows_of(want), f64_round=9), f"seed {seed} gj"
+    assert_rows_match(rows_of(got), rows_of(want), f"gj seed {seed}")
 
     # window (sorted partitions)
     n = int(rng.integers(1, 20000))
