@@ -37,8 +37,13 @@ def assert_rows_match(got, want, ctx):
             g[exact].append(floats)
         return g
 
+    def kkey(t):
+        return tuple((0, "") if v is None else
+                     (1, v.decode("latin1")) if isinstance(v, bytes) else
+                     (2, str(v)) for v in t)
+
     gg, ww = split(got), split(want)
-    assert sorted(gg.keys()) == sorted(ww.keys()), ctx
+    assert sorted(gg.keys(), key=kkey) == sorted(ww.keys(), key=kkey), ctx
     for k in gg:
         a = sorted(gg[k], key=lambda t: [(-1e300 if x is None else x)
                                          for x in t])
