@@ -61,6 +61,10 @@ struct GxApi {
     decltype(&gxop_fwindow_close)     fwin_close;
     decltype(&gxop_window_consume)    win_consume;
     decltype(&gxop_window_close)      win_close;
+    decltype(&gxop_chunk_serialize)   ser;
+    decltype(&gxop_chunk_deserialize) deser;
+    decltype(&gxop_chunk_free)        chunk_free;
+    decltype(&gxop_buf_free)          buf_free;
     decltype(&gxop_scan_create)   scan_create;
     decltype(&gxop_scan_consume)  scan_consume;
     decltype(&gxop_scan_close)    scan_close;
@@ -111,6 +115,10 @@ struct GxApi {
         BIND(fwin_close, gxop_fwindow_close)
         BIND(win_consume, gxop_window_consume)
         BIND(win_close, gxop_window_close)
+        BIND(ser, gxop_chunk_serialize)
+        BIND(deser, gxop_chunk_deserialize)
+        BIND(chunk_free, gxop_chunk_free)
+        BIND(buf_free, gxop_buf_free)
         BIND(scan_create, gxop_scan_create)
         BIND(scan_consume, gxop_scan_consume)
         BIND(scan_close, gxop_scan_close)
@@ -627,6 +635,43 @@ static void t_fwindow(int device) {
                 (long long)checked);
 }
 
+/* wire format: serialize -> deserialize -> reserialize must be identical
+ * bytes, and the LongBlock golden frame must match the reference layout */
+static void t_serde() {
+    int64_t vals[3] = {7, 0, -1};
+    uint8_t nulls[3] = {0, 1, 0};
+    gx_block b = mk_i64(vals, nulls);
+    gx_chunk c{3, 1, &b};
+    uint8_t *buf = nullptr;
+    int64_t len = 0;
+    CHECK(api.ser(&c, &buf, &len) == 0, "serialize");
+    if (!buf) return;
+    /* golden frame: [3]['\0'][25][25] [1][3][0x40][7][-1] */
+    const uint8_t expect[] = {
+        3,0,0,0, 0, 25,0,0,0, 25,0,0,0,
+        1,0,0,0, 3,0,0,0, 0x40,
+        7,0,0,0,0,0,0,0,
+        0xFF,0xFF,0xFF,0xFF,0xFF,0xFF,0xFF,0xFF};
+    CHECK(len == (int64_t)sizeof(expect), "frame length %lld", (long long)len);
+    CHECK(std::memcmp(buf, expect, sizeof(expect)) == 0, "golden bytes");
+    int32_t types[1] = {GX_I64};
+    gx_chunk *back = nullptr;
+    int64_t consumed = 0;
+    CHECK(api.deser(buf, len, types, 1, &back, &consumed) == 0, "deser");
+    CHECK(back && consumed == len, "deser consumed");
+    if (back) {
+        uint8_t *buf2 = nullptr;
+        int64_t len2 = 0;
+        CHECK(api.ser(back, &buf2, &len2) == 0, "reserialize");
+        CHECK(len2 == len && std::memcmp(buf, buf2, (size_t)len) == 0,
+              "roundtrip bytes");
+        api.buf_free(buf2);
+        api.chunk_free(back);
+    }
+    api.buf_free(buf);
+    std::printf("  serde: golden frame + roundtrip ok\n");
+}
+
 /* ---- bench: join probe throughput through the pure C ABI --------------- */
 
 static void bench_join(int device, int64_t build_rows, int64_t probe_rows,
@@ -723,6 +768,7 @@ int main(int argc, char **argv) {
         t_groupjoin(device);
         t_window(device);
         t_fwindow(device);
+        t_serde();
         if (g_fail) {
             std::printf("SELFTEST FAILED: %d check(s)\n", g_fail);
             return 1;
