@@ -103,3 +103,31 @@ def shuffle_columns(lib, tensors, types, key_cols, device=-1, group=None):
     send_cols, counts = partition_concat(lib, tensors, types, key_cols, world,
                                          device=device)
     return all_to_all_columns(send_cols, counts, group=group)
+
+
+def final_agg_specs(n_group_cols, aggs):
+    """Partial->final aggregation mapping for the MPP two-phase plan (the
+    reference planner splits an exchanged aggregation into a partial
+    HashAggExec below the shuffle and a final one above it; re-aggregation
+    functions follow SQL: COUNT re-aggregates as SUM, SUM as SUM, MIN/MAX
+    as themselves).
+
+    Input: the PARTIAL phase's agg specs [(func, input_col)].
+    Output: (final_specs, final_input_types_suffix) where final specs
+    index the partial OUTPUT schema: group cols first, then one column
+    per partial aggregator."""
+    from . import abi
+    finals = []
+    out_types = []
+    for i, (func, _col) in enumerate(aggs):
+        col = n_group_cols + i
+        if func in (abi.COUNT_ROW, abi.COUNT_COL):
+            finals.append((abi.SUM_I64, col))
+            out_types.append(0)  # I64
+        elif func in (abi.SUM_I64, abi.MIN_I64, abi.MAX_I64):
+            finals.append((func, col))
+            out_types.append(0)
+        else:  # SUM_F64 / MIN_F64 / MAX_F64
+            finals.append((func, col))
+            out_types.append(2)  # F64
+    return finals, out_types

@@ -1,0 +1,66 @@
+"""Partial -> shuffle -> final aggregation equals single-phase aggregation
+(the MPP two-phase plan the reference planner emits around an exchange;
+re-agg mapping in galaxysql_amd.exchange.final_agg_specs: COUNT->SUM,
+SUM->SUM, MIN/MAX->self). Simulated with 3 'ranks' on the oracle — the
+RCCL exchange mechanics themselves are covered by test_exchange_cpu."""
+import numpy as np
+
+from galaxysql_amd import abi
+from galaxysql_amd.chunk import I64, I32, F64, chunks_from_columns, \
+    multiset, rows_of
+from galaxysql_amd.exchange import final_agg_specs
+from galaxysql_amd.operators import run_agg, PartitioningExchanger
+
+
+AGGS = [(abi.COUNT_ROW, -1), (abi.COUNT_COL, 2), (abi.SUM_I64, 2),
+        (abi.SUM_F64, 3), (abi.MIN_I64, 2), (abi.MAX_F64, 3)]
+GROUP_COLS = [0, 1]
+TYPES = [I64, I32, I64, F64]
+
+
+def _data(rng, n):
+    g1 = rng.integers(0, 40, n)
+    g2 = rng.integers(0, 5, n).astype(np.int32)
+    v = rng.integers(-100, 100, n)
+    nulls = (rng.random(n) < 0.2).astype(np.uint8)
+    f = np.round(rng.standard_normal(n), 3)
+    return chunks_from_columns(TYPES, [(g1, None), (g2, None), (v, nulls),
+                                       (f, None)])
+
+
+def test_two_phase_equals_single_phase():
+    lib = abi.load_oracle()
+    rng = np.random.default_rng(61)
+    chunks = _data(rng, 9000)
+
+    # single phase
+    want = run_agg(lib, GROUP_COLS, AGGS, TYPES, chunks)
+
+    # phase 1: 3 ranks each aggregate a share of the chunks
+    ranks = 3
+    partials = [[] for _ in range(ranks)]
+    for i, c in enumerate(chunks):
+        partials[i % ranks].append(c)
+    partial_out = [run_agg(lib, GROUP_COLS, AGGS, TYPES, share)
+                   for share in partials]
+
+    # shuffle partial rows by group-key hash (Java-exact routing)
+    finals, _ = final_agg_specs(len(GROUP_COLS), AGGS)
+    ptypes = [I64, I32] + [I64, I64, I64, F64, I64, F64]
+    buckets = [[] for _ in range(ranks)]
+    ex = PartitioningExchanger(lib, ranks, key_cols=GROUP_COLS,
+                               input_types=ptypes)
+    for rank_out in partial_out:
+        for c in rank_out:
+            for p, out in enumerate(ex.consume_chunk(c)):
+                if out is not None and out.n_rows:
+                    buckets[p].append(out)
+
+    # phase 2: final agg per destination rank; union of results
+    got = []
+    for p in range(ranks):
+        if buckets[p]:
+            got.extend(run_agg(lib, GROUP_COLS, finals, ptypes, buckets[p]))
+
+    assert multiset(rows_of(got), f64_round=6) == \
+        multiset(rows_of(want), f64_round=6)
