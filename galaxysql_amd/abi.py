@@ -209,7 +209,8 @@ PROJ_DEC_TO_SCALED, PROJ_SCALED_TO_DEC = 4, 5  # gx_proj.c = decimal scale
 FRAME_WHOLE_PARTITION, FRAME_ROWS_SLIDING, FRAME_ROWS_UNBOUNDED_FOLLOWING = \
     0, 1, 2
 # Agg funcs (gx_agg_func)
-COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64, MAX_F64 = range(8)
+COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64, \
+    MAX_F64, AVG_F64 = range(9)
 
 ORACLE_PATH = os.path.join(_REPO, "oracle", "libgxoracle.so")
 HIP_PATH = os.path.join(_REPO, "galaxysql_amd", "csrc", "libgxhip.so")

@@ -121,6 +121,11 @@ def final_agg_specs(n_group_cols, aggs):
     out_types = []
     for i, (func, _col) in enumerate(aggs):
         col = n_group_cols + i
+        if func == abi.AVG_F64:
+            raise ValueError(
+                "AVG must be planned as partial SUM+COUNT around an "
+                "exchange (the reference planner does the same split); "
+                "it never crosses a shuffle as AVG")
         if func in (abi.COUNT_ROW, abi.COUNT_COL):
             finals.append((abi.SUM_I64, col))
             out_types.append(0)  # I64

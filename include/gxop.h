@@ -178,7 +178,12 @@ typedef enum gx_agg_func {
     GX_AGG_MIN_I64   = 4,  /* Long2LongMin: init NULL             */
     GX_AGG_MAX_I64   = 5,  /* Long2LongMax: init NULL             */
     GX_AGG_MIN_F64   = 6,  /* Double2DoubleMin: init NULL         */
-    GX_AGG_MAX_F64   = 7   /* Double2DoubleMax: init NULL         */
+    GX_AGG_MAX_F64   = 7,  /* Double2DoubleMax: init NULL         */
+    GX_AGG_AVG_F64   = 8   /* Avg over doubles: state {sum, count},
+                              NULL when no non-null input; for the
+                              two-phase exchange plan the planner splits
+                              AVG into partial SUM+COUNT (standard MPP),
+                              so AVG never crosses a shuffle as-is */
 } gx_agg_func;
 
 typedef struct gx_agg_spec {

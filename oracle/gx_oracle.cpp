@@ -734,6 +734,9 @@ struct AggOp : gx_op {
                 s.i64.push_back(0); s.isnull.push_back(0); break; /* Long2LongSum0 init 0 */
             case GX_AGG_SUM_F64: case GX_AGG_MIN_F64: case GX_AGG_MAX_F64:
                 s.f64.push_back(0); s.isnull.push_back(1); break; /* init NULL */
+            case GX_AGG_AVG_F64:
+                s.f64.push_back(0); s.i64.push_back(0);
+                s.isnull.push_back(1); break; /* {sum, count}; NULL at 0 */
             case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
                 s.i64.push_back(0); s.isnull.push_back(1); break; /* init NULL */
             }
@@ -811,6 +814,16 @@ struct AggOp : gx_op {
                 else s.f64[gid] += v;
             }
             break;
+        case GX_AGG_AVG_F64:
+            if (!c->is_null(row)) {
+                double v = c->type == GX_F64 ? c->f64v[row]
+                         : c->type == GX_I32 ? (double)c->i32v[row]
+                                             : (double)c->i64v[row];
+                s.f64[gid] += v;
+                s.i64[gid]++;
+                s.isnull[gid] = 0;
+            }
+            break;
         case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             if (!c->is_null(row)) {
                 int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[row] : c->i64v[row];
@@ -871,6 +884,8 @@ struct AggOp : gx_op {
                     !(aggs[a].func == GX_AGG_COUNT_ROW || aggs[a].func == GX_AGG_COUNT_COL
                       || aggs[a].func == GX_AGG_SUM_I64))
                     cols[col++].append_null();
+                else if (aggs[a].func == GX_AGG_AVG_F64)
+                    cols[col++].append_f64(s.f64[g] / (double)s.i64[g]);
                 else if (is_i64) cols[col++].append_i64(s.i64[g]);
                 else cols[col++].append_f64(s.f64[g]);
             }
@@ -982,6 +997,16 @@ struct GroupJoinOp : gx_op {
                 else s.f64[gid] += v;
             }
             break;
+        case GX_AGG_AVG_F64:
+            if (!c->is_null(row)) {
+                double v = c->type == GX_F64 ? c->f64v[row]
+                         : c->type == GX_I32 ? (double)c->i32v[row]
+                                             : (double)c->i64v[row];
+                s.f64[gid] += v;
+                s.i64[gid]++;
+                s.isnull[gid] = 0;
+            }
+            break;
         case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             if (!c->is_null(row)) {
                 int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[row]
@@ -1057,6 +1082,8 @@ struct GroupJoinOp : gx_op {
                                aggs[a].func == GX_AGG_COUNT_COL ||
                                aggs[a].func == GX_AGG_SUM_I64))
                     cols[col++].append_null();
+                else if (aggs[a].func == GX_AGG_AVG_F64)
+                    cols[col++].append_f64(s.f64[g] / (double)s.i64[g]);
                 else if (otypes[group_cols_.size() + a] == GX_I64)
                     cols[col++].append_i64(i64v);
                 else
@@ -1106,7 +1133,8 @@ struct WindowOp : gx_op {
         switch (aggs[a].func) {
         case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL: case GX_AGG_SUM_I64:
             s.i64 = 0; s.isnull = 0; break;
-        default: s.i64 = 0; s.f64 = 0; s.isnull = 1; break;
+        default: s.i64 = 0; s.f64 = 0; s.isnull = 1; break; /* incl. AVG:
+            i64 = count, f64 = sum */
         }
     }
 
@@ -1145,6 +1173,14 @@ struct WindowOp : gx_op {
                 if (s.isnull) { s.f64 = v; s.isnull = 0; }
                 else s.f64 = sp.func == GX_AGG_MIN_F64 ? std::min(s.f64, v)
                                                        : std::max(s.f64, v);
+            }
+            break;
+        case GX_AGG_AVG_F64:
+            if (!c->is_null(r)) {
+                double v = c->type == GX_F64 ? c->f64v[r]
+                         : c->type == GX_I32 ? (double)c->i32v[r]
+                                             : (double)c->i64v[r];
+                s.f64 += v; s.i64++; s.isnull = 0;
             }
             break;
         }
@@ -1212,6 +1248,8 @@ struct WindowOp : gx_op {
                 RunState &s = run[a];
                 bool i64out = otypes[col] == GX_I64;
                 if (s.isnull) cols[col].append_null();
+                else if (aggs[a].func == GX_AGG_AVG_F64)
+                    cols[col].append_f64(s.f64 / (double)s.i64);
                 else if (i64out) cols[col].append_i64(s.i64);
                 else cols[col].append_f64(s.f64);
             }
@@ -1313,6 +1351,14 @@ struct FWindowOp : gx_op {
                                                        : std::max(dv, v);
                 }
                 break;
+            case GX_AGG_AVG_F64:
+                if (!c->is_null(r)) {
+                    double v = c->type == GX_F64 ? c->f64v[r]
+                             : c->type == GX_I32 ? (double)c->i32v[r]
+                                                 : (double)c->i64v[r];
+                    dv += v; iv++; isnull = false;
+                }
+                break;
             }
         }
     }
@@ -1352,6 +1398,8 @@ struct FWindowOp : gx_op {
                 range_agg(f, lo, hi, iv, dv, isnull);
                 size_t col = input_types.size() + a;
                 if (isnull) cols[col].append_null();
+                else if (f.func == GX_AGG_AVG_F64)
+                    cols[col].append_f64(dv / (double)iv);
                 else if (otypes[col] == GX_I64) cols[col].append_i64(iv);
                 else cols[col].append_f64(dv);
             }

@@ -77,6 +77,7 @@ def test_fwindow_oracle_vs_numpy():
 
 
 FRAMES = [(abi.SUM_I64, 1, abi.FRAME_WHOLE_PARTITION),
+          (abi.AVG_F64, 1, abi.FRAME_WHOLE_PARTITION),
           (abi.COUNT_COL, 1, abi.FRAME_WHOLE_PARTITION),
           (abi.SUM_F64, 2, abi.FRAME_WHOLE_PARTITION),
           (abi.MAX_I64, 1, abi.FRAME_WHOLE_PARTITION),

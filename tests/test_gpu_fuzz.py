@@ -60,7 +60,8 @@ PAYLOADS = [I64, I32, F64, SLICE]
 AGG_FUNCS = [(abi.COUNT_ROW, None), (abi.COUNT_COL, I64),
              (abi.SUM_I64, I64), (abi.SUM_F64, F64),
              (abi.MIN_I64, I64), (abi.MAX_I64, I64),
-             (abi.MIN_F64, F64), (abi.MAX_F64, F64)]
+             (abi.MIN_F64, F64), (abi.MAX_F64, F64),
+             (abi.AVG_F64, F64), (abi.AVG_F64, I64)]
 
 
 def _col(rng, t, n, card, null_frac):

@@ -11,7 +11,7 @@ import pytest
 
 from galaxysql_amd import abi
 from galaxysql_amd.chunk import Block, Chunk, I64, F64, SLICE, \
-    chunks_from_columns, rows_of
+    chunks_from_columns, multiset, rows_of
 from galaxysql_amd.operators import run_window
 
 
@@ -115,7 +115,7 @@ def test_window_oracle_vs_numpy():
 
 AGGS = [(abi.COUNT_ROW, -1), (abi.COUNT_COL, 1), (abi.SUM_I64, 1),
         (abi.MIN_I64, 1), (abi.MAX_I64, 1), (abi.SUM_F64, 2),
-        (abi.MIN_F64, 2)]
+        (abi.MIN_F64, 2), (abi.AVG_F64, 1), (abi.AVG_F64, 2)]
 
 
 @pytest.mark.gpu
@@ -163,3 +163,23 @@ def test_gpu_window_row_number_large():
                                                            np.arange(n),
                                                            0)) + 1
     assert np.array_equal(rn, expect)
+
+
+def test_avg_hand_case_oracle():
+    """Avg (calc/aggfunctions/Avg.java): NULL until a non-null input, then
+    running mean — via HashAgg, window, and whole-partition frame."""
+    from galaxysql_amd.operators import run_agg, run_fwindow
+    lib = abi.load_oracle()
+    part = [1, 1, 1, 2]
+    val = [10, None, 5, 7]
+    chunks = [Chunk([Block.of(I64, part), Block.of(I64, val)])]
+    agg = rows_of(run_agg(lib, [0], [(abi.AVG_F64, 1)], [I64, I64], chunks))
+    assert multiset(agg) == multiset([(1, 7.5), (2, 7.0)])
+    win = rows_of(run_window(lib, [0], [(abi.AVG_F64, 1)], [I64, I64],
+                             chunks))
+    assert win == [(1, 10, 10.0), (1, None, 10.0), (1, 5, 7.5),
+                   (2, 7, 7.0)]
+    fw = rows_of(run_fwindow(lib, [0],
+                             [(abi.AVG_F64, 1, abi.FRAME_WHOLE_PARTITION)],
+                             [I64, I64], chunks))
+    assert fw == [(1, 10, 7.5), (1, None, 7.5), (1, 5, 7.5), (2, 7, 7.0)]
