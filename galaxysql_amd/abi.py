@@ -210,7 +210,7 @@ FRAME_WHOLE_PARTITION, FRAME_ROWS_SLIDING, FRAME_ROWS_UNBOUNDED_FOLLOWING = \
     0, 1, 2
 # Agg funcs (gx_agg_func)
 COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64, \
-    MAX_F64, AVG_F64 = range(9)
+    MAX_F64, AVG_F64, BIT_AND, BIT_OR, BIT_XOR = range(12)
 
 ORACLE_PATH = os.path.join(_REPO, "oracle", "libgxoracle.so")
 HIP_PATH = os.path.join(_REPO, "galaxysql_amd", "csrc", "libgxhip.so")

@@ -179,11 +179,18 @@ typedef enum gx_agg_func {
     GX_AGG_MAX_I64   = 5,  /* Long2LongMax: init NULL             */
     GX_AGG_MIN_F64   = 6,  /* Double2DoubleMin: init NULL         */
     GX_AGG_MAX_F64   = 7,  /* Double2DoubleMax: init NULL         */
-    GX_AGG_AVG_F64   = 8   /* Avg over doubles: state {sum, count},
+    GX_AGG_AVG_F64   = 8,  /* Avg over doubles: state {sum, count},
                               NULL when no non-null input; for the
                               two-phase exchange plan the planner splits
                               AVG into partial SUM+COUNT (standard MPP),
                               so AVG never crosses a shuffle as-is */
+    GX_AGG_BIT_AND   = 9,  /* SpecificType2UInt64BitAnd: init all-ones,
+                              AND of non-null inputs, NEVER NULL (the
+                              reference emits the init value for empty) */
+    GX_AGG_BIT_OR    = 10, /* ...BitOr: init 0, never NULL */
+    GX_AGG_BIT_XOR   = 11  /* ...BitXor: init 0, never NULL; the only
+                              BIT agg valid in sliding frames (invertible)
+                              — still excluded there this round */
 } gx_agg_func;
 
 typedef struct gx_agg_spec {

@@ -737,6 +737,10 @@ struct AggOp : gx_op {
             case GX_AGG_AVG_F64:
                 s.f64.push_back(0); s.i64.push_back(0);
                 s.isnull.push_back(1); break; /* {sum, count}; NULL at 0 */
+            case GX_AGG_BIT_AND:
+                s.i64.push_back(-1); s.isnull.push_back(0); break;
+            case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
+                s.i64.push_back(0); s.isnull.push_back(0); break;
             case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
                 s.i64.push_back(0); s.isnull.push_back(1); break; /* init NULL */
             }
@@ -814,6 +818,15 @@ struct AggOp : gx_op {
                 else s.f64[gid] += v;
             }
             break;
+        case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
+            if (!c->is_null(row)) {
+                int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[row]
+                                              : c->i64v[row];
+                if (sp.func == GX_AGG_BIT_AND) s.i64[gid] &= v;
+                else if (sp.func == GX_AGG_BIT_OR) s.i64[gid] |= v;
+                else s.i64[gid] ^= v;
+            }
+            break;
         case GX_AGG_AVG_F64:
             if (!c->is_null(row)) {
                 double v = c->type == GX_F64 ? c->f64v[row]
@@ -866,6 +879,7 @@ struct AggOp : gx_op {
             switch (sp.func) {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
             case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
                 otypes.push_back(GX_I64); break;
             default: otypes.push_back(GX_F64); break;
             }
@@ -882,7 +896,10 @@ struct AggOp : gx_op {
                 bool is_i64 = otypes[col] == GX_I64;
                 if (s.isnull[g] &&
                     !(aggs[a].func == GX_AGG_COUNT_ROW || aggs[a].func == GX_AGG_COUNT_COL
-                      || aggs[a].func == GX_AGG_SUM_I64))
+                      || aggs[a].func == GX_AGG_SUM_I64
+                      || aggs[a].func == GX_AGG_BIT_AND
+                      || aggs[a].func == GX_AGG_BIT_OR
+                      || aggs[a].func == GX_AGG_BIT_XOR))
                     cols[col++].append_null();
                 else if (aggs[a].func == GX_AGG_AVG_F64)
                     cols[col++].append_f64(s.f64[g] / (double)s.i64[g]);
@@ -964,12 +981,17 @@ struct GroupJoinOp : gx_op {
             s.f64.assign(build.n_rows, 0);
             bool init_null = !(aggs[a].func == GX_AGG_COUNT_ROW ||
                                aggs[a].func == GX_AGG_COUNT_COL ||
-                               aggs[a].func == GX_AGG_SUM_I64);
+                               aggs[a].func == GX_AGG_SUM_I64 ||
+                               aggs[a].func == GX_AGG_BIT_AND ||
+                               aggs[a].func == GX_AGG_BIT_OR ||
+                               aggs[a].func == GX_AGG_BIT_XOR);
             s.isnull.assign(build.n_rows, init_null ? 1 : 0);
             if (aggs[a].func == GX_AGG_MIN_I64)
                 s.i64.assign(build.n_rows, INT64_MAX);
             if (aggs[a].func == GX_AGG_MAX_I64)
                 s.i64.assign(build.n_rows, INT64_MIN);
+            if (aggs[a].func == GX_AGG_BIT_AND)
+                s.i64.assign(build.n_rows, -1);
         }
         built = true;
         return 0;
@@ -995,6 +1017,15 @@ struct GroupJoinOp : gx_op {
                                              : (double)c->i64v[row];
                 if (s.isnull[gid]) { s.f64[gid] = v; s.isnull[gid] = 0; }
                 else s.f64[gid] += v;
+            }
+            break;
+        case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
+            if (!c->is_null(row)) {
+                int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[row]
+                                              : c->i64v[row];
+                if (sp.func == GX_AGG_BIT_AND) s.i64[gid] &= v;
+                else if (sp.func == GX_AGG_BIT_OR) s.i64[gid] |= v;
+                else s.i64[gid] ^= v;
             }
             break;
         case GX_AGG_AVG_F64:
@@ -1057,6 +1088,7 @@ struct GroupJoinOp : gx_op {
             switch (sp.func) {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
             case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
                 otypes.push_back(GX_I64); break;
             default: otypes.push_back(GX_F64); break;
             }
@@ -1132,7 +1164,10 @@ struct WindowOp : gx_op {
         RunState &s = run[a];
         switch (aggs[a].func) {
         case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL: case GX_AGG_SUM_I64:
+        case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
             s.i64 = 0; s.isnull = 0; break;
+        case GX_AGG_BIT_AND:
+            s.i64 = -1; s.isnull = 0; break;
         default: s.i64 = 0; s.f64 = 0; s.isnull = 1; break; /* incl. AVG:
             i64 = count, f64 = sum */
         }
@@ -1183,6 +1218,15 @@ struct WindowOp : gx_op {
                 s.f64 += v; s.i64++; s.isnull = 0;
             }
             break;
+        case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
+            if (!c->is_null(r)) {
+                int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[r]
+                                              : c->i64v[r];
+                if (sp.func == GX_AGG_BIT_AND) s.i64 &= v;
+                else if (sp.func == GX_AGG_BIT_OR) s.i64 |= v;
+                else s.i64 ^= v;
+            }
+            break;
         }
     }
 
@@ -1229,6 +1273,7 @@ struct WindowOp : gx_op {
             switch (sp.func) {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
             case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
                 otypes.push_back(GX_I64); break;
             default: otypes.push_back(GX_F64); break;
             }
@@ -1312,11 +1357,15 @@ struct FWindowOp : gx_op {
                    int64_t &iv, double &dv, bool &isnull) const {
         bool never_null = f.func == GX_AGG_COUNT_ROW ||
                           f.func == GX_AGG_COUNT_COL ||
-                          f.func == GX_AGG_SUM_I64;
+                          f.func == GX_AGG_SUM_I64 ||
+                          f.func == GX_AGG_BIT_AND ||
+                          f.func == GX_AGG_BIT_OR ||
+                          f.func == GX_AGG_BIT_XOR;
         iv = 0; dv = 0; isnull = !never_null;
         const Column *c = f.input_col >= 0 ? &input.cols[f.input_col] : nullptr;
         if (f.func == GX_AGG_MIN_I64) iv = INT64_MAX;
         if (f.func == GX_AGG_MAX_I64) iv = INT64_MIN;
+        if (f.func == GX_AGG_BIT_AND) iv = -1;
         for (size_t r = lo; r < hi; r++) {
             switch (f.func) {
             case GX_AGG_COUNT_ROW: iv++; break;
@@ -1359,6 +1408,15 @@ struct FWindowOp : gx_op {
                     dv += v; iv++; isnull = false;
                 }
                 break;
+            case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
+                if (!c->is_null(r)) {
+                    int64_t v = c->type == GX_I32 ? (int64_t)c->i32v[r]
+                                                  : c->i64v[r];
+                    if (f.func == GX_AGG_BIT_AND) iv &= v;
+                    else if (f.func == GX_AGG_BIT_OR) iv |= v;
+                    else iv ^= v;
+                }
+                break;
             }
         }
     }
@@ -1374,6 +1432,7 @@ struct FWindowOp : gx_op {
             switch (f.func) {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
             case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
                 otypes.push_back(GX_I64); break;
             default: otypes.push_back(GX_F64); break;
             }

@@ -61,7 +61,8 @@ AGG_FUNCS = [(abi.COUNT_ROW, None), (abi.COUNT_COL, I64),
              (abi.SUM_I64, I64), (abi.SUM_F64, F64),
              (abi.MIN_I64, I64), (abi.MAX_I64, I64),
              (abi.MIN_F64, F64), (abi.MAX_F64, F64),
-             (abi.AVG_F64, F64), (abi.AVG_F64, I64)]
+             (abi.AVG_F64, F64), (abi.AVG_F64, I64),
+             (abi.BIT_AND, I64), (abi.BIT_OR, I64), (abi.BIT_XOR, I64)]
 
 
 def _col(rng, t, n, card, null_frac):

@@ -115,7 +115,8 @@ def test_window_oracle_vs_numpy():
 
 AGGS = [(abi.COUNT_ROW, -1), (abi.COUNT_COL, 1), (abi.SUM_I64, 1),
         (abi.MIN_I64, 1), (abi.MAX_I64, 1), (abi.SUM_F64, 2),
-        (abi.MIN_F64, 2), (abi.AVG_F64, 1), (abi.AVG_F64, 2)]
+        (abi.MIN_F64, 2), (abi.AVG_F64, 1), (abi.AVG_F64, 2),
+        (abi.BIT_AND, 1), (abi.BIT_OR, 1), (abi.BIT_XOR, 1)]
 
 
 @pytest.mark.gpu
@@ -183,3 +184,25 @@ def test_avg_hand_case_oracle():
                              [(abi.AVG_F64, 1, abi.FRAME_WHOLE_PARTITION)],
                              [I64, I64], chunks))
     assert fw == [(1, 10, 7.5), (1, None, 7.5), (1, 5, 7.5), (2, 7, 7.0)]
+
+
+def test_bit_aggs_hand_case_oracle():
+    """BIT_AND/OR/XOR follow the SHIPPED aggregator code
+    (SpecificType2UInt64BitAnd.java:42-60: init all-ones / 0, skip nulls,
+    writeResultTo always emits — NEVER NULL; the @Ignore'd
+    testBitRelated's expected chunk contradicts that code and is not
+    transcribed)."""
+    from galaxysql_amd.operators import run_agg
+    lib = abi.load_oracle()
+    g = [1, 1, 1, 2]
+    v = [0b0110, None, 0b0011, None]
+    chunks = [Chunk([Block.of(I64, g), Block.of(I64, v)])]
+    rows = rows_of(run_agg(lib, [0], [(abi.BIT_AND, 1), (abi.BIT_OR, 1),
+                                      (abi.BIT_XOR, 1)], [I64, I64], chunks))
+    assert multiset(rows) == multiset([
+        (1, 0b0010, 0b0111, 0b0101),
+        (2, -1, 0, 0),            # no non-null input: init values
+    ])
+    win = rows_of(run_window(lib, [0], [(abi.BIT_XOR, 1)], [I64, I64],
+                             chunks))
+    assert win == [(1, 6, 6), (1, None, 6), (1, 3, 5), (2, None, 0)]
