@@ -133,6 +133,8 @@ class GxWindowCfg(C.Structure):
     _fields_ = [
         ("n_part_cols", C.c_int32),
         ("part_cols", C.POINTER(C.c_int32)),
+        ("n_order_cols", C.c_int32),
+        ("order_cols", C.POINTER(C.c_int32)),
         ("n_aggs", C.c_int32),
         ("aggs", C.POINTER(GxAggSpec)),
         ("reset", C.POINTER(C.c_uint8)),
@@ -210,7 +212,7 @@ FRAME_WHOLE_PARTITION, FRAME_ROWS_SLIDING, FRAME_ROWS_UNBOUNDED_FOLLOWING = \
     0, 1, 2
 # Agg funcs (gx_agg_func)
 COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64, \
-    MAX_F64, AVG_F64, BIT_AND, BIT_OR, BIT_XOR = range(12)
+    MAX_F64, AVG_F64, BIT_AND, BIT_OR, BIT_XOR, RANK, DENSE_RANK = range(14)
 
 ORACLE_PATH = os.path.join(_REPO, "oracle", "libgxoracle.so")
 HIP_PATH = os.path.join(_REPO, "galaxysql_amd", "csrc", "libgxhip.so")

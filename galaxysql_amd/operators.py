@@ -270,11 +270,13 @@ class NonFrameOverWindowExec:
     (abi.COUNT_ROW, -1) cumulative."""
 
     def __init__(self, lib, part_cols, aggs, input_types, reset=None,
-                 device=-1, stream=0):
+                 order_cols=None, device=-1, stream=0):
         from .abi import GxWindowCfg
         self._lib = lib
         self._keep = []
         pc = (C.c_int32 * max(1, len(part_cols)))(*(part_cols or [0]))
+        oc_list = order_cols or []
+        oc = (C.c_int32 * max(1, len(oc_list)))(*(oc_list or [0]))
         sp = (GxAggSpec * max(1, len(aggs)))()
         for i, (f, col) in enumerate(aggs):
             sp[i] = GxAggSpec(f, col)
@@ -282,10 +284,11 @@ class NonFrameOverWindowExec:
             *[1 if (reset and reset[i]) else 0 for i in range(len(aggs))])
         it = (C.c_int32 * len(input_types))(*input_types)
         cfg = GxWindowCfg(n_part_cols=len(part_cols), part_cols=pc,
+                          n_order_cols=len(oc_list), order_cols=oc,
                           n_aggs=len(aggs), aggs=sp, reset=rs,
                           n_input_cols=len(input_types), input_types=it,
                           device=device, stream=stream)
-        self._keep += [pc, sp, rs, it, cfg]
+        self._keep += [pc, oc, sp, rs, it, cfg]
         self._op = lib.lib.gxop_window_create(C.byref(cfg))
         if not self._op:
             raise RuntimeError(f"gxop_window_create: {lib.error()}")
@@ -313,9 +316,9 @@ class NonFrameOverWindowExec:
 
 
 def run_window(lib, part_cols, aggs, input_types, input_chunks, reset=None,
-               **kw):
+               order_cols=None, **kw):
     op = NonFrameOverWindowExec(lib, part_cols, aggs, input_types,
-                                reset=reset, **kw)
+                                reset=reset, order_cols=order_cols, **kw)
     try:
         return [op.consume_chunk(ch) for ch in input_chunks]
     finally:

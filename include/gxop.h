@@ -188,9 +188,15 @@ typedef enum gx_agg_func {
                               AND of non-null inputs, NEVER NULL (the
                               reference emits the init value for empty) */
     GX_AGG_BIT_OR    = 10, /* ...BitOr: init 0, never NULL */
-    GX_AGG_BIT_XOR   = 11  /* ...BitXor: init 0, never NULL; the only
+    GX_AGG_BIT_XOR   = 11, /* ...BitXor: init 0, never NULL; the only
                               BIT agg valid in sliding frames (invertible)
                               — still excluded there this round */
+    GX_AGG_RANK      = 12, /* Rank (calc/aggfunctions/Rank.java:40-74):
+                              1-based position of the first row of the
+                              current equal ORDER-BY run; needs
+                              gx_window_cfg.order_cols; window op only */
+    GX_AGG_DENSE_RANK = 13 /* DenseRank: running count of distinct
+                              ORDER-BY runs in the partition */
 } gx_agg_func;
 
 typedef struct gx_agg_spec {
@@ -309,6 +315,10 @@ int gxop_part_close(gx_op *op);
 typedef struct gx_window_cfg {
     int32_t n_part_cols;
     const int32_t *part_cols;
+    /* ORDER BY columns, consulted only by RANK/DENSE_RANK (null-safe
+     * equality per Objects.equals — Rank.sameRank:61-74). 0 = none. */
+    int32_t n_order_cols;
+    const int32_t *order_cols;
     int32_t n_aggs;
     const gx_agg_spec *aggs;   /* window functions over the agg subset */
     const uint8_t *reset;      /* per agg: 1 = CURRENT ROW..CURRENT ROW */

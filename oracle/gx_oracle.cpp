@@ -880,6 +880,7 @@ struct AggOp : gx_op {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
             case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
+            case GX_AGG_RANK: case GX_AGG_DENSE_RANK:
                 otypes.push_back(GX_I64); break;
             default: otypes.push_back(GX_F64); break;
             }
@@ -1089,6 +1090,7 @@ struct GroupJoinOp : gx_op {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
             case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
+            case GX_AGG_RANK: case GX_AGG_DENSE_RANK:
                 otypes.push_back(GX_I64); break;
             default: otypes.push_back(GX_F64); break;
             }
@@ -1140,22 +1142,28 @@ struct GroupJoinOp : gx_op {
  * (last partition key + running state) survives across chunks. */
 struct WindowOp : gx_op {
     gx_window_cfg cfg;
-    std::vector<int32_t> part_cols_, input_types;
+    std::vector<int32_t> part_cols_, order_cols_, carry_cols_, input_types;
     std::vector<gx_agg_spec> aggs;
     std::vector<uint8_t> reset_;
 
     bool have_carry = false;
-    Store carry_key;                 /* 1 row: last partition key cols */
+    Store carry_key;   /* 1 row: LAST ROW's partition + order key cols */
     struct RunState { int64_t i64 = 0; double f64 = 0; uint8_t isnull = 0; };
     std::vector<RunState> run;
+    int64_t part_count = 0;   /* rows since partition start (Rank.count) */
 
     WindowOp(const gx_window_cfg *c) : gx_op(OP_WINDOW), cfg(*c) {
         part_cols_.assign(c->part_cols, c->part_cols + c->n_part_cols);
+        if (c->n_order_cols > 0)
+            order_cols_.assign(c->order_cols, c->order_cols + c->n_order_cols);
         input_types.assign(c->input_types, c->input_types + c->n_input_cols);
         aggs.assign(c->aggs, c->aggs + c->n_aggs);
         reset_.assign(c->reset, c->reset + c->n_aggs);
+        carry_cols_ = part_cols_;
+        carry_cols_.insert(carry_cols_.end(), order_cols_.begin(),
+                           order_cols_.end());
         std::vector<int32_t> kt;
-        for (int32_t pc : part_cols_) kt.push_back(input_types[pc]);
+        for (int32_t pc : carry_cols_) kt.push_back(input_types[pc]);
         carry_key.init((int32_t)kt.size(), kt.data());
         run.resize(aggs.size());
     }
@@ -1238,10 +1246,20 @@ struct WindowOp : gx_op {
         return carry_key.keys_equal(0, in, r, kidx, pcols);
     }
 
+    /* previous row (carry) has the same ORDER-BY values (Rank.sameRank) */
+    bool same_run(const Store &in, size_t r) {
+        if (!have_carry) return false;
+        std::vector<int> kidx;
+        for (size_t k = 0; k < order_cols_.size(); k++)
+            kidx.push_back((int)(part_cols_.size() + k));
+        std::vector<int> ocols(order_cols_.begin(), order_cols_.end());
+        return carry_key.keys_equal(0, in, r, kidx, ocols);
+    }
+
     void save_carry(const Store &in, size_t r) {
-        for (size_t k = 0; k < part_cols_.size(); k++) {
+        for (size_t k = 0; k < carry_cols_.size(); k++) {
             Column &dst = carry_key.cols[k];
-            const Column &src = in.cols[part_cols_[k]];
+            const Column &src = in.cols[carry_cols_[k]];
             dst.i64v.clear(); dst.i32v.clear(); dst.f64v.clear();
             dst.off.clear(); dst.bytes.clear(); dst.null_.clear();
             dst.null_.push_back(src.is_null(r) ? 1 : 0);
@@ -1274,6 +1292,7 @@ struct WindowOp : gx_op {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
             case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
+            case GX_AGG_RANK: case GX_AGG_DENSE_RANK:
                 otypes.push_back(GX_I64); break;
             default: otypes.push_back(GX_F64); break;
             }
@@ -1283,14 +1302,29 @@ struct WindowOp : gx_op {
 
         for (size_t r = 0; r < in.n_rows; r++) {
             bool change = !same_partition(in, r);
-            if (change) save_carry(in, r);
+            bool run_change = change || !same_run(in, r);
+            save_carry(in, r);
+            if (change) part_count = 0;
+            part_count++;
             for (size_t c = 0; c < input_types.size(); c++)
                 cols[c].append_from(in.cols[c], r);
             for (size_t a = 0; a < aggs.size(); a++) {
-                if (reset_[a] || change) reset_state(a);
-                acc(a, in, r);
                 size_t col = input_types.size() + a;
                 RunState &s = run[a];
+                if (aggs[a].func == GX_AGG_RANK ||
+                    aggs[a].func == GX_AGG_DENSE_RANK) {
+                    /* Rank.accumulate:40-47 / DenseRank; resetAccumulators
+                     * makes every row rank 1 (count and lastRow reset) */
+                    if (reset_[a]) { cols[col].append_i64(1); continue; }
+                    if (change) s.i64 = 0;
+                    if (run_change)
+                        s.i64 = aggs[a].func == GX_AGG_RANK ? part_count
+                                                            : s.i64 + 1;
+                    cols[col].append_i64(s.i64);
+                    continue;
+                }
+                if (reset_[a] || change) reset_state(a);
+                acc(a, in, r);
                 bool i64out = otypes[col] == GX_I64;
                 if (s.isnull) cols[col].append_null();
                 else if (aggs[a].func == GX_AGG_AVG_F64)
@@ -1433,6 +1467,7 @@ struct FWindowOp : gx_op {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
             case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
+            case GX_AGG_RANK: case GX_AGG_DENSE_RANK:
                 otypes.push_back(GX_I64); break;
             default: otypes.push_back(GX_F64); break;
             }
@@ -1751,7 +1786,18 @@ int gxop_join_tail(gx_op *op, gx_result **out) {
 }
 int gxop_join_close(gx_op *op) { delete op; return 0; }
 
+static int agg_funcs_ok(const gx_agg_spec *aggs, int32_t n) {
+    for (int32_t i = 0; i < n; i++)
+        if (aggs[i].func == GX_AGG_RANK || aggs[i].func == GX_AGG_DENSE_RANK)
+            return 0;  /* rank family is window-only (needs ORDER BY runs) */
+    return 1;
+}
+
 gx_op *gxop_agg_create(const gx_agg_cfg *cfg) {
+    if (cfg && !agg_funcs_ok(cfg->aggs, cfg->n_aggs)) {
+        set_err("RANK/DENSE_RANK are window-only");
+        return nullptr;
+    }
     if (!cfg) { set_err("bad agg cfg"); return nullptr; }
     return new AggOp(cfg);
 }
@@ -1772,6 +1818,10 @@ int gxop_agg_close(gx_op *op) { delete op; return 0; }
 
 
 gx_op *gxop_groupjoin_create(const gx_groupjoin_cfg *cfg) {
+    if (cfg && !agg_funcs_ok(cfg->aggs, cfg->n_aggs)) {
+        set_err("RANK/DENSE_RANK are window-only");
+        return nullptr;
+    }
     if (!cfg || cfg->n_keys <= 0 ||
         (cfg->join_type != GX_JOIN_INNER && cfg->join_type != GX_JOIN_LEFT)) {
         set_err("bad groupjoin cfg");
@@ -1826,6 +1876,13 @@ static int fwindow_cfg_ok(const gx_fwindow_cfg *c) {
 }
 
 gx_op *gxop_fwindow_create(const gx_fwindow_cfg *cfg) {
+    if (cfg)
+        for (int32_t i = 0; i < cfg->n_frames; i++)
+            if (cfg->frames[i].func == GX_AGG_RANK ||
+                cfg->frames[i].func == GX_AGG_DENSE_RANK) {
+                set_err("RANK/DENSE_RANK are window-only");
+                return nullptr;
+            }
     if (!fwindow_cfg_ok(cfg)) { set_err("bad fwindow cfg (unsupported frame/func)"); return nullptr; }
     return new FWindowOp(cfg);
 }

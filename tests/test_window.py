@@ -206,3 +206,55 @@ def test_bit_aggs_hand_case_oracle():
     win = rows_of(run_window(lib, [0], [(abi.BIT_XOR, 1)], [I64, I64],
                              chunks))
     assert win == [(1, 6, 6), (1, None, 6), (1, 3, 5), (2, None, 0)]
+
+
+RANK_AGGS = [(abi.RANK, -1), (abi.DENSE_RANK, -1), (abi.COUNT_ROW, -1)]
+
+
+def test_rank_hand_case_oracle():
+    """Rank/DenseRank (Rank.java:40-74): 1-based first-of-run position /
+    distinct-run count, null-safe order equality, reset per partition."""
+    lib = abi.load_oracle()
+    part = [1] * 6 + [2] * 2
+    order = [5, 5, 7, 7, 7, 9, 1, 1]
+    chunks = chunks_from_columns([I64, I64],
+                                 [(np.array(part, np.int64), None),
+                                  (np.array(order, np.int64), None)],
+                                 chunk_size=3)  # run straddles chunks
+    rows = rows_of(run_window(lib, [0], RANK_AGGS, [I64, I64], chunks,
+                              order_cols=[1]))
+    assert [(r[2], r[3], r[4]) for r in rows] == [
+        (1, 1, 1), (1, 1, 2), (3, 2, 3), (3, 2, 4), (3, 2, 5), (6, 3, 6),
+        (1, 1, 1), (1, 1, 2)]
+
+
+def test_rank_rejected_outside_window():
+    from galaxysql_amd.operators import run_agg
+    lib = abi.load_oracle()
+    with pytest.raises(RuntimeError):
+        run_agg(lib, [0], [(abi.RANK, -1)], [I64],
+                [Chunk([Block.of(I64, [1])])])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("chunk_size", [4000, 13])
+def test_gpu_rank_matches_oracle(chunk_size):
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    rng = np.random.default_rng(44)
+    n = 30000
+    parts = np.sort(rng.integers(0, n // 40, n)).astype(np.int64)
+    # order col sorted WITHIN partitions with repeats
+    order = np.concatenate([np.sort(rng.integers(0, 9, (parts == p).sum()))
+                            for p in np.unique(parts)]).astype(np.int64)
+    onulls = (rng.random(n) < 0.05).astype(np.uint8)
+    # nulls sorted to the front within partition? keep arbitrary: both
+    # impls compare the same sequence, so parity holds regardless
+    chunks = chunks_from_columns([I64, I64],
+                                 [(parts, None), (order, onulls)],
+                                 chunk_size=chunk_size)
+    got = rows_of(run_window(hip, [0], RANK_AGGS, [I64, I64], chunks,
+                             order_cols=[1], device=0))
+    want = rows_of(run_window(ora, [0], RANK_AGGS, [I64, I64], chunks,
+                              order_cols=[1], device=-1))
+    assert got == want
