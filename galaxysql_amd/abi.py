@@ -159,6 +159,8 @@ class GxFWindowCfg(C.Structure):
     _fields_ = [
         ("n_part_cols", C.c_int32),
         ("part_cols", C.POINTER(C.c_int32)),
+        ("n_order_cols", C.c_int32),
+        ("order_cols", C.POINTER(C.c_int32)),
         ("n_frames", C.c_int32),
         ("frames", C.POINTER(GxFrameSpec)),
         ("n_input_cols", C.c_int32),
@@ -211,8 +213,10 @@ PROJ_DEC_TO_SCALED, PROJ_SCALED_TO_DEC = 4, 5  # gx_proj.c = decimal scale
 FRAME_WHOLE_PARTITION, FRAME_ROWS_SLIDING, FRAME_ROWS_UNBOUNDED_FOLLOWING = \
     0, 1, 2
 # Agg funcs (gx_agg_func)
-COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64, \
-    MAX_F64, AVG_F64, BIT_AND, BIT_OR, BIT_XOR, RANK, DENSE_RANK = range(14)
+(COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64,
+ MAX_F64, AVG_F64, BIT_AND, BIT_OR, BIT_XOR, RANK, DENSE_RANK,
+ FIRST_VALUE, LAST_VALUE, NTH_VALUE, LAG, LEAD, NTILE, CUME_DIST,
+ PERCENT_RANK) = range(22)
 
 ORACLE_PATH = os.path.join(_REPO, "oracle", "libgxoracle.so")
 HIP_PATH = os.path.join(_REPO, "galaxysql_amd", "csrc", "libgxhip.so")

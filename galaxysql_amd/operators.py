@@ -331,21 +331,24 @@ class OverWindowFramesExec:
     kind, preceding, following)."""
 
     def __init__(self, lib, part_cols, frames, input_types, device=-1,
-                 stream=0):
+                 order_cols=None, stream=0):
         from .abi import GxFrameSpec, GxFWindowCfg
         self._lib = lib
         self._keep = []
         pc = (C.c_int32 * max(1, len(part_cols)))(*(part_cols or [0]))
+        oc_list = order_cols or []
+        oc = (C.c_int32 * max(1, len(oc_list)))(*(oc_list or [0]))
         fr = (GxFrameSpec * len(frames))()
         for i, spec in enumerate(frames):
             spec = tuple(spec) + (0,) * (5 - len(spec))
             fr[i] = GxFrameSpec(*spec[:5])
         it = (C.c_int32 * len(input_types))(*input_types)
         cfg = GxFWindowCfg(n_part_cols=len(part_cols), part_cols=pc,
+                           n_order_cols=len(oc_list), order_cols=oc,
                            n_frames=len(frames), frames=fr,
                            n_input_cols=len(input_types), input_types=it,
                            device=device, stream=stream)
-        self._keep += [pc, fr, it, cfg]
+        self._keep += [pc, oc, fr, it, cfg]
         self._op = lib.lib.gxop_fwindow_create(C.byref(cfg))
         if not self._op:
             raise RuntimeError(f"gxop_fwindow_create: {lib.error()}")
@@ -385,8 +388,10 @@ class OverWindowFramesExec:
             pass
 
 
-def run_fwindow(lib, part_cols, frames, input_types, input_chunks, **kw):
-    op = OverWindowFramesExec(lib, part_cols, frames, input_types, **kw)
+def run_fwindow(lib, part_cols, frames, input_types, input_chunks,
+                order_cols=None, **kw):
+    op = OverWindowFramesExec(lib, part_cols, frames, input_types,
+                              order_cols=order_cols, **kw)
     try:
         for ch in input_chunks:
             op.consume_chunk(ch)

@@ -195,8 +195,24 @@ typedef enum gx_agg_func {
                               1-based position of the first row of the
                               current equal ORDER-BY run; needs
                               gx_window_cfg.order_cols; window op only */
-    GX_AGG_DENSE_RANK = 13 /* DenseRank: running count of distinct
+    GX_AGG_DENSE_RANK = 13,/* DenseRank: running count of distinct
                               ORDER-BY runs in the partition */
+    /* navigation / distribution functions (calc/aggfunctions/
+     * FirstValue|LastValue|NThValue|Lag|Lead|NTile|CumeDist|PercentRank
+     * .java) — FRAME-WINDOW ONLY (they need partition bounds, so they run
+     * in the buffered gxop_fwindow op; gx_frame_spec.preceding carries the
+     * parameter: NTH's n, LAG/LEAD's offset, NTILE's bucket count).
+     * FIRST/LAST/NTH/LAG/LEAD emit the INPUT column's type (any type,
+     * strings included — they are index gathers); NTILE emits BIGINT;
+     * CUME_DIST/PERCENT_RANK emit DOUBLE and need fwindow order_cols. */
+    GX_AGG_FIRST_VALUE = 14,
+    GX_AGG_LAST_VALUE  = 15,
+    GX_AGG_NTH_VALUE   = 16, /* 1-based n; NULL beyond the frame */
+    GX_AGG_LAG         = 17, /* NULL before partition start */
+    GX_AGG_LEAD        = 18, /* NULL past partition end */
+    GX_AGG_NTILE       = 19, /* MySQL split: first size%n buckets get +1 */
+    GX_AGG_CUME_DIST   = 20, /* rows <= current ORDER run / partition size */
+    GX_AGG_PERCENT_RANK = 21 /* (rank-1) / (partition size-1); 0 if size 1 */
 } gx_agg_func;
 
 typedef struct gx_agg_spec {
@@ -369,6 +385,10 @@ typedef struct gx_frame_spec {
 typedef struct gx_fwindow_cfg {
     int32_t n_part_cols;
     const int32_t *part_cols;
+    /* ORDER BY run detection for CUME_DIST / PERCENT_RANK (null-safe
+     * equality); others ignore it. */
+    int32_t n_order_cols;
+    const int32_t *order_cols;
     int32_t n_frames;
     const gx_frame_spec *frames;
     int32_t n_input_cols;

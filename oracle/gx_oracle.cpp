@@ -1346,17 +1346,21 @@ struct WindowOp : gx_op {
  * Emission preserves input order (the operator appends window columns). */
 struct FWindowOp : gx_op {
     gx_fwindow_cfg cfg;
-    std::vector<int32_t> part_cols_, input_types;
+    std::vector<int32_t> part_cols_, order_cols_, input_types;
     std::vector<gx_frame_spec> frames;
     Store input;
     bool finished = false;
     size_t emit_cursor = 0;
     std::vector<size_t> seg_start_of_row;   /* first row of my segment */
     std::vector<size_t> seg_end_of_row;     /* one past last row */
+    std::vector<size_t> run_start_of_row;   /* ORDER-BY run bounds */
+    std::vector<size_t> run_end_of_row;     /* INCLUSIVE last row of run */
     static constexpr int32_t CHUNK_SIZE = 1000;
 
     FWindowOp(const gx_fwindow_cfg *c) : gx_op(OP_FWINDOW), cfg(*c) {
         part_cols_.assign(c->part_cols, c->part_cols + c->n_part_cols);
+        if (c->n_order_cols > 0)
+            order_cols_.assign(c->order_cols, c->order_cols + c->n_order_cols);
         input_types.assign(c->input_types, c->input_types + c->n_input_cols);
         frames.assign(c->frames, c->frames + c->n_frames);
         input.init((int32_t)input_types.size(), input_types.data());
@@ -1382,6 +1386,21 @@ struct FWindowOp : gx_op {
             seg_start_of_row[i] = start;
         }
         for (size_t j = start; j < n; j++) seg_end_of_row[j] = n;
+        if (!order_cols_.empty()) {
+            std::vector<int> oc(order_cols_.begin(), order_cols_.end());
+            run_start_of_row.resize(n);
+            run_end_of_row.resize(n);
+            size_t rs = 0;
+            for (size_t i = 0; i < n; i++) {
+                if (i > 0 && (seg_start_of_row[i] != seg_start_of_row[i - 1] ||
+                              !input.keys_equal(i - 1, input, i, oc, oc))) {
+                    for (size_t j = rs; j < i; j++) run_end_of_row[j] = i - 1;
+                    rs = i;
+                }
+                run_start_of_row[i] = rs;
+            }
+            for (size_t j = rs; j < n; j++) run_end_of_row[j] = n - 1;
+        }
         finished = true;
         return 0;
     }
@@ -1467,8 +1486,11 @@ struct FWindowOp : gx_op {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
             case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
-            case GX_AGG_RANK: case GX_AGG_DENSE_RANK:
+            case GX_AGG_RANK: case GX_AGG_DENSE_RANK: case GX_AGG_NTILE:
                 otypes.push_back(GX_I64); break;
+            case GX_AGG_FIRST_VALUE: case GX_AGG_LAST_VALUE:
+            case GX_AGG_NTH_VALUE: case GX_AGG_LAG: case GX_AGG_LEAD:
+                otypes.push_back(input_types[f.input_col]); break;
             default: otypes.push_back(GX_F64); break;
             }
         }
@@ -1481,6 +1503,52 @@ struct FWindowOp : gx_op {
             for (size_t a = 0; a < frames.size(); a++) {
                 const gx_frame_spec &f = frames[a];
                 size_t s = seg_start_of_row[i], e = seg_end_of_row[i];
+                size_t col2 = input_types.size() + a;
+                if (f.func >= GX_AGG_FIRST_VALUE) {
+                    /* navigation / distribution over the whole partition */
+                    int64_t size = (int64_t)(e - s), pos = (int64_t)(i - s);
+                    int64_t src = -1;
+                    switch (f.func) {
+                    case GX_AGG_FIRST_VALUE: src = (int64_t)s; break;
+                    case GX_AGG_LAST_VALUE: src = (int64_t)e - 1; break;
+                    case GX_AGG_NTH_VALUE:
+                        src = (int64_t)s + f.preceding - 1;
+                        if (src >= (int64_t)e) src = -1;
+                        break;
+                    case GX_AGG_LAG:
+                        src = (int64_t)i - f.preceding;
+                        if (src < (int64_t)s) src = -1;
+                        break;
+                    case GX_AGG_LEAD:
+                        src = (int64_t)i + f.preceding;
+                        if (src >= (int64_t)e) src = -1;
+                        break;
+                    case GX_AGG_NTILE: {
+                        int64_t nt = f.preceding;
+                        int64_t base = size / nt, rem = size % nt;
+                        int64_t bucket;
+                        if (base == 0) bucket = pos + 1;
+                        else if (pos < rem * (base + 1))
+                            bucket = pos / (base + 1) + 1;
+                        else bucket = rem + (pos - rem * (base + 1)) / base + 1;
+                        cols[col2].append_i64(bucket);
+                        continue; }
+                    case GX_AGG_CUME_DIST:
+                        cols[col2].append_f64(
+                            (double)((int64_t)run_end_of_row[i] -
+                                     (int64_t)s + 1) / (double)size);
+                        continue;
+                    case GX_AGG_PERCENT_RANK:
+                        cols[col2].append_f64(size <= 1 ? 0.0 :
+                            (double)((int64_t)run_start_of_row[i] -
+                                     (int64_t)s) / (double)(size - 1));
+                        continue;
+                    }
+                    if (src < 0) cols[col2].append_null();
+                    else cols[col2].append_from(input.cols[f.input_col],
+                                                (size_t)src);
+                    continue;
+                }
                 size_t lo = s, hi = e;
                 if (f.kind == GX_FRAME_ROWS_SLIDING) {
                     lo = i >= s + (size_t)f.preceding ? i - (size_t)f.preceding : s;
@@ -1863,6 +1931,19 @@ static int fwindow_cfg_ok(const gx_fwindow_cfg *c) {
     if (!c || c->n_frames <= 0) return 0;
     for (int32_t i = 0; i < c->n_frames; i++) {
         const gx_frame_spec &f = c->frames[i];
+        if (f.func >= GX_AGG_FIRST_VALUE) {
+            /* navigation funcs: whole-partition only; NTH/LAG/LEAD/NTILE
+             * need a positive parameter; CUME/PERCENT need ORDER cols */
+            if (f.kind != GX_FRAME_WHOLE_PARTITION) return 0;
+            if ((f.func == GX_AGG_NTH_VALUE || f.func == GX_AGG_LAG ||
+                 f.func == GX_AGG_LEAD || f.func == GX_AGG_NTILE) &&
+                f.preceding <= 0)
+                return 0;
+            if ((f.func == GX_AGG_CUME_DIST ||
+                 f.func == GX_AGG_PERCENT_RANK) && c->n_order_cols <= 0)
+                return 0;
+            continue;
+        }
         if (f.kind == GX_FRAME_WHOLE_PARTITION) continue;
         /* sliding / following: exact-additive funcs only (see gxop.h) */
         if (f.func != GX_AGG_COUNT_ROW && f.func != GX_AGG_COUNT_COL &&

@@ -103,3 +103,72 @@ def test_gpu_fwindow_matches_oracle():
                 assert abs(a - b) < 1e-9, (i, g, w)
             else:
                 assert a == b, (i, g, w)
+
+
+def test_navigation_hand_case_oracle():
+    """FIRST/LAST/NTH_VALUE, LAG/LEAD, NTILE, CUME_DIST, PERCENT_RANK
+    (calc/aggfunctions/FirstValue|Lag|NTile|CumeDist|PercentRank.java),
+    whole-partition navigation over buffered input — strings included
+    (index gathers)."""
+    from galaxysql_amd.chunk import SLICE
+    lib = abi.load_oracle()
+    part = [1] * 5 + [2] * 2
+    order = [3, 3, 5, 5, 9, 1, 1]
+    name = ["a", "b", "c", "d", "e", "f", "g"]
+    chunks = [Chunk([Block.of(I64, part), Block.of(I64, order),
+                     Block.of(SLICE, name)])]
+    W = abi.FRAME_WHOLE_PARTITION
+    out = rows_of(run_fwindow(
+        lib, [0],
+        [(abi.FIRST_VALUE, 2, W), (abi.LAST_VALUE, 2, W),
+         (abi.NTH_VALUE, 2, W, 2), (abi.LAG, 1, W, 1),
+         (abi.LEAD, 2, W, 2), (abi.NTILE, -1, W, 2),
+         (abi.CUME_DIST, -1, W), (abi.PERCENT_RANK, -1, W)],
+        [I64, I64, SLICE], chunks, order_cols=[1]))
+    assert out == [
+        (1, 3, b"a", b"a", b"e", b"b", None, b"c", 1, 0.4, 0.0),
+        (1, 3, b"b", b"a", b"e", b"b", 3, b"d", 1, 0.4, 0.0),
+        (1, 5, b"c", b"a", b"e", b"b", 3, b"e", 1, 0.8, 0.5),
+        (1, 5, b"d", b"a", b"e", b"b", 5, None, 2, 0.8, 0.5),
+        (1, 9, b"e", b"a", b"e", b"b", 5, None, 2, 1.0, 1.0),
+        (2, 1, b"f", b"f", b"g", b"g", None, None, 1, 1.0, 0.0),
+        (2, 1, b"g", b"f", b"g", b"g", 1, None, 2, 1.0, 0.0),
+    ]
+
+
+NAV_FRAMES = None
+
+
+@pytest.mark.gpu
+def test_gpu_navigation_matches_oracle():
+    from galaxysql_amd.chunk import SLICE
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    rng = np.random.default_rng(53)
+    n = 20000
+    parts = np.sort(rng.integers(0, n // 31, n)).astype(np.int64)
+    order = np.concatenate([np.sort(rng.integers(0, 7, (parts == p).sum()))
+                            for p in np.unique(parts)]).astype(np.int64)
+    vals = rng.integers(-50, 50, n)
+    vnulls = (rng.random(n) < 0.1).astype(np.uint8)
+    names = Block.of(SLICE, [f"n{int(v) % 13}" for v in vals])
+    chunks = chunks_from_columns(
+        [I64, I64, I64, SLICE],
+        [(parts, None), (order, None), (vals, vnulls), names],
+        chunk_size=777)
+    W = abi.FRAME_WHOLE_PARTITION
+    frames = [(abi.FIRST_VALUE, 2, W), (abi.LAST_VALUE, 3, W),
+              (abi.NTH_VALUE, 3, W, 3), (abi.LAG, 2, W, 2),
+              (abi.LEAD, 3, W, 1), (abi.NTILE, -1, W, 4),
+              (abi.CUME_DIST, -1, W), (abi.PERCENT_RANK, -1, W)]
+    got = rows_of(run_fwindow(hip, [0], frames, [I64, I64, I64, SLICE],
+                              chunks, order_cols=[1], device=0))
+    want = rows_of(run_fwindow(ora, [0], frames, [I64, I64, I64, SLICE],
+                               chunks, order_cols=[1], device=-1))
+    assert len(got) == len(want)
+    for i, (g, w) in enumerate(zip(got, want)):
+        for a, b in zip(g, w):
+            if isinstance(a, float) and b is not None:
+                assert abs(a - b) < 1e-12, (i, g, w)
+            else:
+                assert a == b, (i, g, w)
