@@ -361,13 +361,14 @@ int gxop_window_close(gx_op *op);
  *   GX_FRAME_WHOLE_PARTITION — UnboundedOverFrame: every row gets the
  *     partition total (all 8 agg funcs);
  *   GX_FRAME_ROWS_SLIDING — RowSlidingOverFrame: ROWS BETWEEN p PRECEDING
- *     AND f FOLLOWING, exact-additive funcs only (COUNT_ROW / COUNT_COL /
- *     SUM_I64) via segmented prefix differences;
+ *     AND f FOLLOWING: exact-additive funcs (COUNT_ROW / COUNT_COL /
+ *     SUM_I64) via segmented prefix differences, and MIN/MAX (i64/f64)
+ *     via sparse-table range queries (the fixed span bounds the level
+ *     count);
  *   GX_FRAME_ROWS_UNBOUNDED_FOLLOWING — RowUnboundedFollowingOverFrame:
- *     CURRENT ROW .. UNBOUNDED FOLLOWING, same additive funcs.
- * RANGE frames, sliding MIN/MAX (needs a deque/sparse-table pass) and
- * sliding SUM(DOUBLE) (prefix differences change fp rounding order) are
- * round-2; the create call rejects those combinations loudly. */
+ *     CURRENT ROW .. UNBOUNDED FOLLOWING, additive funcs.
+ * RANGE frames and sliding SUM(DOUBLE) (prefix differences change fp
+ * rounding order) are round-2; the create call rejects those loudly. */
 typedef enum gx_frame_kind {
     GX_FRAME_WHOLE_PARTITION = 0,
     GX_FRAME_ROWS_SLIDING = 1,

@@ -1945,13 +1945,19 @@ static int fwindow_cfg_ok(const gx_fwindow_cfg *c) {
             continue;
         }
         if (f.kind == GX_FRAME_WHOLE_PARTITION) continue;
-        /* sliding / following: exact-additive funcs only (see gxop.h) */
-        if (f.func != GX_AGG_COUNT_ROW && f.func != GX_AGG_COUNT_COL &&
-            f.func != GX_AGG_SUM_I64)
-            return 0;
-        if (f.kind == GX_FRAME_ROWS_SLIDING &&
-            (f.preceding < 0 || f.following < 0))
-            return 0;
+        /* sliding: additive funcs via prefix differences, MIN/MAX via
+         * sparse tables (fixed span); following: additive only */
+        bool additive = f.func == GX_AGG_COUNT_ROW ||
+                        f.func == GX_AGG_COUNT_COL ||
+                        f.func == GX_AGG_SUM_I64;
+        bool minmax = f.func == GX_AGG_MIN_I64 || f.func == GX_AGG_MAX_I64 ||
+                      f.func == GX_AGG_MIN_F64 || f.func == GX_AGG_MAX_F64;
+        if (f.kind == GX_FRAME_ROWS_SLIDING) {
+            if (!additive && !minmax) return 0;
+            if (f.preceding < 0 || f.following < 0) return 0;
+        } else {
+            if (!additive) return 0;
+        }
     }
     return 1;
 }

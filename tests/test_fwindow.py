@@ -41,8 +41,14 @@ def test_fwindow_hand_case():
 def test_fwindow_rejects_unsupported():
     lib = abi.load_oracle()
     with pytest.raises(RuntimeError):
+        # sliding SUM(DOUBLE): prefix differences change fp rounding
         run_fwindow(lib, [0],
-                    [(abi.MIN_I64, 1, abi.FRAME_ROWS_SLIDING, 1, 1)],
+                    [(abi.SUM_F64, 1, abi.FRAME_ROWS_SLIDING, 1, 1)],
+                    [I64, F64], [])
+    with pytest.raises(RuntimeError):
+        # MIN in an unbounded-following frame is still round-2
+        run_fwindow(lib, [0],
+                    [(abi.MIN_I64, 1, abi.FRAME_ROWS_UNBOUNDED_FOLLOWING)],
                     [I64, I64], [])
 
 
@@ -83,6 +89,9 @@ FRAMES = [(abi.SUM_I64, 1, abi.FRAME_WHOLE_PARTITION),
           (abi.MAX_I64, 1, abi.FRAME_WHOLE_PARTITION),
           (abi.SUM_I64, 1, abi.FRAME_ROWS_SLIDING, 3, 2),
           (abi.COUNT_ROW, -1, abi.FRAME_ROWS_SLIDING, 0, 0),
+          (abi.MIN_I64, 1, abi.FRAME_ROWS_SLIDING, 4, 1),
+          (abi.MAX_F64, 2, abi.FRAME_ROWS_SLIDING, 2, 6),
+          (abi.MIN_F64, 2, abi.FRAME_ROWS_SLIDING, 0, 3),
           (abi.SUM_I64, 1, abi.FRAME_ROWS_UNBOUNDED_FOLLOWING)]
 
 
