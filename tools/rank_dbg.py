@@ -1,24 +1,25 @@
 import sys, numpy as np
 sys.path.insert(0, "/root/repo")
 from galaxysql_amd import abi
-from galaxysql_amd.chunk import I64, rows_of, chunks_from_columns
-from galaxysql_amd.operators import run_window
-rng = np.random.default_rng(44)
+from galaxysql_amd.chunk import I64, F64, rows_of, chunks_from_columns
+from galaxysql_amd.operators import run_fwindow
+rng = np.random.default_rng(52)
 n = 30000
-parts = np.sort(rng.integers(0, n // 40, n)).astype(np.int64)
-order = np.concatenate([np.sort(rng.integers(0, 9, (parts == p).sum()))
-                        for p in np.unique(parts)]).astype(np.int64)
-onulls = (rng.random(n) < 0.05).astype(np.uint8)
-chunks = chunks_from_columns([I64, I64], [(parts, None), (order, onulls)],
-                             chunk_size=13)
-AGGS = [(abi.RANK, -1), (abi.DENSE_RANK, -1), (abi.COUNT_ROW, -1)]
-got = rows_of(run_window(abi.load_hip(), [0], AGGS, [I64, I64], chunks,
-                         order_cols=[1], device=0))
-want = rows_of(run_window(abi.load_oracle(), [0], AGGS, [I64, I64], chunks,
-                          order_cols=[1], device=-1))
-bad = [i for i, (g, w) in enumerate(zip(got, want)) if g != w]
-print("mismatches:", len(bad), "first:", bad[:5])
-for i in bad[:5]:
-    print(i, "got", got[i], "want", want[i], "ctx",
-          [(int(parts[j]), None if onulls[j] else int(order[j]))
-           for j in range(max(0, i - 3), min(n, i + 2))])
+parts = np.sort(rng.integers(0, n // 23 + 1, n)).astype(np.int64)
+vals = rng.integers(-50, 50, n).astype(np.int64)
+nulls = (rng.random(n) < 0.1).astype(np.uint8)
+fvals = np.round(rng.random(n) * 7, 3)
+chunks = chunks_from_columns([I64, I64, F64],
+                             [(parts, None), (vals, nulls), (fvals, None)],
+                             chunk_size=997)
+W = abi.FRAME_WHOLE_PARTITION
+FR = [(abi.SUM_F64, 2, W),
+      (abi.MIN_I64, 1, abi.FRAME_ROWS_SLIDING, 4, 1),
+      (abi.MAX_F64, 2, abi.FRAME_ROWS_SLIDING, 2, 6),
+      (abi.MIN_F64, 2, abi.FRAME_ROWS_SLIDING, 0, 3)]
+for lib, dev, nm in [(abi.load_oracle(), -1, "ora"), (abi.load_hip(), 0, "hip")]:
+    out = rows_of(run_fwindow(lib, [0], FR, [I64, I64, F64], chunks, device=dev))
+    print(nm, [tuple(round(x, 3) if isinstance(x, float) else x for x in r[3:])
+               for r in out[:4]])
+print("inputs", [(int(parts[i]), None if nulls[i] else int(vals[i]),
+                  float(fvals[i])) for i in range(6)])
