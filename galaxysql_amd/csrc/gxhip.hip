@@ -33,7 +33,6 @@
 #include <algorithm>
 #include <cstdlib>
 #include <deque>
-#include <functional>
 #include <hipcub/hipcub.hpp>
 
 thread_local std::string gx_err;
@@ -842,11 +841,9 @@ struct JoinOp : gx_op {
     bool built = false;
     bool pass_nothing = false, pass_through = false;
     bool tail_done = false;
-    /* group-join hooks: null-safe key matching, and a pair sink that
-     * replaces materialize() (receives staged probe chunk + pair lists) */
+    /* group-join hook: HashGroupJoinExec matching is null-safe
+     * (Chunk.equals) — build inserts every row and equal NULL keys match */
     bool null_safe_keys = false;
-    std::function<int(const StagedChunk &, const uint32_t *,
-                      const uint32_t *, int64_t)> pair_sink;
 
     /* live probe-kernel stats for bench.py's roofline leg */
     hipEvent_t ev0 = nullptr, ev1 = nullptr;
@@ -1252,14 +1249,6 @@ struct JoinOp : gx_op {
                 if (meta[0] <= cap) {
                     probe_rows_total += n;
                     matches_total += meta[0];
-                    if (pair_sink) {
-                        if (pair_sink(probe_st, (const uint32_t *)d_pidx.p,
-                                      (const uint32_t *)d_bpos.p,
-                                      (int64_t)meta[0])) { attempt = 99; break; }
-                        HIP_OK(hipStreamSynchronize(stream));
-                        rc = 0;
-                        break;
-                    }
                     HipResult *h = materialize(probe_st, (uint32_t *)d_pidx.p,
                                                (uint32_t *)d_bpos.p, meta[0]);
                     if (!h) { attempt = 99; break; }

@@ -88,9 +88,6 @@ def _side(rng, key_types, n, card, null_frac, extra_payloads):
         types.append(t)
         blocks.append(_col(rng, t, n, 1000, null_frac / 2))
     chunk_size = int(rng.integers(100, 2000))
-    cols = []
-    for b in blocks:
-        cols.append(b)
     chunks = []
     for start in range(0, n, chunk_size):
         sub = []
