@@ -267,7 +267,8 @@ class NonFrameOverWindowExec:
     """Running window aggregates over partition-sorted input
     (NonFrameOverWindowExec.java:34-160). aggs: (abi.COUNT_ROW.., col);
     reset[a]=True = CURRENT ROW..CURRENT ROW mode. ROW_NUMBER() =
-    (abi.COUNT_ROW, -1) cumulative."""
+    (abi.COUNT_ROW, -1) cumulative; RANK/DENSE_RANK consult order_cols
+    (null-safe run detection)."""
 
     def __init__(self, lib, part_cols, aggs, input_types, reset=None,
                  order_cols=None, device=-1, stream=0):
@@ -328,7 +329,8 @@ def run_window(lib, part_cols, aggs, input_types, input_chunks, reset=None,
 class OverWindowFramesExec:
     """Frame windows (OverWindowFramesExec.java + operator/frame/):
     consume* -> finish barrier -> result_chunks. frames: (func, input_col,
-    kind, preceding, following)."""
+    kind, preceding, following); for NTH_VALUE/LAG/LEAD/NTILE `preceding`
+    carries the parameter; CUME_DIST/PERCENT_RANK need order_cols."""
 
     def __init__(self, lib, part_cols, frames, input_types, device=-1,
                  order_cols=None, stream=0):
