@@ -585,6 +585,62 @@ static void t_window(int device) {
                 (long long)checked);
 }
 
+/* rank family over sorted runs, closed form */
+static void t_rank(int device) {
+    const int N = 20000, PARTLEN = 40, RUNLEN = 5;
+    std::vector<int64_t> part(N), ord(N);
+    for (int i = 0; i < N; i++) {
+        part[i] = i / PARTLEN;
+        ord[i] = (i % PARTLEN) / RUNLEN;
+    }
+    int32_t pcols[1] = {0};
+    int32_t ocols[1] = {1};
+    gx_agg_spec sp[2] = {{GX_AGG_RANK, -1}, {GX_AGG_DENSE_RANK, -1}};
+    uint8_t rs[2] = {0, 0};
+    int32_t itypes[2] = {GX_I64, GX_I64};
+    gx_window_cfg cfg{};
+    cfg.n_part_cols = 1; cfg.part_cols = pcols;
+    cfg.n_order_cols = 1; cfg.order_cols = ocols;
+    cfg.n_aggs = 2; cfg.aggs = sp; cfg.reset = rs;
+    cfg.n_input_cols = 2; cfg.input_types = itypes;
+    cfg.device = device;
+    gx_op *op = api.win_create(&cfg);
+    CHECK(op, "rank window_create");
+    if (!op) return;
+    int64_t checked = 0;
+    const int SPLIT = N / 2 + 7; /* partition AND run straddle the carry */
+    for (int c = 0; c < 2; c++) {
+        int from = c == 0 ? 0 : SPLIT;
+        int to = c == 0 ? SPLIT : N;
+        gx_block bb[2] = {mk_i64(part.data() + from),
+                          mk_i64(ord.data() + from)};
+        gx_chunk ch{to - from, 2, bb};
+        gx_result *res = nullptr;
+        CHECK(api.win_consume(op, &ch, &res) == 0, "rank consume");
+        if (!res) continue;
+        api.result_to_host(res);
+        for (int r = 0; r < res->chunk.n_rows; r++, checked++) {
+            int64_t gi = from + r;
+            int64_t pos = gi % PARTLEN;
+            int64_t exp_rank = (pos / RUNLEN) * RUNLEN + 1;
+            int64_t exp_dense = pos / RUNLEN + 1;
+            if (cell_i64(res, 2, r) != exp_rank ||
+                cell_i64(res, 3, r) != exp_dense) {
+                CHECK(false, "rank row %lld: (%lld,%lld) != (%lld,%lld)",
+                      (long long)gi, (long long)cell_i64(res, 2, r),
+                      (long long)cell_i64(res, 3, r), (long long)exp_rank,
+                      (long long)exp_dense);
+                break;
+            }
+        }
+        api.result_release(res);
+    }
+    CHECK(checked == N, "rank rows");
+    api.win_close(op);
+    std::printf("  rank: %lld rank/dense_rank values ok\n",
+                (long long)checked);
+}
+
 /* frame window: whole-partition totals + sliding counts, closed form */
 static void t_fwindow(int device) {
     const int N = 30000, PARTLEN = 29;
@@ -768,6 +824,7 @@ int main(int argc, char **argv) {
         t_groupjoin(device);
         t_window(device);
         t_fwindow(device);
+        t_rank(device);
         t_serde();
         if (g_fail) {
             std::printf("SELFTEST FAILED: %d check(s)\n", g_fail);
