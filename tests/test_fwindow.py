@@ -181,3 +181,37 @@ def test_gpu_navigation_matches_oracle():
                 assert abs(a - b) < 1e-12, (i, g, w)
             else:
                 assert a == b, (i, g, w)
+
+
+def test_ntile_cume_percent_vs_numpy():
+    """NTILE/CUME_DIST/PERCENT_RANK against direct numpy formulas."""
+    lib = abi.load_oracle()
+    rng = np.random.default_rng(54)
+    n = 3000
+    parts = np.sort(rng.integers(0, 60, n)).astype(np.int64)
+    order = np.concatenate([np.sort(rng.integers(0, 5, (parts == p).sum()))
+                            for p in np.unique(parts)]).astype(np.int64)
+    chunks = chunks_from_columns([I64, I64], [(parts, None), (order, None)])
+    W = abi.FRAME_WHOLE_PARTITION
+    rows = rows_of(run_fwindow(lib, [0],
+                               [(abi.NTILE, -1, W, 3),
+                                (abi.CUME_DIST, -1, W),
+                                (abi.PERCENT_RANK, -1, W)],
+                               [I64, I64], chunks, order_cols=[1]))
+    for i, r in enumerate(rows):
+        seg = np.nonzero(parts == parts[i])[0]
+        size = len(seg)
+        pos = i - seg[0]
+        base, rem = divmod(size, 3)
+        if base == 0:
+            exp_tile = pos + 1
+        elif pos < rem * (base + 1):
+            exp_tile = pos // (base + 1) + 1
+        else:
+            exp_tile = rem + (pos - rem * (base + 1)) // base + 1
+        run = seg[order[seg] == order[i]]
+        exp_cume = (run[-1] - seg[0] + 1) / size
+        exp_pct = 0.0 if size == 1 else (run[0] - seg[0]) / (size - 1)
+        assert r[2] == exp_tile, i
+        assert abs(r[3] - exp_cume) < 1e-12, i
+        assert abs(r[4] - exp_pct) < 1e-12, i
