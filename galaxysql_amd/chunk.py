@@ -154,6 +154,12 @@ def dec40_encode(scaled: int, scale: int) -> bytes:
     import struct
     neg = scaled < 0
     a = -scaled if neg else scaled
+    if a >= 10 ** 18:
+        raise ValueError(
+            "wide DECIMAL: >18 significant digits does not fit the "
+            "DecimalBox simple layout / scaled-int64 fast path "
+            "(reference falls back to full Decimal arithmetic, "
+            "DecimalBox.java:43-71)")
     ip, rem = divmod(a, 10 ** scale)
     fr = rem * 10 ** (9 - scale)
     if ip >= 10 ** 9:
@@ -180,6 +186,10 @@ def dec40_decode(p, scale: int) -> int:
         ip, fr = w0 * 10 ** 9 + w1, w2
     else:
         ip, fr = w0, w1
+    if integers > 18 or ip >= 10 ** (18 - scale):
+        raise ValueError(
+            "wide DECIMAL: >18 significant digits does not fit the "
+            "scaled-int64 fast path (DecimalBox.java:43-71)")
     v = ip * 10 ** scale + fr // 10 ** (9 - scale)
     return -v if isneg else v
 

@@ -129,10 +129,16 @@ def final_agg_specs(n_group_cols, aggs):
         if func in (abi.COUNT_ROW, abi.COUNT_COL):
             finals.append((abi.SUM_I64, col))
             out_types.append(0)  # I64
-        elif func in (abi.SUM_I64, abi.MIN_I64, abi.MAX_I64):
+        elif func in (abi.SUM_I64, abi.MIN_I64, abi.MAX_I64,
+                      abi.BIT_AND, abi.BIT_OR, abi.BIT_XOR):
+            # bit aggs re-aggregate as themselves over I64 partial values
             finals.append((func, col))
             out_types.append(0)
-        else:  # SUM_F64 / MIN_F64 / MAX_F64
+        elif func in (abi.SUM_F64, abi.MIN_F64, abi.MAX_F64):
             finals.append((func, col))
             out_types.append(2)  # F64
+        else:
+            raise ValueError(
+                f"agg func {func} has no defined partial->final mapping "
+                "for the two-phase exchanged plan")
     return finals, out_types

@@ -215,7 +215,9 @@ def deserialize_chunk(buf: bytes, types, pos: int = 0):
     end = pos + size
     (n_blocks,) = struct.unpack_from("<i", buf, pos)
     pos += 4
-    assert n_blocks == len(types), (n_blocks, types)
+    if n_blocks != len(types):
+        raise ValueError(
+            f"frame declares {n_blocks} blocks but {len(types)} types given")
     blocks = []
     for t in types:
         if t == I64:
@@ -231,7 +233,9 @@ def deserialize_chunk(buf: bytes, types, pos: int = 0):
         else:
             raise ValueError(f"unsupported block type {t}")
         blocks.append(b)
-    assert pos == end, (pos, end)
+    if pos != end:
+        raise ValueError(
+            f"frame body misparsed: ended at {pos}, frame declares {end}")
     ch = Chunk(blocks) if blocks else Chunk([])
     ch.n_rows = n_rows
     return ch, pos

@@ -1647,7 +1647,13 @@ static const int64_t POW10[19] = {1LL,10LL,100LL,1000LL,10000LL,100000LL,
     1000000000000000LL,10000000000000000LL,100000000000000000LL,
     1000000000000000000LL};
 
-static inline int64_t dec40_to_scaled(const uint8_t *p, int scale) {
+/* Wide-DECIMAL fence: values needing a third integer word (integers > 18)
+ * or more than 18 total significant digits do not fit the scaled-int64
+ * fast path; the reference falls back to full 9-limb Decimal arithmetic
+ * there (DecimalBox.java:43-71). We reject loudly (set *err) instead of
+ * silently truncating. */
+static inline int64_t dec40_to_scaled(const uint8_t *p, int scale,
+                                      bool *err) {
     int32_t w[3];
     std::memcpy(w, p, 12);
     uint8_t integers = p[36];
@@ -1659,6 +1665,10 @@ static inline int64_t dec40_to_scaled(const uint8_t *p, int scale) {
     } else {
         ip = w[0];
         fr = w[1];
+    }
+    if (integers > 18 || ip >= POW10[18 - scale]) {
+        *err = true;
+        return 0;
     }
     /* frac word holds the fraction digits x 10^(9-scale) */
     int64_t v = ip * POW10[scale] + fr / POW10[9 - scale];
@@ -1751,6 +1761,7 @@ struct ScanOp : gx_op {
         std::vector<OutCol> cols(out_types.size());
         for (size_t c = 0; c < out_types.size(); c++) cols[c].type = out_types[c];
         int32_t kept = 0;
+        bool wide_err = false;
         for (size_t r = 0; r < in.n_rows; r++) {
             bool pass = true;
             for (auto &p : preds) {
@@ -1796,7 +1807,7 @@ struct ScanOp : gx_op {
                 case GX_PROJ_DEC_TO_SCALED: {
                     if (a.is_null(r)) cols[c].append_null();
                     else cols[c].append_i64(dec40_to_scaled(
-                        a.bytes.data() + (size_t)r * 40, pj.c));
+                        a.bytes.data() + (size_t)r * 40, pj.c, &wide_err));
                     break; }
                 case GX_PROJ_SCALED_TO_DEC: {
                     if (a.is_null(r)) cols[c].append_null();
@@ -1820,6 +1831,13 @@ struct ScanOp : gx_op {
                     break; }
                 }
             }
+        }
+        if (wide_err) {
+            set_err("wide DECIMAL: value exceeds 18 significant digits / "
+                    "the DecimalBox simple layout; the scaled-int64 fast "
+                    "path cannot represent it (reference falls back to full "
+                    "Decimal arithmetic, DecimalBox.java:43-71)");
+            return -1;
         }
         *out = make_result(std::move(cols), kept);
         return 0;

@@ -134,3 +134,61 @@ def test_gpu_decimal_copy_through_scan():
            for i in range(out.n_rows)}
     for k in range(40):
         assert got[k] == dec40_encode(k * 7 - 50, 2)
+
+
+# ---- wide-DECIMAL fence --------------------------------
+# Values with >18 significant digits (or a 3rd integer word) exceed the
+# DecimalBox "simple" layout the scaled-int64 fast path mirrors
+# (DecimalBox.java:43-71); the reference falls back to full 9-limb Decimal
+# arithmetic there. We must reject LOUDLY, not truncate silently.
+
+def _wide_dec40(scale: int) -> bytes:
+    """Raw 2-word record whose integer part has 18-scale+1 digits: decodes
+    to >18 total significant digits at `scale`."""
+    import struct
+    ip = 10 ** (18 - scale)  # too big by exactly one digit
+    p = bytearray(40)
+    struct.pack_into("<iii", p, 0, ip // 10 ** 9, ip % 10 ** 9, 0)
+    p[36] = 18
+    p[37] = scale
+    p[38] = scale
+    return bytes(p)
+
+
+def test_dec40_encode_rejects_wide():
+    with pytest.raises(ValueError, match="wide DECIMAL"):
+        dec40_encode(10 ** 18, 2)
+    with pytest.raises(ValueError, match="wide DECIMAL"):
+        dec40_decode(np.frombuffer(_wide_dec40(2), np.uint8), 2)
+    # boundary value (exactly 18 digits) still fine
+    assert dec40_decode(np.frombuffer(dec40_encode(10 ** 18 - 1, 2),
+                                      np.uint8), 2) == 10 ** 18 - 1
+
+
+def _wide_chunk():
+    vals = np.zeros((3, 40), dtype=np.uint8)
+    vals[0] = np.frombuffer(dec40_encode(1234, 2), np.uint8)
+    vals[1] = np.frombuffer(_wide_dec40(2), np.uint8)
+    vals[2] = np.frombuffer(dec40_encode(-5, 2), np.uint8)
+    return Chunk([Block(DECIMAL, values=vals)])
+
+
+def test_oracle_rejects_wide_decimal():
+    lib = abi.load_oracle()
+    sc = ScanExec(lib, preds=[],
+                  projs=[(abi.PROJ_DEC_TO_SCALED, 0, -1, 2)],
+                  input_types=[DECIMAL], device=-1)
+    with pytest.raises(RuntimeError, match="wide DECIMAL"):
+        sc.consume_chunk(_wide_chunk())
+    sc.close()
+
+
+@pytest.mark.gpu
+def test_gpu_rejects_wide_decimal():
+    lib = abi.load_hip()
+    sc = ScanExec(lib, preds=[],
+                  projs=[(abi.PROJ_DEC_TO_SCALED, 0, -1, 2)],
+                  input_types=[DECIMAL], device=0)
+    with pytest.raises(RuntimeError, match="wide DECIMAL"):
+        sc.consume_chunk(_wide_chunk())
+    sc.close()
