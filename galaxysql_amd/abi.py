@@ -53,6 +53,22 @@ class GxEquiKey(C.Structure):
     ]
 
 
+class GxJoinCond(C.Structure):
+    """Residual (non-equi) join condition term (gx_join_cond in gxop.h):
+    AND of terms over the reference's condition-row layout (leftSide cols
+    then rightSide cols, JoinRelType.java:145-151)."""
+    _fields_ = [
+        ("col_a", C.c_int32),
+        ("cmp", C.c_int32),
+        ("col_b", C.c_int32),
+        ("v_i64", C.c_int64),
+        ("v_f64", C.c_double),
+        ("v_bytes", C.POINTER(C.c_uint8)),
+        ("v_len", C.c_int32),
+        ("const_is_null", C.c_int32),
+    ]
+
+
 class GxJoinCfg(C.Structure):
     _fields_ = [
         ("join_type", C.c_int32),
@@ -71,6 +87,8 @@ class GxJoinCfg(C.Structure):
         ("n_out_proj", C.c_int32),
         ("out_proj", C.POINTER(C.c_int32)),
         ("memory_budget_bytes", C.c_int64),
+        ("n_conds", C.c_int32),
+        ("conds", C.POINTER(GxJoinCond)),
     ]
 
 
@@ -207,6 +225,8 @@ INNER, LEFT, RIGHT, SEMI, ANTI = 0, 1, 2, 3, 4
 # Comparisons (gx_cmp)
 LT, LE, GT, GE, EQ, NE = 0, 1, 2, 3, 4, 5
 CONTAINS = 6  # SLICE LIKE '%pat%'
+# join-condition-only null-safe compares (Objects.equals semantics)
+EQ_NULLSAFE, NE_NULLSAFE = 7, 8
 # Projections (gx_proj_op)
 PROJ_COPY, PROJ_REV_F64, PROJ_REV_SCALED4, PROJ_Q9_AMOUNT4 = 0, 1, 2, 3
 PROJ_DEC_TO_SCALED, PROJ_SCALED_TO_DEC = 4, 5  # gx_proj.c = decimal scale

@@ -173,6 +173,98 @@ __device__ static inline bool rows_key_equal(const KeyViews &a, int64_t i,
     return true;
 }
 
+/* ---- residual (non-equi) join condition -------------------------------
+ * AbstractJoinExec.checkJoinCondition:227-250: evaluated per matched
+ * candidate; a failing candidate is skipped and does not set `matched`.
+ * Host code maps the condition-row column indices (leftSide cols then
+ * rightSide cols) onto probe/build sides before launch. */
+#define GX_MAX_CONDS 4
+
+struct JoinCondDev {
+    int32_t cmp;
+    int32_t a_is_build;
+    int32_t b_is_const;
+    int32_t b_is_build;
+    int32_t const_is_null;
+    DevColView a, b;
+    int64_t v_i64;
+    double v_f64;
+    const uint8_t *v_bytes;
+    int32_t v_len;
+};
+
+__device__ static inline bool cond_cmp_i64d(int64_t a, int32_t cmp, int64_t b) {
+    switch (cmp) {
+    case GX_CMP_LT: return a < b;
+    case GX_CMP_LE: return a <= b;
+    case GX_CMP_GT: return a > b;
+    case GX_CMP_GE: return a >= b;
+    case GX_CMP_NE: case GX_CMP_NE_NULLSAFE: return a != b;
+    default: return a == b; /* EQ / EQ_NULLSAFE */
+    }
+}
+
+__device__ static inline bool cond_cmp_f64d(double a, int32_t cmp, double b) {
+    switch (cmp) {
+    case GX_CMP_LT: return a < b;
+    case GX_CMP_LE: return a <= b;
+    case GX_CMP_GT: return a > b;
+    case GX_CMP_GE: return a >= b;
+    case GX_CMP_NE: case GX_CMP_NE_NULLSAFE: return a != b;
+    default: return a == b;
+    }
+}
+
+__device__ static bool join_cond_term(const JoinCondDev &c, int64_t pi,
+                                      uint32_t bpos) {
+    int64_t ia = c.a_is_build ? (int64_t)bpos : pi;
+    int64_t ib = c.b_is_const ? 0 : (c.b_is_build ? (int64_t)bpos : pi);
+    bool na = col_is_null(c.a, ia);
+    bool nb = c.b_is_const ? (bool)c.const_is_null : col_is_null(c.b, ib);
+    bool nullsafe = c.cmp == GX_CMP_EQ_NULLSAFE || c.cmp == GX_CMP_NE_NULLSAFE;
+    if (na || nb) {
+        if (!nullsafe) return false; /* SQL: NULL fails every comparison */
+        bool eq = na && nb;          /* Objects.equals: NULL == NULL only */
+        return c.cmp == GX_CMP_EQ_NULLSAFE ? eq : !eq;
+    }
+    switch (c.a.type) {
+    case GX_SLICE: {
+        int32_t ab = slice_begin(c.a, ia), ae = c.a.offsets[ia];
+        const uint8_t *bp;
+        int32_t blen;
+        if (c.b_is_const) {
+            bp = c.v_bytes;
+            blen = c.v_len;
+        } else {
+            int32_t bb = slice_begin(c.b, ib);
+            bp = c.b.bytes + bb;
+            blen = c.b.offsets[ib] - bb;
+        }
+        bool eq = (ae - ab) == blen;
+        for (int32_t k = 0; eq && k < blen; k++) eq = c.a.bytes[ab + k] == bp[k];
+        return (c.cmp == GX_CMP_NE || c.cmp == GX_CMP_NE_NULLSAFE) ? !eq : eq;
+    }
+    case GX_F64: {
+        double av = ((const double *)c.a.values)[ia];
+        double bv = c.b_is_const ? c.v_f64 : ((const double *)c.b.values)[ib];
+        return cond_cmp_f64d(av, c.cmp, bv);
+    }
+    default: {
+        int64_t av = c.a.type == GX_I32
+            ? (int64_t)((const int32_t *)c.a.values)[ia]
+            : ((const int64_t *)c.a.values)[ia];
+        int64_t bv;
+        if (c.b_is_const)
+            bv = c.v_i64;
+        else
+            bv = c.b.type == GX_I32
+                ? (int64_t)((const int32_t *)c.b.values)[ib]
+                : ((const int64_t *)c.b.values)[ib];
+        return cond_cmp_i64d(av, c.cmp, bv);
+    }
+    }
+}
+
 /* ======================= generic small kernels ========================= */
 
 __global__ void k_hash_rows(KeyViews keys, int64_t n, int32_t *hashes,
@@ -199,9 +291,27 @@ __global__ void k_any_null(DevColView c, int64_t n, uint32_t *flag) {
 
 /* ========================== join: build =============================== */
 
-/* CSR join table entry, one 16-B aligned load per candidate:
+/* Join table entry, one 16-B aligned load per candidate:
  * fast path (single I64-typed key): key inline; generic: key compared via
- * build key columns at pos. */
+ * build key columns at pos.
+ *
+ * Two layouts share this struct:
+ *
+ * INLINE-BUCKET (plain joins, the k_probe path): the table is an array of
+ * 64-B buckets = 4 JoinEntry slots; bucket metadata lives in the slots'
+ * otherwise-unused pad words (e[0].pad = entry count, e[1].pad = start of
+ * this bucket's overflow run in the separate overflow array). With
+ * n_buckets = pow2(n) (load <= 1/bucket-line) a probe resolves count,
+ * keys and positions from ONE random 64-B line — r01 PMC showed the CSR
+ * layout (starts[] + entries[]) spending 2.7-3.2x the algorithmic bytes
+ * on 64-B lines fetched per 16-B entry; this layout removes the whole
+ * starts[] indirection. Overflow (count > 4, ~0.4% of buckets at load 1)
+ * continues into entries[ovf_start ..]. e[0] is loaded unconditionally
+ * (count rides in it); e[1..3] load lazily from the same line.
+ *
+ * CSR (null_safe_keys group-join builds only): starts[b]..starts[b+1]
+ * index a contiguous entries[] run — kept because the group-join kernels
+ * (gxhip_groupjoin.inc) enumerate whole-table entry lists linearly. */
 struct __align__(16) JoinEntry {
     int64_t key;
     uint32_t pos;
@@ -232,6 +342,54 @@ __global__ void k_join_scatter(const int32_t *hashes, const uint8_t *keynull,
         e.pos = (uint32_t)i;
         e.pad = 0;
         entries[at] = e;
+    }
+}
+
+/* ---- inline-bucket build (plain joins) -------------------------------- */
+
+/* overflow run length per bucket (entries beyond the 4 inline slots) */
+__global__ void k_ovf_counts(const uint32_t *counts, int64_t n_buckets,
+                             uint32_t *ovf) {
+    for (int64_t b = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         b < n_buckets; b += (int64_t)gridDim.x * blockDim.x)
+        ovf[b] = counts[b] > 4 ? counts[b] - 4 : 0;
+}
+
+/* stamp bucket metadata into the slot pads: e[0].pad = count,
+ * e[1].pad = overflow start (both land in the line's first 32-B sector).
+ * Runs BEFORE the scatter, which writes only key+pos of each slot. */
+__global__ void k_bucket_meta(const uint32_t *counts, const uint32_t *ovf_starts,
+                              int64_t n_buckets, JoinEntry *table) {
+    for (int64_t b = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         b < n_buckets; b += (int64_t)gridDim.x * blockDim.x) {
+        table[b * 4 + 0].pad = counts[b];
+        table[b * 4 + 1].pad = ovf_starts[b];
+    }
+}
+
+__global__ void k_join_scatter_ib(const int32_t *hashes, const uint8_t *keynull,
+                                  int64_t n, uint32_t *cursors,
+                                  const uint32_t *ovf_starts, uint32_t mask,
+                                  JoinEntry *table, JoinEntry *ovf,
+                                  DevColView key0, int fast_i64) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        if (keynull[i]) continue;
+        uint32_t b = (uint32_t)gx_mix(hashes[i]) & mask;
+        uint32_t rank = atomicAdd(&cursors[b], 1u);
+        int64_t key = fast_i64 ? ((const int64_t *)key0.values)[i]
+                               : (int64_t)hashes[i];
+        if (rank < 4) {
+            JoinEntry *t = &table[(uint64_t)b * 4 + rank];
+            t->key = key;       /* pad untouched: carries bucket metadata */
+            t->pos = (uint32_t)i;
+        } else {
+            JoinEntry e;
+            e.key = key;
+            e.pos = (uint32_t)i;
+            e.pad = 0;
+            ovf[ovf_starts[b] + (rank - 4)] = e;
+        }
     }
 }
 
@@ -316,8 +474,8 @@ __global__ void k_radix_scatter(const int32_t *hashes, const uint8_t *keynull,
  * null-extended side of LEFT/RIGHT unmatched rows; SEMI/ANTI emit only
  * probe_idx). */
 struct ProbeParams {
-    const uint32_t *starts;     /* n_buckets+1 */
-    const JoinEntry *entries;
+    const JoinEntry *table;     /* inline buckets: 4 slots x n_buckets */
+    const JoinEntry *entries;   /* overflow runs (count > 4) */
     uint32_t mask;
     int64_t n_probe;
     const int32_t *hashes;      /* probe row hashes */
@@ -339,7 +497,16 @@ struct ProbeParams {
     uint32_t *counter;          /* total pairs (may exceed cap) */
     uint32_t *err;              /* 1 = single-join >1 match */
     uint32_t *build_matched;    /* bitmap (u32 words), build_outer only */
+    int32_t n_conds;            /* residual condition terms (AND) */
+    JoinCondDev conds[GX_MAX_CONDS];
 };
+
+__device__ static inline bool join_conds_pass(const ProbeParams &P,
+                                              int64_t pi, uint32_t bpos) {
+    for (int t = 0; t < P.n_conds; t++)
+        if (!join_cond_term(P.conds[t], pi, bpos)) return false;
+    return true;
+}
 
 /* Per-wave LDS pair staging: batch STAGE pairs per global atomicAdd.
  * A bare per-iteration atomic on one counter serializes the whole grid
@@ -406,21 +573,29 @@ __global__ void k_probe(ProbeParams P) {
         bool matched = false;
         uint32_t it = 0, end = 0;
         int64_t want_key = 0;
+        /* one random 64-B line resolves the whole bucket: e0 carries the
+         * entry count in its pad; e1 (lazily loaded, same line) carries
+         * the overflow start. */
+        JoinEntry e0 = {0, 0, 0};
+        uint64_t ebase = 0;
+        uint32_t ovf0 = 0;
         if (P.staged) {
             RadixRow r = active ? P.staged[base_i] : RadixRow{0x80000000u, 0, 0};
             i = r.rowid & 0x7FFFFFFFu;
             if (active && !(r.rowid & 0x80000000u)) {
                 uint32_t b = (uint32_t)gx_mix(r.hash) & P.mask;
-                it = P.starts[b];
-                end = P.starts[b + 1];
+                ebase = (uint64_t)b * 4;
+                e0 = P.table[ebase];
+                end = e0.pad;
                 want_key = r.key;
             }
         } else {
             i = active ? base_i : 0;
             if (active && !P.keynull[i]) {
                 uint32_t b = (uint32_t)gx_mix(P.hashes[i]) & P.mask;
-                it = P.starts[b];
-                end = P.starts[b + 1];
+                ebase = (uint64_t)b * 4;
+                e0 = P.table[ebase];
+                end = e0.pad;
                 if (P.fast_i64)
                     want_key = ((const int64_t *)P.probe_keys.col[0].values)[i];
             }
@@ -432,13 +607,26 @@ __global__ void k_probe(ProbeParams P) {
             bool is_match = false;
             uint32_t bpos = 0;
             if (have) {
-                JoinEntry e = P.entries[it++];
+                JoinEntry e;
+                if (it == 0) {
+                    e = e0;
+                } else if (it < 4) {
+                    e = P.table[ebase + it];
+                    if (it == 1) ovf0 = e.pad;
+                } else {
+                    e = P.entries[ovf0 + (it - 4)];
+                }
+                it++;
                 if (P.fast_i64)
                     is_match = (e.key == want_key);
                 else
                     is_match = ((int32_t)e.key == P.hashes[i]) &&
                                rows_key_equal(P.build_keys, e.pos,
                                               P.probe_keys, i);
+                /* residual condition: a failing candidate is skipped and
+                 * never counts as a match (checkJoinCondition:206-208) */
+                if (is_match && P.n_conds)
+                    is_match = join_conds_pass(P, i, e.pos);
                 bpos = e.pos;
             }
             if (is_match) {
@@ -824,13 +1012,17 @@ struct StagedChunk {
 struct JoinOp : gx_op {
     gx_join_cfg cfg;
     std::vector<gx_equi_key> keys;
+    std::vector<gx_join_cond> conds;             /* residual condition */
+    std::vector<std::vector<uint8_t>> cond_bytes; /* deep-copied SLICE consts */
+    std::vector<DevBuf> d_cond_pats;
     std::vector<int32_t> out_proj;  /* projection pushdown; empty = full */
     std::vector<int32_t> outer_types, inner_types;
     std::vector<int> build_key_cols, probe_key_cols;
 
     DevStore build;          /* build-side columns in HBM */
     DevBuf d_hashes, d_keynull;
-    DevBuf d_counts, d_starts, d_entries;
+    DevBuf d_table;          /* inline-bucket table (plain joins) */
+    DevBuf d_counts, d_starts, d_entries; /* CSR (null_safe) / IB overflow */
     DevBuf d_bitmap;         /* build_outer matched bitmap */
     DevBuf d_scan_tmp;
     DevBuf d_pidx, d_bpos, d_meta, d_ph, d_pn; /* reused across probe calls */
@@ -852,6 +1044,17 @@ struct JoinOp : gx_op {
 
     JoinOp(const gx_join_cfg *c) : gx_op(OP_JOIN, c->device, c->stream), cfg(*c) {
         keys.assign(c->keys, c->keys + c->n_keys);
+        if (c->n_conds > 0) {
+            conds.assign(c->conds, c->conds + c->n_conds);
+            cond_bytes.resize(conds.size());
+            for (size_t t = 0; t < conds.size(); t++) {
+                if (conds[t].col_b < 0 && conds[t].v_bytes && conds[t].v_len > 0) {
+                    cond_bytes[t].assign(conds[t].v_bytes,
+                                         conds[t].v_bytes + conds[t].v_len);
+                    conds[t].v_bytes = cond_bytes[t].data();
+                }
+            }
+        }
         if (c->n_out_proj > 0)
             out_proj.assign(c->out_proj, c->out_proj + c->n_out_proj);
         outer_types.assign(c->outer_types, c->outer_types + c->n_outer_cols);
@@ -866,14 +1069,50 @@ struct JoinOp : gx_op {
     }
     ~JoinOp() override {
         build.release();
+        d_table.release();
         d_hashes.release(); d_keynull.release(); d_counts.release();
         d_starts.release(); d_entries.release(); d_bitmap.release();
         d_scan_tmp.release();
         d_pidx.release(); d_bpos.release(); d_meta.release();
         d_ph.release(); d_pn.release();
         d_staged.release(); d_radix_cnt.release();
+        for (auto &b : d_cond_pats) b.release();
         if (ev0) (void)hipEventDestroy(ev0);
         if (ev1) (void)hipEventDestroy(ev1);
+    }
+
+    /* condition-row column (JoinRelType.leftSide cols then rightSide cols)
+     * -> (build-side?, source column) honoring build_outer */
+    int cond_map(int32_t col, bool &is_build) const {
+        const int n_outer = (int)outer_types.size();
+        const int n_inner = (int)inner_types.size();
+        const int n_left = cfg.join_type == GX_JOIN_RIGHT ? n_inner : n_outer;
+        bool is_outer_col;
+        int src;
+        if (cfg.join_type == GX_JOIN_RIGHT) {
+            is_outer_col = col >= n_left;
+            src = is_outer_col ? col - n_left : col;
+        } else {
+            is_outer_col = col < n_left;
+            src = is_outer_col ? col : col - n_left;
+        }
+        is_build = is_outer_col == (cfg.build_outer != 0);
+        return src;
+    }
+
+    int upload_cond_pats() {
+        if (conds.empty() || !d_cond_pats.empty()) return 0;
+        d_cond_pats.resize(conds.size());
+        for (size_t t = 0; t < conds.size(); t++) {
+            const gx_join_cond &c = conds[t];
+            if (c.col_b < 0 && c.v_bytes && c.v_len > 0) {
+                if (d_cond_pats[t].grow((size_t)c.v_len, stream)) return -1;
+                HIP_OK(hipMemcpyAsync(d_cond_pats[t].p, c.v_bytes,
+                                      (size_t)c.v_len, hipMemcpyHostToDevice,
+                                      stream));
+            }
+        }
+        return 0;
     }
 
     KeyViews key_views(const DevStore &s, const std::vector<int> &colidx) const {
@@ -913,8 +1152,16 @@ struct JoinOp : gx_op {
              * rows) — give the probe kernel a valid all-empty table. */
             n_buckets = 2;
             mask = 1;
-            if (d_starts.grow((size_t)(n_buckets + 1) * 4, stream)) return -1;
-            HIP_OK(hipMemsetAsync(d_starts.p, 0, (size_t)(n_buckets + 1) * 4, stream));
+            if (null_safe_keys) {
+                if (d_starts.grow((size_t)(n_buckets + 1) * 4, stream)) return -1;
+                HIP_OK(hipMemsetAsync(d_starts.p, 0, (size_t)(n_buckets + 1) * 4, stream));
+            } else {
+                if (d_table.grow((size_t)n_buckets * 4 * sizeof(JoinEntry),
+                                 stream)) return -1;
+                HIP_OK(hipMemsetAsync(d_table.p, 0,
+                                      (size_t)n_buckets * 4 * sizeof(JoinEntry),
+                                      stream));
+            }
         }
         if (n > 0) {
             if (d_hashes.grow((size_t)n * 4, stream) ||
@@ -923,35 +1170,87 @@ struct JoinOp : gx_op {
             hipLaunchKernelGGL(k_hash_rows, dim3(gx_grid(n)), dim3(256), 0, stream,
                                bk, n, (int32_t *)d_hashes.p, (uint8_t *)d_keynull.p,
                                (int)null_safe_keys);
-
-            n_buckets = gx_pow2(n * 2); /* avg load 0.5 */
-            mask = (uint32_t)(n_buckets - 1);
-            if (d_counts.grow((size_t)(n_buckets + 1) * 4, stream) ||
-                d_starts.grow((size_t)(n_buckets + 1) * 4, stream) ||
-                d_entries.grow((size_t)n * sizeof(JoinEntry), stream)) return -1;
-            HIP_OK(hipMemsetAsync(d_counts.p, 0, (size_t)(n_buckets + 1) * 4, stream));
-            hipLaunchKernelGGL(k_join_hist, dim3(gx_grid(n)), dim3(256), 0, stream,
-                               (const int32_t *)d_hashes.p, (const uint8_t *)d_keynull.p,
-                               n, (uint32_t *)d_counts.p, mask);
-            /* exclusive scan counts[0..n_buckets] -> starts (incl. total) */
-            size_t tmp_bytes = 0;
-            HIP_OK(hipcub::DeviceScan::ExclusiveSum(nullptr, tmp_bytes,
-                                                    (uint32_t *)d_counts.p,
-                                                    (uint32_t *)d_starts.p,
-                                                    n_buckets + 1, stream));
-            if (d_scan_tmp.grow(tmp_bytes, stream)) return -1;
-            HIP_OK(hipcub::DeviceScan::ExclusiveSum(d_scan_tmp.p, tmp_bytes,
-                                                    (uint32_t *)d_counts.p,
-                                                    (uint32_t *)d_starts.p,
-                                                    n_buckets + 1, stream));
-            /* reuse counts as cursors (= starts) */
-            HIP_OK(hipMemcpyAsync(d_counts.p, d_starts.p, (size_t)n_buckets * 4,
-                                  hipMemcpyDeviceToDevice, stream));
-            hipLaunchKernelGGL(k_join_scatter, dim3(gx_grid(n)), dim3(256), 0, stream,
-                               (const int32_t *)d_hashes.p, (const uint8_t *)d_keynull.p,
-                               n, (uint32_t *)d_counts.p, mask,
-                               (JoinEntry *)d_entries.p,
-                               build.view(build_key_cols[0]), (int)fast_i64);
+            if (null_safe_keys) {
+                /* CSR layout for the group-join kernels (linear entry
+                 * enumeration); plain joins use the inline-bucket table. */
+                n_buckets = gx_pow2(n * 2); /* avg load 0.5 */
+                mask = (uint32_t)(n_buckets - 1);
+                if (d_counts.grow((size_t)(n_buckets + 1) * 4, stream) ||
+                    d_starts.grow((size_t)(n_buckets + 1) * 4, stream) ||
+                    d_entries.grow((size_t)n * sizeof(JoinEntry), stream)) return -1;
+                HIP_OK(hipMemsetAsync(d_counts.p, 0, (size_t)(n_buckets + 1) * 4, stream));
+                hipLaunchKernelGGL(k_join_hist, dim3(gx_grid(n)), dim3(256), 0, stream,
+                                   (const int32_t *)d_hashes.p, (const uint8_t *)d_keynull.p,
+                                   n, (uint32_t *)d_counts.p, mask);
+                /* exclusive scan counts[0..n_buckets] -> starts (incl. total) */
+                size_t tmp_bytes = 0;
+                HIP_OK(hipcub::DeviceScan::ExclusiveSum(nullptr, tmp_bytes,
+                                                        (uint32_t *)d_counts.p,
+                                                        (uint32_t *)d_starts.p,
+                                                        n_buckets + 1, stream));
+                if (d_scan_tmp.grow(tmp_bytes, stream)) return -1;
+                HIP_OK(hipcub::DeviceScan::ExclusiveSum(d_scan_tmp.p, tmp_bytes,
+                                                        (uint32_t *)d_counts.p,
+                                                        (uint32_t *)d_starts.p,
+                                                        n_buckets + 1, stream));
+                /* reuse counts as cursors (= starts) */
+                HIP_OK(hipMemcpyAsync(d_counts.p, d_starts.p, (size_t)n_buckets * 4,
+                                      hipMemcpyDeviceToDevice, stream));
+                hipLaunchKernelGGL(k_join_scatter, dim3(gx_grid(n)), dim3(256), 0, stream,
+                                   (const int32_t *)d_hashes.p, (const uint8_t *)d_keynull.p,
+                                   n, (uint32_t *)d_counts.p, mask,
+                                   (JoinEntry *)d_entries.p,
+                                   build.view(build_key_cols[0]), (int)fast_i64);
+            } else {
+                /* inline-bucket build: hist -> overflow-run scan -> bucket
+                 * metadata stamp -> scatter (key+pos only; pads carry the
+                 * metadata). */
+                n_buckets = gx_pow2(n); /* 4 slots/bucket, load <= 1 */
+                mask = (uint32_t)(n_buckets - 1);
+                if (d_counts.grow((size_t)(n_buckets + 1) * 4, stream) ||
+                    d_starts.grow((size_t)(n_buckets + 1) * 4, stream) ||
+                    d_table.grow((size_t)n_buckets * 4 * sizeof(JoinEntry),
+                                 stream)) return -1;
+                HIP_OK(hipMemsetAsync(d_counts.p, 0, (size_t)(n_buckets + 1) * 4, stream));
+                hipLaunchKernelGGL(k_join_hist, dim3(gx_grid(n)), dim3(256), 0, stream,
+                                   (const int32_t *)d_hashes.p, (const uint8_t *)d_keynull.p,
+                                   n, (uint32_t *)d_counts.p, mask);
+                hipLaunchKernelGGL(k_ovf_counts, dim3(gx_grid(n_buckets)), dim3(256),
+                                   0, stream, (const uint32_t *)d_counts.p,
+                                   n_buckets, (uint32_t *)d_starts.p);
+                size_t tmp_bytes = 0;
+                HIP_OK(hipcub::DeviceScan::ExclusiveSum(nullptr, tmp_bytes,
+                                                        (uint32_t *)d_starts.p,
+                                                        (uint32_t *)d_starts.p,
+                                                        n_buckets + 1, stream));
+                if (d_scan_tmp.grow(tmp_bytes, stream)) return -1;
+                HIP_OK(hipcub::DeviceScan::ExclusiveSum(d_scan_tmp.p, tmp_bytes,
+                                                        (uint32_t *)d_starts.p,
+                                                        (uint32_t *)d_starts.p,
+                                                        n_buckets + 1, stream));
+                uint32_t total_ovf = 0;
+                HIP_OK(hipMemcpyAsync(&total_ovf,
+                                      (uint32_t *)d_starts.p + n_buckets, 4,
+                                      hipMemcpyDeviceToHost, stream));
+                HIP_OK(hipStreamSynchronize(stream));
+                if (d_entries.grow((size_t)std::max<uint32_t>(total_ovf, 1) *
+                                       sizeof(JoinEntry), stream)) return -1;
+                hipLaunchKernelGGL(k_bucket_meta, dim3(gx_grid(n_buckets)),
+                                   dim3(256), 0, stream,
+                                   (const uint32_t *)d_counts.p,
+                                   (const uint32_t *)d_starts.p, n_buckets,
+                                   (JoinEntry *)d_table.p);
+                HIP_OK(hipMemsetAsync(d_counts.p, 0, (size_t)n_buckets * 4, stream));
+                hipLaunchKernelGGL(k_join_scatter_ib, dim3(gx_grid(n)), dim3(256),
+                                   0, stream,
+                                   (const int32_t *)d_hashes.p,
+                                   (const uint8_t *)d_keynull.p,
+                                   n, (uint32_t *)d_counts.p,
+                                   (const uint32_t *)d_starts.p, mask,
+                                   (JoinEntry *)d_table.p,
+                                   (JoinEntry *)d_entries.p,
+                                   build.view(build_key_cols[0]), (int)fast_i64);
+            }
         }
 
         /* ANTI NOT-IN: build contains NULL -> pass nothing
@@ -1127,8 +1426,7 @@ struct JoinOp : gx_op {
             /* radix-stage the probe rows by bucket range when the table far
              * exceeds the Infinity Cache (fast path only) — slices of
              * starts+entries become L3-resident while their rows probe. */
-            const int64_t table_bytes =
-                (n_buckets + 1) * 4 + build.n_rows * (int64_t)sizeof(JoinEntry);
+            const int64_t table_bytes = n_buckets * 4 * (int64_t)sizeof(JoinEntry);
             int64_t n_part_radix = 1;
             int radix_shift = 0;
             /* MEASURED OFF by default: bucket-range staging made each
@@ -1184,7 +1482,7 @@ struct JoinOp : gx_op {
                     d_bpos.grow((size_t)cap * 4, stream)) { attempt = 99; break; }
                 HIP_OK(hipMemsetAsync(d_meta.p, 0, 8, stream));
                 ProbeParams P;
-                P.starts = (const uint32_t *)d_starts.p;
+                P.table = (const JoinEntry *)d_table.p;
                 P.entries = (const JoinEntry *)d_entries.p;
                 P.mask = mask;
                 P.n_probe = n;
@@ -1212,6 +1510,30 @@ struct JoinOp : gx_op {
                 P.counter = (uint32_t *)d_meta.p;
                 P.err = (uint32_t *)d_meta.p + 1;
                 P.build_matched = cfg.build_outer ? (uint32_t *)d_bitmap.p : nullptr;
+                P.n_conds = (int32_t)conds.size();
+                if (!conds.empty() && upload_cond_pats()) { attempt = 99; break; }
+                for (size_t t = 0; t < conds.size(); t++) {
+                    const gx_join_cond &src = conds[t];
+                    JoinCondDev &d = P.conds[t];
+                    std::memset(&d, 0, sizeof(d));
+                    d.cmp = src.cmp;
+                    d.const_is_null = src.const_is_null;
+                    bool a_build = false, b_build = false;
+                    int a_src = cond_map(src.col_a, a_build);
+                    d.a_is_build = a_build;
+                    d.a = a_build ? build.view(a_src) : probe_st.views[a_src];
+                    d.b_is_const = src.col_b < 0;
+                    if (d.b_is_const) {
+                        d.v_i64 = src.v_i64;
+                        d.v_f64 = src.v_f64;
+                        d.v_bytes = (const uint8_t *)d_cond_pats[t].p;
+                        d.v_len = src.v_len;
+                    } else {
+                        int b_src = cond_map(src.col_b, b_build);
+                        d.b_is_build = b_build;
+                        d.b = b_build ? build.view(b_src) : probe_st.views[b_src];
+                    }
+                }
                 if (!ev0) {
                     HIP_OK(hipEventCreate(&ev0));
                     HIP_OK(hipEventCreate(&ev1));
@@ -1335,6 +1657,11 @@ extern "C" {
 gx_op *gxop_join_create(const gx_join_cfg *cfg) {
     if (!cfg || cfg->n_keys <= 0 || cfg->n_keys > GX_MAX_KEYS) {
         gx_set_err("bad join cfg");
+        return nullptr;
+    }
+    if (cfg->n_conds < 0 || cfg->n_conds > GX_MAX_CONDS ||
+        (cfg->n_conds > 0 && !cfg->conds)) {
+        gx_set_err("bad join condition list (max 4 AND terms)");
         return nullptr;
     }
     if (cfg->device < 0) { gx_set_err("gxhip requires a GPU device"); return nullptr; }

@@ -27,6 +27,43 @@ class EquiJoinKey:
         self.unified_type = unified_type
 
 
+class JoinCond:
+    """Residual (non-equi) condition term (gx_join_cond): the AND of terms
+    is evaluated per matched candidate (AbstractJoinExec.checkJoinCondition
+    :227-250). col_a/col_b index the condition row: leftSide cols then
+    rightSide cols (JoinRelType.java:145-151). col_b = -1 compares against
+    `value` (int/float/str/bytes, or None for SQL NULL)."""
+
+    def __init__(self, col_a, cmp, col_b=-1, value=None):
+        self.col_a = col_a
+        self.cmp = cmp
+        self.col_b = col_b
+        self.value = value
+
+
+def _conds_to_ctypes(conds, keep):
+    arr = (abi.GxJoinCond * len(conds))()
+    for i, c in enumerate(conds):
+        t = abi.GxJoinCond(col_a=c.col_a, cmp=c.cmp, col_b=c.col_b)
+        if c.col_b < 0:
+            v = c.value
+            if v is None:
+                t.const_is_null = 1
+            elif isinstance(v, float):
+                t.v_f64 = v
+            elif isinstance(v, (bytes, str)):
+                raw = v.encode() if isinstance(v, str) else v
+                buf = (C.c_uint8 * max(1, len(raw)))(*raw)
+                keep.append(buf)
+                t.v_bytes = buf
+                t.v_len = len(raw)
+            else:
+                t.v_i64 = int(v)
+        arr[i] = t
+    keep.append(arr)
+    return arr
+
+
 class ParallelHashJoinExec:
     """One operator instance (the C side holds the shared build state).
 
@@ -37,7 +74,7 @@ class ParallelHashJoinExec:
     def __init__(self, lib, join_type, join_keys, outer_types, inner_types,
                  max_one_row=False, build_outer=False, anti_null_col=-1,
                  device=-1, stream=0, expected_build_rows=0, out_proj=None,
-                 memory_budget_bytes=0):
+                 memory_budget_bytes=0, conds=None):
         self._lib = lib
         self._keep = []
         keys = (GxEquiKey * len(join_keys))()
@@ -55,6 +92,9 @@ class ParallelHashJoinExec:
             expected_build_rows=expected_build_rows,
             n_out_proj=len(out_proj or []), out_proj=op_arr,
             memory_budget_bytes=memory_budget_bytes)
+        if conds:
+            cfg.n_conds = len(conds)
+            cfg.conds = _conds_to_ctypes(conds, self._keep)
         self._keep += [keys, ot, it, op_arr, cfg]
         self._op = lib.lib.gxop_join_create(C.byref(cfg))
         if not self._op:

@@ -113,6 +113,32 @@ typedef struct gx_equi_key {
                                  on this path — must be 0 */
 } gx_equi_key;
 
+/* Residual (non-equi) join condition term, evaluated per matched candidate
+ * AFTER the equi-key match (AbstractBufferedJoinExec.java:206-208 ->
+ * AbstractJoinExec.checkJoinCondition:227-250): a candidate failing the
+ * condition is skipped and does NOT count as a match (so it feeds LEFT
+ * null-row emission, SEMI/ANTI matched flags and the singleJoin >1-row
+ * check exactly like the reference). The full condition is the AND of the
+ * terms. Column indices address the reference's condition row layout:
+ * JoinRelType.leftSide columns then rightSide columns (leftSide = outer
+ * unless RIGHT — polardbx-calcite JoinRelType.java:145-151).
+ *
+ * cmp uses gx_cmp below. Plain comparisons follow SQL semantics (any NULL
+ * operand fails the term); GX_CMP_EQ_NULLSAFE / GX_CMP_NE_NULLSAFE follow
+ * Objects.equals semantics (NULL equals NULL, NULL never equals a value) —
+ * the semantics the reference's own condition fixtures use
+ * (HashJoinTest.java:300-306, :774-780). */
+typedef struct gx_join_cond {
+    int32_t col_a;       /* condition-row column */
+    int32_t cmp;         /* gx_cmp */
+    int32_t col_b;       /* second condition-row column, or -1 = constant */
+    int64_t v_i64;       /* constant for I64/I32 columns */
+    double v_f64;        /* constant for F64 columns */
+    const uint8_t *v_bytes; /* constant for SLICE columns (EQ/NE only) */
+    int32_t v_len;
+    int32_t const_is_null;  /* constant is SQL NULL */
+} gx_join_cond;
+
 typedef struct gx_join_cfg {
     int32_t join_type;        /* gx_join_type */
     int32_t single_join;      /* maxOneRow: output = outer + first inner col;
@@ -148,6 +174,9 @@ typedef struct gx_join_cfg {
      * MemoryAllocatorCtx the same way the reference picks
      * HybridHashJoinExec over ParallelHashJoinExec at plan time. */
     int64_t memory_budget_bytes;
+    /* residual condition: AND of terms; n_conds = 0 -> pure equi-join */
+    int32_t n_conds;
+    const gx_join_cond *conds;
 } gx_join_cfg;
 
 typedef struct gx_op gx_op;  /* opaque operator instance */
@@ -415,7 +444,10 @@ int gxop_fwindow_close(gx_op *op);
 
 typedef enum gx_cmp {
     GX_CMP_LT = 0, GX_CMP_LE, GX_CMP_GT, GX_CMP_GE, GX_CMP_EQ, GX_CMP_NE,
-    GX_CMP_CONTAINS = 6   /* SLICE columns: LIKE '%pattern%' byte scan */
+    GX_CMP_CONTAINS = 6,  /* SLICE columns: LIKE '%pattern%' byte scan */
+    /* join-condition-only null-safe compares (Objects.equals semantics:
+     * NULL == NULL, NULL != value — see gx_join_cond) */
+    GX_CMP_EQ_NULLSAFE = 7, GX_CMP_NE_NULLSAFE = 8
 } gx_cmp;
 
 typedef struct gx_pred {

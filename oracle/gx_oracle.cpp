@@ -385,9 +385,33 @@ namespace {
  * matchNext (AbstractHashJoinExec.java:80-106) + Synchronizer.buildHashTable
  * (ParallelHashJoinExec.java:406-426) with numPartitions=1 (single-threaded:
  * insertion order = consume order, as in the reference's SingleExecTest). */
+/* residual-condition comparisons (see gx_join_cond in gxop.h) */
+static inline bool cond_cmp_i64o(int64_t a, int32_t cmp, int64_t b) {
+    switch (cmp) {
+    case GX_CMP_LT: return a < b;
+    case GX_CMP_LE: return a <= b;
+    case GX_CMP_GT: return a > b;
+    case GX_CMP_GE: return a >= b;
+    case GX_CMP_NE: case GX_CMP_NE_NULLSAFE: return a != b;
+    default: return a == b;
+    }
+}
+static inline bool cond_cmp_f64o(double a, int32_t cmp, double b) {
+    switch (cmp) {
+    case GX_CMP_LT: return a < b;
+    case GX_CMP_LE: return a <= b;
+    case GX_CMP_GT: return a > b;
+    case GX_CMP_GE: return a >= b;
+    case GX_CMP_NE: case GX_CMP_NE_NULLSAFE: return a != b;
+    default: return a == b;
+    }
+}
+
 struct JoinOp : gx_op {
     gx_join_cfg cfg;
     std::vector<gx_equi_key> keys;
+    std::vector<gx_join_cond> conds;              /* residual condition */
+    std::vector<std::vector<uint8_t>> cond_bytes; /* deep-copied SLICE consts */
     std::vector<int32_t> out_proj;  /* projection pushdown; empty = full */
     std::vector<int32_t> outer_types, inner_types;
 
@@ -412,6 +436,17 @@ struct JoinOp : gx_op {
 
     JoinOp(const gx_join_cfg *c) : gx_op(OP_JOIN), cfg(*c) {
         keys.assign(c->keys, c->keys + c->n_keys);
+        if (c->n_conds > 0) {
+            conds.assign(c->conds, c->conds + c->n_conds);
+            cond_bytes.resize(conds.size());
+            for (size_t t = 0; t < conds.size(); t++) {
+                if (conds[t].col_b < 0 && conds[t].v_bytes && conds[t].v_len > 0) {
+                    cond_bytes[t].assign(conds[t].v_bytes,
+                                         conds[t].v_bytes + conds[t].v_len);
+                    conds[t].v_bytes = cond_bytes[t].data();
+                }
+            }
+        }
         if (c->n_out_proj > 0)
             out_proj.assign(c->out_proj, c->out_proj + c->n_out_proj);
         outer_types.assign(c->outer_types, c->outer_types + c->n_outer_cols);
@@ -463,6 +498,82 @@ struct JoinOp : gx_op {
         }
         if (cfg.build_outer) matched_build.assign(build.n_rows, 0);
         return 0;
+    }
+
+    /* condition-row column (JoinRelType.leftSide cols then rightSide cols,
+     * polardbx-calcite JoinRelType.java:145-151) -> (build-side?, src col) */
+    int cond_map(int32_t col, bool &is_build) const {
+        const int n_outer = (int)outer_types.size();
+        const int n_inner = (int)inner_types.size();
+        const int n_left = cfg.join_type == GX_JOIN_RIGHT ? n_inner : n_outer;
+        bool is_outer_col;
+        int src;
+        if (cfg.join_type == GX_JOIN_RIGHT) {
+            is_outer_col = col >= n_left;
+            src = is_outer_col ? col - n_left : col;
+        } else {
+            is_outer_col = col < n_left;
+            src = is_outer_col ? col : col - n_left;
+        }
+        is_build = is_outer_col == (cfg.build_outer != 0);
+        return src;
+    }
+
+    /* AbstractJoinExec.checkJoinCondition:227-250: evaluated per matched
+     * candidate; failing candidates are skipped (don't set `matched`). */
+    bool cond_pass(const Store &probe_store, size_t prow, size_t bpos) const {
+        for (const auto &c : conds) {
+            bool a_build = false, b_build = false;
+            int a_src = cond_map(c.col_a, a_build);
+            const Column &A = a_build ? build.cols[a_src]
+                                      : probe_store.cols[a_src];
+            size_t ia = a_build ? bpos : prow;
+            const Column *B = nullptr;
+            size_t ib = 0;
+            bool na = A.is_null(ia), nb;
+            if (c.col_b < 0) {
+                nb = c.const_is_null != 0;
+            } else {
+                int b_src = cond_map(c.col_b, b_build);
+                B = &(b_build ? build.cols[b_src] : probe_store.cols[b_src]);
+                ib = b_build ? bpos : prow;
+                nb = B->is_null(ib);
+            }
+            bool nullsafe = c.cmp == GX_CMP_EQ_NULLSAFE ||
+                            c.cmp == GX_CMP_NE_NULLSAFE;
+            bool ok;
+            if (na || nb) {
+                if (!nullsafe) return false; /* SQL NULL fails comparisons */
+                bool eq = na && nb;          /* Objects.equals semantics */
+                ok = c.cmp == GX_CMP_EQ_NULLSAFE ? eq : !eq;
+            } else if (A.type == GX_SLICE) {
+                int32_t ab = A.begin_off(ia), ae = A.off[ia];
+                const uint8_t *bp;
+                int32_t blen;
+                if (c.col_b < 0) {
+                    bp = c.v_bytes;
+                    blen = c.v_len;
+                } else {
+                    int32_t bb = B->begin_off(ib);
+                    bp = B->bytes.data() + bb;
+                    blen = B->off[ib] - bb;
+                }
+                bool eq = (ae - ab) == blen &&
+                          std::memcmp(A.bytes.data() + ab, bp, (size_t)blen) == 0;
+                ok = (c.cmp == GX_CMP_NE || c.cmp == GX_CMP_NE_NULLSAFE) ? !eq : eq;
+            } else if (A.type == GX_F64) {
+                double bv = c.col_b < 0 ? c.v_f64 : B->f64v[ib];
+                ok = cond_cmp_f64o(A.f64v[ia], c.cmp, bv);
+            } else {
+                int64_t av = A.type == GX_I32 ? (int64_t)A.i32v[ia] : A.i64v[ia];
+                int64_t bv;
+                if (c.col_b < 0) bv = c.v_i64;
+                else bv = B->type == GX_I32 ? (int64_t)B->i32v[ib] : B->i64v[ib];
+                ok = cond_cmp_i64o(av, c.cmp, bv);
+            }
+            if (!ok) return false;
+        }
+        return true;
     }
 
     int32_t match_init(int32_t h) const {
@@ -592,6 +703,8 @@ struct JoinOp : gx_op {
             for (; m != -1; m = links[m]) {
                 if (!build.keys_equal((size_t)m, probe_store, r,
                                       build_key_cols, probe_key_cols))
+                    continue;
+                if (!conds.empty() && !cond_pass(probe_store, r, (size_t)m))
                     continue;
                 /* NOTE: probe rows with a NULL key can never reach here —
                  * build NULL keys were never inserted and Block.equals(null,

@@ -5,13 +5,16 @@ import os
 
 from galaxysql_amd import abi
 from galaxysql_amd.chunk import Block, Chunk, I64, I32, F64, SLICE, multiset
-from galaxysql_amd.operators import EquiJoinKey, run_join, run_agg
+from galaxysql_amd.operators import EquiJoinKey, JoinCond, run_join, run_agg
 
 GOLDEN = os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden")
 
 TYPE_BY_NAME = {"i64": I64, "i32": I32, "f64": F64, "slice": SLICE}
 JOIN_BY_NAME = {"INNER": abi.INNER, "LEFT": abi.LEFT, "RIGHT": abi.RIGHT,
                 "SEMI": abi.SEMI, "ANTI": abi.ANTI}
+CMP_BY_NAME = {"LT": abi.LT, "LE": abi.LE, "GT": abi.GT, "GE": abi.GE,
+               "EQ": abi.EQ, "NE": abi.NE,
+               "EQ_NULLSAFE": abi.EQ_NULLSAFE, "NE_NULLSAFE": abi.NE_NULLSAFE}
 AGG_BY_NAME = {"COUNT_ROW": abi.COUNT_ROW, "COUNT_COL": abi.COUNT_COL,
                "SUM_I64": abi.SUM_I64, "SUM_F64": abi.SUM_F64,
                "MIN_I64": abi.MIN_I64, "MAX_I64": abi.MAX_I64,
@@ -48,11 +51,13 @@ def run_join_case(lib, case, device=-1, stream=0):
     build_chunks, inner_types = chunks_of(case["inner"])
     probe_chunks, outer_types = chunks_of(case["outer"])
     keys = [EquiJoinKey(k[0], k[1], TYPE_BY_NAME[k[2]]) for k in case["keys"]]
+    conds = [JoinCond(c[0], CMP_BY_NAME[c[1]], c[2], c[3])
+             for c in case.get("conds", [])] or None
     out = run_join(lib, JOIN_BY_NAME[case["join_type"]], keys,
                    build_chunks, probe_chunks, outer_types, inner_types,
                    max_one_row=bool(case.get("single")),
                    anti_null_col=case.get("anti_null_col", -1),
-                   device=device, stream=stream)
+                   device=device, stream=stream, conds=conds)
     rows = []
     for c in out:
         rows.extend(c.rows())
