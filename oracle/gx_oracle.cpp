@@ -29,6 +29,7 @@
 #include <string>
 #include <vector>
 #include <memory>
+#include <mutex>
 #include <algorithm>
 
 namespace {
@@ -372,6 +373,12 @@ enum OpKind { OP_JOIN = 1, OP_AGG = 2, OP_PART = 3, OP_SCAN = 4,
 
 struct gx_op {
     int kind;
+    /* consume/build serialization matching the C-ABI thread contract
+     * (the reference's synchronized(shared) in
+     * ParallelHashJoinExec.consumeChunk:158). probe stays lock-free here:
+     * it only reads the built table, and the all-cores CPU baseline
+     * (bench.py) probes one shared op from every host thread. */
+    std::mutex mu;
     virtual ~gx_op() = default;
 protected:
     explicit gx_op(int k) : kind(k) {}
@@ -465,6 +472,7 @@ struct JoinOp : gx_op {
     int consume(const gx_chunk *ch) { return build.append(ch); }
 
     int do_build() {
+        if (built) return 0; /* first-come barrier */
         const int64_t size = (int64_t)build.n_rows;
         int64_t n = hc_array_size(size, join_load_factor(size));
         mask = (int32_t)(n - 1);
@@ -1073,6 +1081,7 @@ struct GroupJoinOp : gx_op {
     int consume(const gx_chunk *ch) { return build.append(ch); }
 
     int do_build() {
+        if (built) return 0; /* first-come barrier */
         const int64_t size = (int64_t)build.n_rows;
         if (size == 0 && cfg.join_type == GX_JOIN_INNER) pass_nothing = true;
         int64_t n = hc_array_size(std::max<int64_t>(size, 1),
@@ -1969,10 +1978,12 @@ gx_op *gxop_join_create(const gx_join_cfg *cfg) {
 }
 int gxop_join_consume(gx_op *op, const gx_chunk *c) {
     if (!op || op->kind != OP_JOIN) { set_err("not a join op"); return -1; }
+    std::lock_guard<std::mutex> lk(op->mu);
     return static_cast<JoinOp *>(op)->consume(c);
 }
 int gxop_join_build(gx_op *op) {
     if (!op || op->kind != OP_JOIN) { set_err("not a join op"); return -1; }
+    std::lock_guard<std::mutex> lk(op->mu);
     return static_cast<JoinOp *>(op)->do_build();
 }
 int gxop_join_probe(gx_op *op, const gx_chunk *c, gx_result **out) {

@@ -24,6 +24,7 @@
 #include <chrono>
 #include <string>
 #include <vector>
+#include <thread>
 #include <dlfcn.h>
 
 #include "../include/gxop.h"
@@ -227,6 +228,86 @@ static void t_join_inner(int device) {
     CHECK(api.join_tail(op, &tail) == 0 && tail == nullptr, "inner tail");
     api.join_close(op);
     std::printf("  join INNER: %lld matches ok\n", (long long)got);
+}
+
+/* Concurrent-consumer contract: the reference's N
+ * driver threads share one Synchronizer — consumeChunk races through
+ * synchronized(shared), buildConsume is first-come, probes run from all
+ * instances against the shared table (ParallelHashJoinExec.java:157-166,
+ * :107-128). Here: 4 threads consume disjoint build chunks into ONE gx_op
+ * with NO external lock (the library serializes internally), all 4 call
+ * join_build (one builds, three no-op), then all 4 probe disjoint slices
+ * concurrently and the match counts must add up. */
+static void t_concurrent(int device) {
+    const int T = 4, B = 40000, PSLICE = 50000;
+    std::vector<int64_t> bk(B), bv(B);
+    for (int i = 0; i < B; i++) { bk[i] = i; bv[i] = (int64_t)i * 10; }
+    std::vector<std::vector<int64_t>> pk(T), pv(T);
+    std::vector<int64_t> expect(T, 0);
+    for (int t = 0; t < T; t++) {
+        pk[t].resize(PSLICE);
+        pv[t].resize(PSLICE);
+        for (int i = 0; i < PSLICE; i++) {
+            pk[t][i] = (int64_t)(rnd() % (2 * B));
+            pv[t][i] = i;
+            if (pk[t][i] < B) expect[t]++;
+        }
+    }
+    gx_equi_key key{0, 0, GX_I64, 0};
+    int32_t t2[2] = {GX_I64, GX_I64};
+    gx_join_cfg cfg{};
+    cfg.join_type = GX_JOIN_INNER;
+    cfg.n_keys = 1; cfg.keys = &key;
+    cfg.n_outer_cols = 2; cfg.outer_types = t2;
+    cfg.n_inner_cols = 2; cfg.inner_types = t2;
+    cfg.anti_null_col = -1;
+    cfg.device = device;
+    gx_op *op = api.join_create(&cfg);
+    CHECK(op, "conc join_create");
+    if (!op) return;
+
+    const int SLICE = B / T;
+    std::vector<std::thread> th;
+    std::vector<int> crc(T, -1);
+    for (int t = 0; t < T; t++)
+        th.emplace_back([&, t] {
+            gx_block bb[2] = {mk_i64(bk.data() + t * SLICE),
+                              mk_i64(bv.data() + t * SLICE)};
+            gx_chunk bc{SLICE, 2, bb};
+            crc[t] = api.join_consume(op, &bc);
+        });
+    for (auto &x : th) x.join();
+    th.clear();
+    for (int t = 0; t < T; t++) CHECK(crc[t] == 0, "conc consume %d", t);
+
+    std::vector<int> brc(T, -1);
+    for (int t = 0; t < T; t++)
+        th.emplace_back([&, t] { brc[t] = api.join_build(op); });
+    for (auto &x : th) x.join();
+    th.clear();
+    for (int t = 0; t < T; t++) CHECK(brc[t] == 0, "conc build %d", t);
+
+    std::vector<int64_t> got(T, -1);
+    std::vector<int> prc(T, -1);
+    for (int t = 0; t < T; t++)
+        th.emplace_back([&, t] {
+            gx_block pb[2] = {mk_i64(pk[t].data()), mk_i64(pv[t].data())};
+            gx_chunk pc{PSLICE, 2, pb};
+            gx_result *res = nullptr;
+            prc[t] = api.join_probe(op, &pc, &res);
+            if (prc[t] == 0 && res) {
+                got[t] = res->chunk.n_rows;
+                api.result_release(res);
+            }
+        });
+    for (auto &x : th) x.join();
+    for (int t = 0; t < T; t++) {
+        CHECK(prc[t] == 0, "conc probe rc %d", t);
+        CHECK(got[t] == expect[t], "conc probe %d: %lld != %lld", t,
+              (long long)got[t], (long long)expect[t]);
+    }
+    api.join_close(op);
+    std::printf("  join concurrent (4 threads, shared op): ok\n");
 }
 
 /* LEFT join: probe keys half-missing -> unmatched probe rows carry NULL
@@ -817,6 +898,7 @@ int main(int argc, char **argv) {
 
     if (cmd == "selftest") {
         t_join_inner(device);
+        t_concurrent(device);
         t_join_left(device);
         t_agg(device);
         t_part(device);
