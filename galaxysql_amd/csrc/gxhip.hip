@@ -584,11 +584,15 @@ __global__ void k_probe(ProbeParams P) {
         /* phase A: independent (hash -> bucket -> e0) chains per slot.
          * e0 carries the bucket's entry count in its pad; e1 (lazily
          * loaded in phase B, same 64-B line) carries the overflow start. */
+        /* these slot arrays MUST become registers: without full unrolling
+         * the promote-alloca pass placed them in LDS (measured: 80 KB
+         * group segment -> 2 blocks/CU -> probe 61% SLOWER) */
         uint32_t si[GX_PROBE_ILP];      /* probe row id (n_rows is i32) */
         uint32_t sb[GX_PROBE_ILP];      /* bucket index */
         JoinEntry se0[GX_PROBE_ILP];
         uint32_t send[GX_PROBE_ILP];    /* candidate count */
         int64_t swant[GX_PROBE_ILP];    /* fast-path probe key */
+#pragma unroll
         for (int k = 0; k < GX_PROBE_ILP; k++) {
             int64_t bi = base + (int64_t)k * stride;
             bool active = bi < P.n_probe;
@@ -621,6 +625,7 @@ __global__ void k_probe(ProbeParams P) {
         }
 
         /* phase B: resolve slots one at a time, wave-synchronously */
+#pragma unroll
         for (int k = 0; k < GX_PROBE_ILP; k++) {
             bool active = base + (int64_t)k * stride < P.n_probe;
             if (!__ballot(active)) break;
