@@ -394,46 +394,54 @@ __global__ void k_join_scatter_ib(const int32_t *hashes, const uint8_t *keynull,
     }
 }
 
-/* ---- radix-staged probe ------------------------------------------------
- * For LARGE builds the table (starts+entries) far exceeds the 256 MiB
- * Infinity Cache, so every probe costs 1-2 random HBM lines. Staging the
- * probe rows by BUCKET RANGE first makes each range's table slice
- * L3-resident while its rows are probed: partition p owns buckets
- * [p<<shift, (p+1)<<shift). The staged 16-B row carries everything the
- * fast-path probe needs (rowid, hash, inline key), so the probe reads are
- * payload-sequential + table-slice-local. keynull rows carry the flag in
- * the rowid's top bit (they still emit LEFT/ANTI rows, match nothing). */
+/* ---- bucket-clustered staged probe -------------------------------------
+ * For LARGE builds the inline-bucket table far exceeds the caches, so
+ * every probe costs one random HBM line and the kernel runs at ~half the
+ * random-line ceiling (r2: 11.0 ms / 324M probes). Staging the probe rows
+ * by BUCKET RANGE first makes each range's table slice L2-RESIDENT while
+ * its rows probe: partition p owns buckets [p<<shift, (p+1)<<shift) — a
+ * ~2 MB slice of table lines — and the staged probe kernel processes one
+ * partition per workgroup, so a slice's ~2.5 probes/bucket hit the same
+ * XCD's L2 after the first fetch. The staged 16-B row carries everything
+ * the fast-path probe needs (rowid, hash, inline key). NULL-key rows
+ * (bit31 of rowid) spread round-robin across partitions — they match
+ * nothing and only emit LEFT/ANTI rows. Fast-path only: the hash is
+ * computed inline from the key column (no k_hash_rows pass). */
 struct __align__(16) RadixRow {
     uint32_t rowid;   /* bit31 = keynull */
     int32_t hash;
     int64_t key;
 };
 
-__global__ void k_radix_count(const int32_t *hashes, const uint8_t *keynull,
-                              int64_t n, uint32_t mask, int shift,
-                              int n_parts, uint32_t *counts) {
-    __shared__ uint32_t s_cnt[64];
+#define GX_RADIX_MAXP 4096
+#define GX_RADIX_TILE 16384
+
+__global__ void k_radix_count(DevColView key0, int64_t n, uint32_t mask,
+                              int shift, int n_parts, uint32_t *counts) {
+    __shared__ uint32_t s_cnt[GX_RADIX_MAXP];
     for (int p = threadIdx.x; p < n_parts; p += blockDim.x) s_cnt[p] = 0;
     __syncthreads();
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += (int64_t)gridDim.x * blockDim.x) {
-        uint32_t b = keynull[i] ? 0u : ((uint32_t)gx_mix(hashes[i]) & mask);
-        atomicAdd(&s_cnt[b >> shift], 1u);
+        uint32_t p;
+        if (col_is_null(key0, i)) {
+            p = (uint32_t)i & (uint32_t)(n_parts - 1);
+        } else {
+            int32_t h = gx_hash_i64(((const int64_t *)key0.values)[i]);
+            p = ((uint32_t)gx_mix(h) & mask) >> shift;
+        }
+        atomicAdd(&s_cnt[p], 1u);
     }
     __syncthreads();
     for (int p = threadIdx.x; p < n_parts; p += blockDim.x)
         if (s_cnt[p]) atomicAdd(&counts[p], s_cnt[p]);
 }
 
-#define GX_RADIX_MAXP 64
-#define GX_RADIX_TILE 16384
-
 /* Block-hierarchical scatter: per-tile LDS histogram, ONE global atomicAdd
  * per (block, partition) to reserve a range, then LDS-cursor placement —
- * a flat per-row atomicAdd on <=64 cursor words serialized the whole grid
+ * a flat per-row atomicAdd on the cursor words serialized the whole grid
  * (324M returning atomics over 4 words cost ~3.5 s). */
-__global__ void k_radix_scatter(const int32_t *hashes, const uint8_t *keynull,
-                                DevColView key0, int fast_i64, int64_t n,
+__global__ void k_radix_scatter(DevColView key0, int64_t n,
                                 uint32_t mask, int shift, int n_parts,
                                 uint32_t *cursors, RadixRow *out) {
     __shared__ uint32_t s_cnt[GX_RADIX_MAXP];
@@ -445,8 +453,14 @@ __global__ void k_radix_scatter(const int32_t *hashes, const uint8_t *keynull,
         for (int p = threadIdx.x; p < n_parts; p += blockDim.x) s_cnt[p] = 0;
         __syncthreads();
         for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x) {
-            uint32_t b = keynull[i] ? 0u : ((uint32_t)gx_mix(hashes[i]) & mask);
-            atomicAdd(&s_cnt[b >> shift], 1u);
+            uint32_t p;
+            if (col_is_null(key0, i)) {
+                p = (uint32_t)i & (uint32_t)(n_parts - 1);
+            } else {
+                int32_t h = gx_hash_i64(((const int64_t *)key0.values)[i]);
+                p = ((uint32_t)gx_mix(h) & mask) >> shift;
+            }
+            atomicAdd(&s_cnt[p], 1u);
         }
         __syncthreads();
         for (int p = threadIdx.x; p < n_parts; p += blockDim.x) {
@@ -455,14 +469,20 @@ __global__ void k_radix_scatter(const int32_t *hashes, const uint8_t *keynull,
         }
         __syncthreads();
         for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x) {
-            bool kn = keynull[i];
-            uint32_t b = kn ? 0u : ((uint32_t)gx_mix(hashes[i]) & mask);
-            uint32_t p = b >> shift;
+            bool kn = col_is_null(key0, i);
+            int32_t h = 0;
+            uint32_t p;
+            if (kn) {
+                p = (uint32_t)i & (uint32_t)(n_parts - 1);
+            } else {
+                h = gx_hash_i64(((const int64_t *)key0.values)[i]);
+                p = ((uint32_t)gx_mix(h) & mask) >> shift;
+            }
             uint32_t at = s_base[p] + atomicAdd(&s_cnt[p], 1u);
             RadixRow r;
             r.rowid = (uint32_t)i | (kn ? 0x80000000u : 0u);
-            r.hash = hashes[i];
-            r.key = (!kn && fast_i64) ? ((const int64_t *)key0.values)[i] : 0;
+            r.hash = h;
+            r.key = kn ? 0 : ((const int64_t *)key0.values)[i];
             out[at] = r;
         }
         __syncthreads();
@@ -479,9 +499,11 @@ struct ProbeParams {
     const JoinEntry *entries;   /* overflow runs (count > 4) */
     uint32_t mask;
     int64_t n_probe;
-    const int32_t *hashes;      /* probe row hashes */
-    const uint8_t *keynull;     /* probe row has a NULL key */
+    const int32_t *hashes;      /* probe row hashes (generic path only) */
+    const uint8_t *keynull;     /* probe row has a NULL key (generic path) */
     const RadixRow *staged;     /* radix-staged rows (fast path); else NULL */
+    const uint32_t *part_starts;/* staged: n_parts+1 partition row offsets */
+    int32_t n_parts;
     int fast_i64;               /* entries carry the key inline */
     KeyViews build_keys;        /* for generic compare */
     KeyViews probe_keys;
@@ -552,15 +574,110 @@ __device__ static inline void emit_pair(const ProbeParams &P, EmitStage &E,
     E.cnt += n;
 }
 
-/* Per-thread row batching: each thread owns GX_PROBE_ILP rows per outer
- * iteration and phase A issues their (hash -> bucket -> e0) load chains
- * back to back — 4 independent random-line loads in flight per lane
- * instead of 1. Without it the wave-synchronous candidate walk serializes
- * one HBM latency per row (measured r02: 11.0 ms for 324M probes ≈ 29.5G
- * lines/s, half the ~55G/s random-line ceiling). Phase B then resolves
- * the slots one at a time so the ballot-compacted emission path is
- * unchanged. */
-#define GX_PROBE_ILP 4
+/* Process one probe row (or staged row at index bi), wave-synchronously:
+ * the whole wave calls this together so ballot-compaction in emit_pair
+ * sees every lane. One random 64-B line resolves the whole bucket: e0
+ * carries the entry count in its pad; e1 (lazily loaded, same line)
+ * carries the overflow start. In the fast path the bucket hash is
+ * computed inline from the key column — no k_hash_rows pass, no
+ * hashes/keynull loads.
+ *
+ * (Negative result, r2 A/B: per-thread 4-row batching of the e0 load
+ * chains — extra MLP — ran join2 at 13.8 ms vs 11.0 ms unbatched even
+ * with the slot arrays in registers; the ~8 resident waves/SIMD already
+ * overlap the walk. The lever that pays is bucket-clustered staging.) */
+__device__ static inline void probe_one(const ProbeParams &P, EmitStage &E,
+                                        bool active, int64_t bi) {
+    uint32_t i = 0;
+    bool matched = false;
+    uint32_t it = 0, end = 0;
+    int64_t want_key = 0;
+    JoinEntry e0 = {0, 0, 0};
+    uint64_t ebase = 0;
+    uint32_t ovf0 = 0;
+    if (P.staged) {
+        RadixRow r = active ? P.staged[bi] : RadixRow{0x80000000u, 0, 0};
+        i = r.rowid & 0x7FFFFFFFu;
+        if (active && !(r.rowid & 0x80000000u)) {
+            uint32_t b = (uint32_t)gx_mix(r.hash) & P.mask;
+            ebase = (uint64_t)b * 4;
+            e0 = P.table[ebase];
+            end = e0.pad;
+            want_key = r.key;
+        }
+    } else if (P.fast_i64) {
+        i = active ? (uint32_t)bi : 0;
+        const DevColView &kc = P.probe_keys.col[0];
+        if (active && !col_is_null(kc, i)) {
+            want_key = ((const int64_t *)kc.values)[i];
+            uint32_t b = (uint32_t)gx_mix(gx_hash_i64(want_key)) & P.mask;
+            ebase = (uint64_t)b * 4;
+            e0 = P.table[ebase];
+            end = e0.pad;
+        }
+    } else {
+        i = active ? (uint32_t)bi : 0;
+        if (active && !P.keynull[i]) {
+            uint32_t b = (uint32_t)gx_mix(P.hashes[i]) & P.mask;
+            ebase = (uint64_t)b * 4;
+            e0 = P.table[ebase];
+            end = e0.pad;
+        }
+    }
+
+    /* walk candidates; lanes iterate together so emissions batch */
+    while (__ballot(it < end)) {
+        bool have = active && it < end;
+        bool is_match = false;
+        uint32_t bpos = 0;
+        if (have) {
+            JoinEntry e;
+            if (it == 0) {
+                e = e0;
+            } else if (it < 4) {
+                e = P.table[ebase + it];
+                if (it == 1) ovf0 = e.pad;
+            } else {
+                e = P.entries[ovf0 + (it - 4)];
+            }
+            it++;
+            if (P.fast_i64)
+                is_match = (e.key == want_key);
+            else
+                is_match = ((int32_t)e.key == P.hashes[i]) &&
+                           rows_key_equal(P.build_keys, e.pos,
+                                          P.probe_keys, i);
+            /* residual condition: a failing candidate is skipped and
+             * never counts as a match (checkJoinCondition:206-208) */
+            if (is_match && P.n_conds)
+                is_match = join_conds_pass(P, i, e.pos);
+            bpos = e.pos;
+        }
+        if (is_match) {
+            if (P.single_join && matched) atomicExch(P.err, 1u);
+            if (P.build_matched)
+                atomicOr(&P.build_matched[bpos >> 5], 1u << (bpos & 31));
+            if (P.semi_join) { it = end; } /* first match is enough */
+            matched = true;
+        }
+        bool emit_now = is_match && !P.semi_join;
+        emit_pair(P, E, emit_now, i, bpos);
+    }
+
+    /* post-row emissions (LEFT/RIGHT null rows, SEMI/ANTI rows) */
+    bool want_null_row = active && P.outer_join && !P.build_outer && !matched;
+    emit_pair(P, E, want_null_row, i, 0xFFFFFFFFu);
+    if (P.semi_join) {
+        bool want;
+        if (P.join_type == GX_JOIN_SEMI) want = active && matched;
+        else { /* ANTI */
+            want = active && !matched;
+            if (want && P.anti_null_col >= 0)
+                want = !col_is_null(P.anti_col, i);
+        }
+        emit_pair(P, E, want, i, 0xFFFFFFFFu);
+    }
+}
 
 __global__ void k_probe(ProbeParams P) {
     __shared__ uint32_t s_pairs[2][4][GX_EMIT_STAGE]; /* [p/b][wave][slot] */
@@ -572,125 +689,24 @@ __global__ void k_probe(ProbeParams P) {
         E.cnt = 0;
         E.lane = threadIdx.x & 63;
     }
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    const int64_t sstride = stride * GX_PROBE_ILP;
-    for (int64_t base = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
-         base += sstride) {
-        /* slot 0 holds the smallest row index: if it is past the end, all
-         * slots of all lanes are (every lane stays until the wave drains
-         * so ballot-compaction in emit_pair sees the full wave) */
-        if (!__ballot(base < P.n_probe)) break;
-
-        /* phase A: independent (hash -> bucket -> e0) chains per slot.
-         * e0 carries the bucket's entry count in its pad; e1 (lazily
-         * loaded in phase B, same 64-B line) carries the overflow start. */
-        /* these slot arrays MUST become registers: without full unrolling
-         * the promote-alloca pass placed them in LDS (measured: 80 KB
-         * group segment -> 2 blocks/CU -> probe 61% SLOWER) */
-        uint32_t si[GX_PROBE_ILP];      /* probe row id (n_rows is i32) */
-        uint32_t sb[GX_PROBE_ILP];      /* bucket index */
-        JoinEntry se0[GX_PROBE_ILP];
-        uint32_t send[GX_PROBE_ILP];    /* candidate count */
-        int64_t swant[GX_PROBE_ILP];    /* fast-path probe key */
-#pragma unroll
-        for (int k = 0; k < GX_PROBE_ILP; k++) {
-            int64_t bi = base + (int64_t)k * stride;
-            bool active = bi < P.n_probe;
-            send[k] = 0;
-            swant[k] = 0;
-            sb[k] = 0;
-            if (P.staged) {
-                RadixRow r = active ? P.staged[bi]
-                                    : RadixRow{0x80000000u, 0, 0};
-                si[k] = r.rowid & 0x7FFFFFFFu;
-                if (active && !(r.rowid & 0x80000000u)) {
-                    uint32_t b = (uint32_t)gx_mix(r.hash) & P.mask;
-                    sb[k] = b;
-                    se0[k] = P.table[(uint64_t)b * 4];
-                    send[k] = se0[k].pad;
-                    swant[k] = r.key;
-                }
-            } else {
-                si[k] = active ? (uint32_t)bi : 0;
-                if (active && !P.keynull[si[k]]) {
-                    uint32_t b = (uint32_t)gx_mix(P.hashes[si[k]]) & P.mask;
-                    sb[k] = b;
-                    se0[k] = P.table[(uint64_t)b * 4];
-                    send[k] = se0[k].pad;
-                    if (P.fast_i64)
-                        swant[k] = ((const int64_t *)
-                                        P.probe_keys.col[0].values)[si[k]];
-                }
+    if (P.staged && P.part_starts) {
+        /* bucket-clustered mode: one partition (a ~2 MB table slice) per
+         * workgroup iteration — the slice's repeat probes hit this XCD's
+         * L2 instead of refetching HBM lines */
+        for (int64_t p = blockIdx.x; p < P.n_parts; p += gridDim.x) {
+            const int64_t r0 = P.part_starts[p];
+            const int64_t r1 = P.part_starts[p + 1];
+            for (int64_t base = r0 + threadIdx.x;; base += blockDim.x) {
+                if (!__ballot(base < r1)) break;
+                probe_one(P, E, base < r1, base);
             }
         }
-
-        /* phase B: resolve slots one at a time, wave-synchronously */
-#pragma unroll
-        for (int k = 0; k < GX_PROBE_ILP; k++) {
-            bool active = base + (int64_t)k * stride < P.n_probe;
-            if (!__ballot(active)) break;
-            const uint32_t i = si[k];
-            const uint64_t ebase = (uint64_t)sb[k] * 4;
-            const int64_t want_key = swant[k];
-            bool matched = false;
-            uint32_t it = 0, end = send[k];
-            uint32_t ovf0 = 0;
-
-            /* walk candidates; lanes iterate together so emissions batch */
-            while (__ballot(it < end)) {
-                bool have = active && it < end;
-                bool is_match = false;
-                uint32_t bpos = 0;
-                if (have) {
-                    JoinEntry e;
-                    if (it == 0) {
-                        e = se0[k];
-                    } else if (it < 4) {
-                        e = P.table[ebase + it];
-                        if (it == 1) ovf0 = e.pad;
-                    } else {
-                        e = P.entries[ovf0 + (it - 4)];
-                    }
-                    it++;
-                    if (P.fast_i64)
-                        is_match = (e.key == want_key);
-                    else
-                        is_match = ((int32_t)e.key == P.hashes[i]) &&
-                                   rows_key_equal(P.build_keys, e.pos,
-                                                  P.probe_keys, i);
-                    /* residual condition: a failing candidate is skipped
-                     * and never counts as a match
-                     * (checkJoinCondition:206-208) */
-                    if (is_match && P.n_conds)
-                        is_match = join_conds_pass(P, i, e.pos);
-                    bpos = e.pos;
-                }
-                if (is_match) {
-                    if (P.single_join && matched) atomicExch(P.err, 1u);
-                    if (P.build_matched)
-                        atomicOr(&P.build_matched[bpos >> 5],
-                                 1u << (bpos & 31));
-                    if (P.semi_join) { it = end; } /* first match wins */
-                    matched = true;
-                }
-                bool emit_now = is_match && !P.semi_join;
-                emit_pair(P, E, emit_now, i, bpos);
-            }
-
-            /* post-row emissions (LEFT/RIGHT null rows, SEMI/ANTI rows) */
-            bool want_null_row = active && P.outer_join && !P.build_outer &&
-                                 !matched;
-            emit_pair(P, E, want_null_row, i, 0xFFFFFFFFu);
-            if (P.semi_join) {
-                bool want;
-                if (P.join_type == GX_JOIN_SEMI) want = active && matched;
-                else { /* ANTI */
-                    want = active && !matched;
-                    if (want && P.anti_null_col >= 0)
-                        want = !col_is_null(P.anti_col, i);
-                }
-                emit_pair(P, E, want, i, 0xFFFFFFFFu);
-            }
+    } else {
+        const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+        for (int64_t base = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+             base += stride) {
+            if (!__ballot(base < P.n_probe)) break;
+            probe_one(P, E, base < P.n_probe, base);
         }
     }
     emit_flush(P, E);
@@ -1445,13 +1461,17 @@ struct JoinOp : gx_op {
             return 0;
         }
 
-        /* probe hashes */
-        if (d_ph.grow((size_t)n * 4, stream) || d_pn.grow((size_t)n, stream))
-            return -1;
+        /* probe hashes — generic path only; the fast path computes the
+         * bucket hash inline from the key column (one pass saved) */
         KeyViews pk = key_views_staged(probe_st, probe_key_cols);
-        hipLaunchKernelGGL(k_hash_rows, dim3(gx_grid(n)), dim3(256), 0, stream,
-                           pk, n, (int32_t *)d_ph.p, (uint8_t *)d_pn.p,
-                           (int)null_safe_keys);
+        if (!fast_i64) {
+            if (d_ph.grow((size_t)n * 4, stream) ||
+                d_pn.grow((size_t)n, stream))
+                return -1;
+            hipLaunchKernelGGL(k_hash_rows, dim3(gx_grid(n)), dim3(256), 0,
+                               stream, pk, n, (int32_t *)d_ph.p,
+                               (uint8_t *)d_pn.p, (int)null_safe_keys);
+        }
 
         int rc = -1;
         do {
@@ -1473,23 +1493,25 @@ struct JoinOp : gx_op {
                 std::max<int64_t>((int64_t)n * 2, 1 << 16), INT64_C(1) << 31);
             if (d_meta.grow(8, stream)) break;
 
-            /* radix-stage the probe rows by bucket range when the table far
-             * exceeds the Infinity Cache (fast path only) — slices of
-             * starts+entries become L3-resident while their rows probe. */
+            /* Bucket-clustered staging (fast path): when the table far
+             * exceeds the caches, partition the probe rows so each
+             * workgroup probes one ~2 MB bucket-range slice that stays
+             * L2-resident (see k_radix_count). Cost: ~2 streaming passes
+             * over 16-B staged rows; gain: the ~2.5 probes/bucket stop
+             * refetching random HBM lines. GX_RADIX=0 disables,
+             * GX_RADIX=1 forces, for A/B runs. */
             const int64_t table_bytes = n_buckets * 4 * (int64_t)sizeof(JoinEntry);
             int64_t n_part_radix = 1;
             int radix_shift = 0;
-            /* MEASURED OFF by default: bucket-range staging made each
-             * table slice L3-resident but probe throughput moved only +3%
-             * (the ~3.5 TB/s random-line ceiling is latency/request-path
-             * bound, not HBM-capacity bound), while staging costs ~2 passes
-             * over the probe rows. Kept for A/B runs: GX_RADIX_FORCE=1. */
-            const bool radix_force = getenv("GX_RADIX_FORCE") != nullptr;
-            if (fast_i64 && radix_force) {
-                int64_t target = radix_force ? std::max<int64_t>(table_bytes / 4, 1)
-                                             : (96 << 20);
-                n_part_radix = gx_pow2((table_bytes + target - 1) / target);
-                if (n_part_radix > 64) n_part_radix = 64;
+            const char *radix_env = getenv("GX_RADIX");
+            bool use_radix = fast_i64 &&
+                             (radix_env ? radix_env[0] == '1'
+                                        : (table_bytes > (256 << 20) &&
+                                           n >= (8 << 20)));
+            if (use_radix) {
+                const int64_t slice = 2 << 20;
+                n_part_radix = gx_pow2((table_bytes + slice - 1) / slice);
+                if (n_part_radix > GX_RADIX_MAXP) n_part_radix = GX_RADIX_MAXP;
                 if (n_part_radix > n_buckets) n_part_radix = 1;
                 while ((1 << radix_shift) < n_buckets / n_part_radix)
                     radix_shift++;
@@ -1497,12 +1519,12 @@ struct JoinOp : gx_op {
             std::vector<uint32_t> radix_starts;
             if (n_part_radix > 1) {
                 if (d_staged.grow((size_t)n * sizeof(RadixRow), stream) ||
-                    d_radix_cnt.grow((size_t)n_part_radix * 4, stream)) break;
+                    d_radix_cnt.grow((size_t)(n_part_radix + 1) * 4, stream))
+                    break;
                 HIP_OK(hipMemsetAsync(d_radix_cnt.p, 0, (size_t)n_part_radix * 4,
                                       stream));
                 hipLaunchKernelGGL(k_radix_count, dim3(gx_grid(n)), dim3(256), 0,
-                                   stream, (const int32_t *)d_ph.p,
-                                   (const uint8_t *)d_pn.p, n, mask, radix_shift,
+                                   stream, pk.col[0], n, mask, radix_shift,
                                    (int)n_part_radix, (uint32_t *)d_radix_cnt.p);
                 std::vector<uint32_t> counts((size_t)n_part_radix);
                 HIP_OK(hipMemcpyAsync(counts.data(), d_radix_cnt.p,
@@ -1519,12 +1541,15 @@ struct JoinOp : gx_op {
                                    dim3(gx_grid((n + GX_RADIX_TILE - 1) /
                                                 GX_RADIX_TILE * 256) ),
                                    dim3(256), 0, stream,
-                                   (const int32_t *)d_ph.p,
-                                   (const uint8_t *)d_pn.p,
-                                   pk.col[0], 1, n, mask, radix_shift,
+                                   pk.col[0], n, mask, radix_shift,
                                    (int)n_part_radix,
                                    (uint32_t *)d_radix_cnt.p,
                                    (RadixRow *)d_staged.p);
+                /* cursors now hold end offsets == starts[1..n]; rebuild the
+                 * starts array (incl. leading 0) for the probe kernel */
+                HIP_OK(hipMemcpyAsync(d_radix_cnt.p, radix_starts.data(),
+                                      (size_t)(n_part_radix + 1) * 4,
+                                      hipMemcpyHostToDevice, stream));
             }
 
             for (int attempt = 0; attempt < 4; attempt++) {
@@ -1539,6 +1564,8 @@ struct JoinOp : gx_op {
                 P.hashes = (const int32_t *)d_ph.p;
                 P.keynull = (const uint8_t *)d_pn.p;
                 P.staged = nullptr;
+                P.part_starts = nullptr;
+                P.n_parts = 0;
                 P.fast_i64 = (int)fast_i64;
                 P.build_keys = key_views(build, build_key_cols);
                 P.probe_keys = pk;
@@ -1590,15 +1617,11 @@ struct JoinOp : gx_op {
                 }
                 HIP_OK(hipEventRecord(ev0, stream));
                 if (n_part_radix > 1) {
-                    for (int64_t p = 0; p < n_part_radix; p++) {
-                        int64_t cnt = (int64_t)radix_starts[p + 1] - radix_starts[p];
-                        if (cnt == 0) continue;
-                        ProbeParams Pp = P;
-                        Pp.staged = (const RadixRow *)d_staged.p + radix_starts[p];
-                        Pp.n_probe = cnt;
-                        hipLaunchKernelGGL(k_probe, dim3(gx_grid(cnt)), dim3(256),
-                                           0, stream, Pp);
-                    }
+                    P.staged = (const RadixRow *)d_staged.p;
+                    P.part_starts = (const uint32_t *)d_radix_cnt.p;
+                    P.n_parts = (int32_t)n_part_radix;
+                    hipLaunchKernelGGL(k_probe, dim3((uint32_t)n_part_radix),
+                                       dim3(256), 0, stream, P);
                 } else {
                     hipLaunchKernelGGL(k_probe, dim3(gx_grid(n)), dim3(256), 0,
                                        stream, P);
