@@ -1493,21 +1493,22 @@ struct JoinOp : gx_op {
                 std::max<int64_t>((int64_t)n * 2, 1 << 16), INT64_C(1) << 31);
             if (d_meta.grow(8, stream)) break;
 
-            /* Bucket-clustered staging (fast path): when the table far
-             * exceeds the caches, partition the probe rows so each
-             * workgroup probes one ~2 MB bucket-range slice that stays
-             * L2-resident (see k_radix_count). Cost: ~2 streaming passes
-             * over 16-B staged rows; gain: the ~2.5 probes/bucket stop
-             * refetching random HBM lines. GX_RADIX=0 disables,
-             * GX_RADIX=1 forces, for A/B runs. */
+            /* Bucket-clustered staging (fast path): partition the probe
+             * rows so each workgroup probes one ~2 MB bucket-range slice.
+             * MEASURED OFF by default (r2 A/B, GX_RADIX=1 forces): the
+             * ~2.5 probes/bucket do stop refetching HBM lines, but the
+             * 30+ concurrent workgroups per XCD each hold a DIFFERENT
+             * 2 MB slice — 60+ MB of hot slices against 4 MB of XCD L2 —
+             * so the slices thrash each other and the staged probe ran
+             * 15.5 ms vs 11.0 ms unstaged (plus a 6.1 ms scatter).
+             * Making concurrent slices fit L2 needs <=128 KB slices
+             * (64K partitions, two-level scatter) — not attempted. */
             const int64_t table_bytes = n_buckets * 4 * (int64_t)sizeof(JoinEntry);
             int64_t n_part_radix = 1;
             int radix_shift = 0;
             const char *radix_env = getenv("GX_RADIX");
-            bool use_radix = fast_i64 &&
-                             (radix_env ? radix_env[0] == '1'
-                                        : (table_bytes > (256 << 20) &&
-                                           n >= (8 << 20)));
+            bool use_radix = fast_i64 && radix_env && radix_env[0] == '1';
+            (void)table_bytes;
             if (use_radix) {
                 const int64_t slice = 2 << 20;
                 n_part_radix = gx_pow2((table_bytes + slice - 1) / slice);
