@@ -170,6 +170,8 @@ class GxFrameSpec(C.Structure):
         ("kind", C.c_int32),
         ("preceding", C.c_int64),
         ("following", C.c_int64),
+        ("order_col", C.c_int32),   # RANGE kinds: the ORDER BY column
+        ("order_asc", C.c_int32),   # RANGE kinds: 1 = asc, 0 = desc
     ]
 
 
@@ -232,11 +234,16 @@ PROJ_COPY, PROJ_REV_F64, PROJ_REV_SCALED4, PROJ_Q9_AMOUNT4 = 0, 1, 2, 3
 PROJ_DEC_TO_SCALED, PROJ_SCALED_TO_DEC = 4, 5  # gx_proj.c = decimal scale
 FRAME_WHOLE_PARTITION, FRAME_ROWS_SLIDING, FRAME_ROWS_UNBOUNDED_FOLLOWING = \
     0, 1, 2
+(FRAME_RANGE_SLIDING, FRAME_RANGE_UNBOUNDED_PRECEDING,
+ FRAME_RANGE_UNBOUNDED_FOLLOWING) = 3, 4, 5
 # Agg funcs (gx_agg_func)
 (COUNT_ROW, COUNT_COL, SUM_I64, SUM_F64, MIN_I64, MAX_I64, MIN_F64,
  MAX_F64, AVG_F64, BIT_AND, BIT_OR, BIT_XOR, RANK, DENSE_RANK,
  FIRST_VALUE, LAST_VALUE, NTH_VALUE, LAG, LEAD, NTILE, CUME_DIST,
  PERCENT_RANK) = range(22)
+# SQL SUM over integers, NULL-init (the Sum family AggregateUtils maps
+# SqlKind.SUM to); SUM_I64 is SqlKind.SUM0 (Long2LongSum0, init 0)
+SUM_I64N = 22
 
 ORACLE_PATH = os.path.join(_REPO, "oracle", "libgxoracle.so")
 HIP_PATH = os.path.join(_REPO, "galaxysql_amd", "csrc", "libgxhip.so")

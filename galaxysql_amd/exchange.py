@@ -129,7 +129,7 @@ def final_agg_specs(n_group_cols, aggs):
         if func in (abi.COUNT_ROW, abi.COUNT_COL):
             finals.append((abi.SUM_I64, col))
             out_types.append(0)  # I64
-        elif func in (abi.SUM_I64, abi.MIN_I64, abi.MAX_I64,
+        elif func in (abi.SUM_I64, abi.SUM_I64N, abi.MIN_I64, abi.MAX_I64,
                       abi.BIT_AND, abi.BIT_OR, abi.BIT_XOR):
             # bit aggs re-aggregate as themselves over I64 partial values
             finals.append((func, col))

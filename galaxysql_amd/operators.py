@@ -382,8 +382,10 @@ class OverWindowFramesExec:
         oc = (C.c_int32 * max(1, len(oc_list)))(*(oc_list or [0]))
         fr = (GxFrameSpec * len(frames))()
         for i, spec in enumerate(frames):
-            spec = tuple(spec) + (0,) * (5 - len(spec))
-            fr[i] = GxFrameSpec(*spec[:5])
+            # (func, input_col, kind[, preceding, following,
+            #  order_col, order_asc]) — RANGE kinds need the last two
+            spec = tuple(spec) + (0,) * (7 - len(spec))
+            fr[i] = GxFrameSpec(*spec[:7])
         it = (C.c_int32 * len(input_types))(*input_types)
         cfg = GxFWindowCfg(n_part_cols=len(part_cols), part_cols=pc,
                            n_order_cols=len(oc_list), order_cols=oc,

@@ -241,7 +241,15 @@ typedef enum gx_agg_func {
     GX_AGG_LEAD        = 18, /* NULL past partition end */
     GX_AGG_NTILE       = 19, /* MySQL split: first size%n buckets get +1 */
     GX_AGG_CUME_DIST   = 20, /* rows <= current ORDER run / partition size */
-    GX_AGG_PERCENT_RANK = 21 /* (rank-1) / (partition size-1); 0 if size 1 */
+    GX_AGG_PERCENT_RANK = 21,/* (rank-1) / (partition size-1); 0 if size 1 */
+    GX_AGG_SUM_I64N  = 22  /* SQL SUM over integers, NULL-init (the Sum /
+                              Long2DecimalSum family AggregateUtils maps
+                              SqlKind.SUM to, AggregateUtils.java:175-197):
+                              an empty or all-NULL group/frame emits NULL,
+                              unlike SUM_I64 = Long2LongSum0 (SqlKind.SUM0,
+                              init 0). The window frame classes use the
+                              same null-init Sum (OverWindowFramesExecTest
+                              fixtures). */
 } gx_agg_func;
 
 typedef struct gx_agg_spec {

@@ -853,6 +853,8 @@ struct AggOp : gx_op {
                 s.i64.push_back(0); s.isnull.push_back(0); break;
             case GX_AGG_SUM_I64:
                 s.i64.push_back(0); s.isnull.push_back(0); break; /* Long2LongSum0 init 0 */
+            case GX_AGG_SUM_I64N:
+                s.i64.push_back(0); s.isnull.push_back(1); break; /* Sum: init NULL */
             case GX_AGG_SUM_F64: case GX_AGG_MIN_F64: case GX_AGG_MAX_F64:
                 s.f64.push_back(0); s.isnull.push_back(1); break; /* init NULL */
             case GX_AGG_AVG_F64:
@@ -931,6 +933,13 @@ struct AggOp : gx_op {
                 s.i64[gid] = (int64_t)((uint64_t)s.i64[gid] +
                     (uint64_t)(c->type == GX_I32 ? (int64_t)c->i32v[row] : c->i64v[row]));
             break;
+        case GX_AGG_SUM_I64N: /* null-init Sum (AggregateUtils SqlKind.SUM) */
+            if (!c->is_null(row)) {
+                s.i64[gid] = (int64_t)((uint64_t)s.i64[gid] +
+                    (uint64_t)(c->type == GX_I32 ? (int64_t)c->i32v[row] : c->i64v[row]));
+                s.isnull[gid] = 0;
+            }
+            break;
         case GX_AGG_SUM_F64:
             if (!c->is_null(row)) {
                 double v = c->type == GX_F64 ? c->f64v[row]
@@ -999,7 +1008,8 @@ struct AggOp : gx_op {
         for (auto &sp : aggs) {
             switch (sp.func) {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
-            case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            case GX_AGG_SUM_I64: case GX_AGG_SUM_I64N:
+            case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
             case GX_AGG_RANK: case GX_AGG_DENSE_RANK:
                 otypes.push_back(GX_I64); break;
@@ -1210,7 +1220,8 @@ struct GroupJoinOp : gx_op {
         for (auto &sp : aggs) {
             switch (sp.func) {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
-            case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            case GX_AGG_SUM_I64: case GX_AGG_SUM_I64N:
+            case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
             case GX_AGG_RANK: case GX_AGG_DENSE_RANK:
                 otypes.push_back(GX_I64); break;
@@ -1315,6 +1326,13 @@ struct WindowOp : gx_op {
                 s.i64 = (int64_t)((uint64_t)s.i64 + (uint64_t)(
                     c->type == GX_I32 ? (int64_t)c->i32v[r] : c->i64v[r]));
             break;
+        case GX_AGG_SUM_I64N: /* null-init running Sum */
+            if (!c->is_null(r)) {
+                s.i64 = (int64_t)((uint64_t)s.i64 + (uint64_t)(
+                    c->type == GX_I32 ? (int64_t)c->i32v[r] : c->i64v[r]));
+                s.isnull = 0;
+            }
+            break;
         case GX_AGG_SUM_F64:
             if (!c->is_null(r)) {
                 double v = c->type == GX_F64 ? c->f64v[r]
@@ -1412,7 +1430,8 @@ struct WindowOp : gx_op {
         for (auto &sp : aggs) {
             switch (sp.func) {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
-            case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            case GX_AGG_SUM_I64: case GX_AGG_SUM_I64N:
+            case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
             case GX_AGG_RANK: case GX_AGG_DENSE_RANK:
                 otypes.push_back(GX_I64); break;
@@ -1527,6 +1546,81 @@ struct FWindowOp : gx_op {
         return 0;
     }
 
+    /* RANGE frame index bounds, restated with the reference's own linear
+     * scans (RangeSlidingOverFrame.getBound:117-138,
+     * RangeUnboundedPrecedingOverFrame.getBound:144-168,
+     * RangeUnboundedFollowingOverFrame.getBound:133-155; null-run
+     * handling per updateIndex/updateNullRows in each class — nulls sort
+     * first when ascending, last when descending). Returns INCLUSIVE
+     * [lo, hi]; hi < lo = empty frame. */
+    void range_bounds(const gx_frame_spec &f, size_t i, size_t s, size_t e,
+                      int64_t &lo, int64_t &hi) const {
+        const Column &oc = input.cols[f.order_col];
+        const bool asc = f.order_asc != 0;
+        const int64_t dir = asc ? 1 : -1;
+        auto is_null = [&](int64_t j) { return oc.is_null((size_t)j); };
+        /* value(a) <= value(b) + range — the compare() ladder
+         * (RangeSlidingOverFrame.java:159-186, numeric types) */
+        auto le_plus = [&](int64_t a, int64_t b, int64_t range) {
+            if (oc.type == GX_F64)
+                return oc.f64v[a] <= oc.f64v[b] + (double)range;
+            int64_t va = oc.type == GX_I32 ? (int64_t)oc.i32v[a] : oc.i64v[a];
+            int64_t vb = oc.type == GX_I32 ? (int64_t)oc.i32v[b] : oc.i64v[b];
+            return va <= vb + range;
+        };
+        if (is_null((int64_t)i)) {
+            int64_t rs = (int64_t)i, re = (int64_t)i;
+            while (rs - 1 >= (int64_t)s && is_null(rs - 1)) rs--;
+            while (re + 1 < (int64_t)e && is_null(re + 1)) re++;
+            switch (f.kind) {
+            case GX_FRAME_RANGE_SLIDING: lo = rs; hi = re; break;
+            case GX_FRAME_RANGE_UNBOUNDED_PRECEDING:
+                lo = (int64_t)s; hi = asc ? re : (int64_t)e - 1; break;
+            default: /* RANGE_UNBOUNDED_FOLLOWING */
+                lo = asc ? (int64_t)s : rs; hi = (int64_t)e - 1; break;
+            }
+            return;
+        }
+        switch (f.kind) {
+        case GX_FRAME_RANGE_SLIDING: {
+            int64_t hiI = (int64_t)i;
+            while (hiI >= (int64_t)s && hiI < (int64_t)e && !is_null(hiI) &&
+                   le_plus(hiI, (int64_t)i, f.following))
+                hiI += dir;
+            hiI -= dir;
+            int64_t loI = (int64_t)i;
+            while (loI >= (int64_t)s && loI < (int64_t)e && !is_null(loI) &&
+                   le_plus((int64_t)i, loI, f.preceding))
+                loI -= dir;
+            loI += dir;
+            lo = asc ? loI : hiI;
+            hi = asc ? hiI : loI;
+            break; }
+        case GX_FRAME_RANGE_UNBOUNDED_PRECEDING: {
+            int64_t other = (int64_t)i;
+            while (other >= (int64_t)s && other < (int64_t)e &&
+                   !is_null(other) &&
+                   (asc ? le_plus(other, (int64_t)i, f.following)
+                        : le_plus((int64_t)i, other, f.following)))
+                other += 1;
+            if (other != (int64_t)i) other -= 1;
+            lo = (int64_t)s;
+            hi = other;
+            break; }
+        default: { /* RANGE_UNBOUNDED_FOLLOWING */
+            int64_t other = (int64_t)i;
+            while (other >= (int64_t)s && other < (int64_t)e &&
+                   !is_null(other) &&
+                   (asc ? le_plus((int64_t)i, other, f.preceding)
+                        : le_plus(other, (int64_t)i, f.preceding)))
+                other -= 1;
+            if (other != (int64_t)i) other += 1;
+            lo = other;
+            hi = (int64_t)e - 1;
+            break; }
+        }
+    }
+
     /* accumulate rows [lo, hi) of frame f into (i64 or f64, isnull) */
     void range_agg(const gx_frame_spec &f, size_t lo, size_t hi,
                    int64_t &iv, double &dv, bool &isnull) const {
@@ -1549,6 +1643,13 @@ struct FWindowOp : gx_op {
                 if (!c->is_null(r))
                     iv = (int64_t)((uint64_t)iv + (uint64_t)(
                         c->type == GX_I32 ? (int64_t)c->i32v[r] : c->i64v[r]));
+                break;
+            case GX_AGG_SUM_I64N:
+                if (!c->is_null(r)) {
+                    iv = (int64_t)((uint64_t)iv + (uint64_t)(
+                        c->type == GX_I32 ? (int64_t)c->i32v[r] : c->i64v[r]));
+                    isnull = false;
+                }
                 break;
             case GX_AGG_SUM_F64:
                 if (!c->is_null(r)) {
@@ -1606,7 +1707,8 @@ struct FWindowOp : gx_op {
         for (auto &f : frames) {
             switch (f.func) {
             case GX_AGG_COUNT_ROW: case GX_AGG_COUNT_COL:
-            case GX_AGG_SUM_I64: case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
+            case GX_AGG_SUM_I64: case GX_AGG_SUM_I64N:
+            case GX_AGG_MIN_I64: case GX_AGG_MAX_I64:
             case GX_AGG_BIT_AND: case GX_AGG_BIT_OR: case GX_AGG_BIT_XOR:
             case GX_AGG_RANK: case GX_AGG_DENSE_RANK: case GX_AGG_NTILE:
                 otypes.push_back(GX_I64); break;
@@ -1626,7 +1728,8 @@ struct FWindowOp : gx_op {
                 const gx_frame_spec &f = frames[a];
                 size_t s = seg_start_of_row[i], e = seg_end_of_row[i];
                 size_t col2 = input_types.size() + a;
-                if (f.func >= GX_AGG_FIRST_VALUE) {
+                if (f.func >= GX_AGG_FIRST_VALUE &&
+                    f.func <= GX_AGG_PERCENT_RANK) {
                     /* navigation / distribution over the whole partition */
                     int64_t size = (int64_t)(e - s), pos = (int64_t)(i - s);
                     int64_t src = -1;
@@ -1677,6 +1780,13 @@ struct FWindowOp : gx_op {
                     hi = std::min<size_t>(e, i + (size_t)f.following + 1);
                 } else if (f.kind == GX_FRAME_ROWS_UNBOUNDED_FOLLOWING) {
                     lo = i;
+                } else if (f.kind == GX_FRAME_RANGE_SLIDING ||
+                           f.kind == GX_FRAME_RANGE_UNBOUNDED_PRECEDING ||
+                           f.kind == GX_FRAME_RANGE_UNBOUNDED_FOLLOWING) {
+                    int64_t l, h;
+                    range_bounds(f, i, s, e, l, h);
+                    lo = (size_t)l;
+                    hi = h < l ? (size_t)l : (size_t)(h + 1); /* exclusive */
                 }
                 int64_t iv; double dv; bool isnull;
                 range_agg(f, lo, hi, iv, dv, isnull);
@@ -2073,7 +2183,7 @@ static int fwindow_cfg_ok(const gx_fwindow_cfg *c) {
     if (!c || c->n_frames <= 0) return 0;
     for (int32_t i = 0; i < c->n_frames; i++) {
         const gx_frame_spec &f = c->frames[i];
-        if (f.func >= GX_AGG_FIRST_VALUE) {
+        if (f.func >= GX_AGG_FIRST_VALUE && f.func <= GX_AGG_PERCENT_RANK) {
             /* navigation funcs: whole-partition only; NTH/LAG/LEAD/NTILE
              * need a positive parameter; CUME/PERCENT need ORDER cols */
             if (f.kind != GX_FRAME_WHOLE_PARTITION) return 0;
@@ -2087,18 +2197,28 @@ static int fwindow_cfg_ok(const gx_fwindow_cfg *c) {
             continue;
         }
         if (f.kind == GX_FRAME_WHOLE_PARTITION) continue;
-        /* sliding: additive funcs via prefix differences, MIN/MAX via
-         * sparse tables (fixed span); following: additive only */
+        /* sliding/RANGE: additive + SUM/AVG(F64) + MIN/MAX;
+         * following: additive + SUM/AVG(F64) (matches the HIP create) */
         bool additive = f.func == GX_AGG_COUNT_ROW ||
                         f.func == GX_AGG_COUNT_COL ||
-                        f.func == GX_AGG_SUM_I64;
+                        f.func == GX_AGG_SUM_I64 ||
+                        f.func == GX_AGG_SUM_I64N;
+        bool f64sum = f.func == GX_AGG_SUM_F64 || f.func == GX_AGG_AVG_F64;
         bool minmax = f.func == GX_AGG_MIN_I64 || f.func == GX_AGG_MAX_I64 ||
                       f.func == GX_AGG_MIN_F64 || f.func == GX_AGG_MAX_F64;
+        bool is_range = f.kind == GX_FRAME_RANGE_SLIDING ||
+                        f.kind == GX_FRAME_RANGE_UNBOUNDED_PRECEDING ||
+                        f.kind == GX_FRAME_RANGE_UNBOUNDED_FOLLOWING;
         if (f.kind == GX_FRAME_ROWS_SLIDING) {
-            if (!additive && !minmax) return 0;
+            if (!additive && !minmax && !f64sum) return 0;
             if (f.preceding < 0 || f.following < 0) return 0;
+        } else if (is_range) {
+            if (!additive && !minmax && !f64sum) return 0;
+            if (f.order_col < 0 || f.order_col >= c->n_input_cols) return 0;
+            int32_t ot = c->input_types[f.order_col];
+            if (ot != GX_I64 && ot != GX_I32 && ot != GX_F64) return 0;
         } else {
-            if (!additive) return 0;
+            if (!additive && !f64sum) return 0;
         }
     }
     return 1;

@@ -16,7 +16,8 @@ CMP_BY_NAME = {"LT": abi.LT, "LE": abi.LE, "GT": abi.GT, "GE": abi.GE,
                "EQ": abi.EQ, "NE": abi.NE,
                "EQ_NULLSAFE": abi.EQ_NULLSAFE, "NE_NULLSAFE": abi.NE_NULLSAFE}
 AGG_BY_NAME = {"COUNT_ROW": abi.COUNT_ROW, "COUNT_COL": abi.COUNT_COL,
-               "SUM_I64": abi.SUM_I64, "SUM_F64": abi.SUM_F64,
+               "SUM_I64": abi.SUM_I64, "SUM_I64N": abi.SUM_I64N,
+               "SUM_F64": abi.SUM_F64,
                "MIN_I64": abi.MIN_I64, "MAX_I64": abi.MAX_I64,
                "MIN_F64": abi.MIN_F64, "MAX_F64": abi.MAX_F64}
 

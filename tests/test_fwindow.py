@@ -9,7 +9,8 @@ import numpy as np
 import pytest
 
 from galaxysql_amd import abi
-from galaxysql_amd.chunk import Block, Chunk, I64, F64, chunks_from_columns, \
+from galaxysql_amd.chunk import Block, Chunk, I64, F64, SLICE, \
+    chunks_from_columns, \
     rows_of
 from galaxysql_amd.operators import run_fwindow
 
@@ -40,16 +41,18 @@ def test_fwindow_hand_case():
 
 def test_fwindow_rejects_unsupported():
     lib = abi.load_oracle()
+    # (sliding SUM_F64 and RANGE frames became supported in round 2)
     with pytest.raises(RuntimeError):
-        # sliding SUM(DOUBLE): prefix differences change fp rounding
-        run_fwindow(lib, [0],
-                    [(abi.SUM_F64, 1, abi.FRAME_ROWS_SLIDING, 1, 1)],
-                    [I64, F64], [])
-    with pytest.raises(RuntimeError):
-        # MIN in an unbounded-following frame is still round-2
+        # MIN in an unbounded-following frame is still unsupported
         run_fwindow(lib, [0],
                     [(abi.MIN_I64, 1, abi.FRAME_ROWS_UNBOUNDED_FOLLOWING)],
                     [I64, I64], [])
+    with pytest.raises(RuntimeError):
+        # RANGE frame with a non-numeric order col
+        # (RangeSlidingOverFrame.java:36: numeric types only)
+        run_fwindow(lib, [0],
+                    [(abi.SUM_I64N, 1, abi.FRAME_RANGE_SLIDING, 1, 1, 2, 1)],
+                    [I64, I64, SLICE], [])
 
 
 def _gen(rng, n):
