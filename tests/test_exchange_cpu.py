@@ -15,30 +15,31 @@ WORKER = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                       "_exchange_worker.py")
 
 
-def test_shuffle_world2(tmp_path):
+@pytest.mark.parametrize("world", [2, 8])
+def test_shuffle_world(tmp_path, world):
     subprocess.run(["make", "-C", os.path.join(REPO, "oracle")], check=True,
                    capture_output=True)
     env = dict(os.environ)
-    env.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29612", WORLD_SIZE="2",
-               PYTHONPATH=REPO)
+    env.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(29612 + world),
+               WORLD_SIZE=str(world), PYTHONPATH=REPO, OMP_NUM_THREADS="1")
     procs = []
-    for rank in range(2):
+    for rank in range(world):
         e = dict(env, RANK=str(rank))
         procs.append(subprocess.Popen(
             [sys.executable, WORKER, str(tmp_path)], env=e,
             stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
-    outs = [p.communicate(timeout=120)[0].decode() for p in procs]
+    outs = [p.communicate(timeout=240)[0].decode() for p in procs]
     for rank, p in enumerate(procs):
         assert p.returncode == 0, f"rank {rank} failed:\n{outs[rank]}"
     # workers wrote per-rank received rows; verify completeness here
     recv = []
-    for rank in range(2):
+    for rank in range(world):
         recv.append(np.load(tmp_path / f"recv_{rank}.npz"))
     all_keys = np.concatenate([r["keys"] for r in recv])
     all_pay = np.concatenate([r["pay"] for r in recv])
     sent_keys = np.concatenate([np.load(tmp_path / f"sent_{r}.npz")["keys"]
-                                for r in range(2)])
+                                for r in range(world)])
     sent_pay = np.concatenate([np.load(tmp_path / f"sent_{r}.npz")["pay"]
-                               for r in range(2)])
+                               for r in range(world)])
     assert sorted(zip(all_keys.tolist(), all_pay.tolist())) == \
            sorted(zip(sent_keys.tolist(), sent_pay.tolist()))
