@@ -181,6 +181,7 @@ class C2Chunk(C2):
             args.sf = 1
         super().__init__(args, device, rank)
         self.chunk_size = args.chunk_size
+        self.flush_rows = args.flush_rows
 
     def step(self, lib, local_rank, distributed):
         import ctypes as C
@@ -201,13 +202,38 @@ class C2Chunk(C2):
             op.build_consume()
             n_out = 0
             cs = self.chunk_size
+            fr = self.flush_rows
+            pushed = 0
             for lo in range(0, self.n_probe, cs):
                 ka2 = []
                 sl = [t[lo:lo + cs] for t in p]
                 pc = chunk_from_torch(lib, sl, [I64, I64], ka2)
+                if fr > 0:
+                    # buffered cadence: push chunk, flush per `fr` rows
+                    # (the shim's LocalBufferExec-style amortization)
+                    lib.check(lib.lib.gxop_join_probe_push(op._op,
+                                                           C.byref(pc)),
+                              "push")
+                    pushed += sl[0].numel()
+                    if pushed >= fr:
+                        out = C.POINTER(GxResult)()
+                        lib.check(lib.lib.gxop_join_probe_flush(
+                            op._op, C.byref(out)), "flush")
+                        if out:
+                            n_out += out.contents.chunk.n_rows
+                            lib.lib.gxop_result_release(out)
+                        pushed = 0
+                else:
+                    out = C.POINTER(GxResult)()
+                    lib.check(lib.lib.gxop_join_probe(op._op, C.byref(pc),
+                                                      C.byref(out)), "probe")
+                    if out:
+                        n_out += out.contents.chunk.n_rows
+                        lib.lib.gxop_result_release(out)
+            if fr > 0 and pushed:
                 out = C.POINTER(GxResult)()
-                lib.check(lib.lib.gxop_join_probe(op._op, C.byref(pc),
-                                                  C.byref(out)), "probe")
+                lib.check(lib.lib.gxop_join_probe_flush(op._op, C.byref(out)),
+                          "flush")
                 if out:
                     n_out += out.contents.chunk.n_rows
                     lib.lib.gxop_result_release(out)
@@ -223,8 +249,11 @@ class C2Chunk(C2):
         c = super().config(world)
         c["workload"] = f"C2chunk_sf{self.sf}_probe_cadence"
         c["chunk_size"] = self.chunk_size
-        c["note"] = ("probe pushed in CHUNK_SIZE-row ABI calls; compare "
-                     "value against the monolithic c2 line")
+        c["flush_rows"] = self.flush_rows
+        c["note"] = ("probe pushed in CHUNK_SIZE-row ABI calls "
+                     "(flush_rows=0: one gxop_join_probe per chunk; >0: "
+                     "probe_push per chunk + probe_flush per flush_rows); "
+                     "compare value against the monolithic c2 line")
         return c
 
 
@@ -732,6 +761,9 @@ def main():
     ap.add_argument("--sf", type=int, default=None, help="C2 scale factor")
     ap.add_argument("--chunk-size", type=int, default=1000,
                     help="c2chunk probe cadence (CHUNK_SIZE default 1000)")
+    ap.add_argument("--flush-rows", type=int, default=0,
+                    help="c2chunk: 0 = probe per chunk; >0 = probe_push "
+                         "per chunk + probe_flush per this many rows")
     ap.add_argument("--c3-scale", type=float, default=1.0,
                     help="C3 size fraction of SF100 (1.0 = full)")
     ap.add_argument("--c4-scale", type=float, default=1.0,

@@ -269,6 +269,9 @@ class GxLib:
         L.gxop_join_probe.argtypes = [C.c_void_p, C.POINTER(GxChunk),
                                       C.POINTER(C.POINTER(GxResult))]
         L.gxop_join_tail.argtypes = [C.c_void_p, C.POINTER(C.POINTER(GxResult))]
+        L.gxop_join_probe_push.argtypes = [C.c_void_p, C.POINTER(GxChunk)]
+        L.gxop_join_probe_flush.argtypes = [
+            C.c_void_p, C.POINTER(C.POINTER(GxResult))]
         L.gxop_join_close.argtypes = [C.c_void_p]
         L.gxop_agg_create.restype = C.c_void_p
         L.gxop_agg_create.argtypes = [C.POINTER(GxAggCfg)]

@@ -1083,6 +1083,8 @@ struct JoinOp : gx_op {
     std::vector<int> build_key_cols, probe_key_cols;
 
     DevStore build;          /* build-side columns in HBM */
+    DevStore probe_buf;      /* buffered-probe staging (probe_push/flush) */
+    bool probe_buf_init = false;
     DevBuf d_hashes, d_keynull;
     DevBuf d_table;          /* inline-bucket table (plain joins) */
     DevBuf d_counts, d_starts, d_entries; /* CSR (null_safe) / IB overflow */
@@ -1132,6 +1134,7 @@ struct JoinOp : gx_op {
     }
     ~JoinOp() override {
         build.release();
+        probe_buf.release();
         d_table.release();
         d_hashes.release(); d_keynull.release(); d_counts.release();
         d_starts.release(); d_entries.release(); d_bitmap.release();
@@ -1660,6 +1663,49 @@ struct JoinOp : gx_op {
         return rc;
     }
 
+    /* Buffered probe (gxop.h: the LocalBufferExec pattern): push appends
+     * the chunk's columns into a device store (async copies only); flush
+     * runs ONE probe over the accumulated batch. Amortizes the ~60 us
+     * launch+sync cost of a per-chunk probe call across many CHUNK_SIZE
+     * chunks. */
+    int probe_push(const gx_chunk *ch) {
+        if (ensure_device(device)) return -1;
+        if (!built) { gx_set_err("probe before build"); return -1; }
+        if (!probe_buf_init) {
+            const auto &pt = cfg.build_outer ? inner_types : outer_types;
+            probe_buf.init((int32_t)pt.size(), pt.data(), stream);
+            probe_buf_init = true;
+        }
+        return probe_buf.append(ch);
+    }
+
+    int probe_flush(gx_result **out) {
+        *out = nullptr;
+        if (!probe_buf_init || probe_buf.n_rows == 0) return 0;
+        if (probe_buf.n_rows > INT32_MAX) {
+            gx_set_err("buffered probe exceeds 2^31 rows; flush earlier");
+            return -1;
+        }
+        std::vector<gx_block> blocks(probe_buf.cols.size());
+        for (size_t c = 0; c < probe_buf.cols.size(); c++) {
+            DevColView v = probe_buf.view((int32_t)c);
+            gx_block &b = blocks[c];
+            std::memset(&b, 0, sizeof(b));
+            b.type = v.type;
+            b.mem = GX_MEM_DEVICE;
+            b.values = v.values;
+            b.nulls = v.has_nulls ? v.nulls : nullptr;
+            b.offsets = v.offsets;
+            b.data = v.bytes;
+        }
+        gx_chunk ch{(int32_t)probe_buf.n_rows, (int32_t)blocks.size(),
+                    blocks.data()};
+        int rc = probe(&ch, out);
+        probe_buf.release();
+        probe_buf_init = false;
+        return rc;
+    }
+
     int tail(gx_result **out) {
         *out = nullptr;
         if (!cfg.build_outer || tail_done) { tail_done = true; return 0; }
@@ -1779,6 +1825,22 @@ int gxop_join_tail(gx_op *op, gx_result **out) {
     std::lock_guard<std::mutex> lk(op->mu);
     if (op->variant == 1) return static_cast<HybridJoinOp *>(op)->tail(out);
     return static_cast<JoinOp *>(op)->tail(out);
+}
+int gxop_join_probe_push(gx_op *op, const gx_chunk *c) {
+    if (!op || op->kind != OP_JOIN) { gx_set_err("not a join op"); return -1; }
+    std::lock_guard<std::mutex> lk(op->mu);
+    if (op->variant == 1) {
+        /* hybrid defers all probe output anyway: push = spill the chunk */
+        gx_result *ignored = nullptr;
+        return static_cast<HybridJoinOp *>(op)->probe(c, &ignored);
+    }
+    return static_cast<JoinOp *>(op)->probe_push(c);
+}
+int gxop_join_probe_flush(gx_op *op, gx_result **out) {
+    if (!op || op->kind != OP_JOIN) { gx_set_err("not a join op"); return -1; }
+    std::lock_guard<std::mutex> lk(op->mu);
+    if (op->variant == 1) { *out = nullptr; return 0; } /* drained via tail */
+    return static_cast<JoinOp *>(op)->probe_flush(out);
 }
 int gxop_join_close(gx_op *op) { delete op; return 0; }
 

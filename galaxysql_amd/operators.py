@@ -119,6 +119,25 @@ class ParallelHashJoinExec:
             return None
         return self._lib.result_to_chunk(out)
 
+    def probe_push(self, chunk: Chunk):
+        """Buffered probe (gxop.h: the LocalBufferExec pattern): stage a
+        chunk device-side without launching kernels."""
+        ka = []
+        gc = self._lib.to_gx_chunk(chunk, ka)
+        self._lib.check(self._lib.lib.gxop_join_probe_push(self._op,
+                                                           C.byref(gc)),
+                        "join_probe_push")
+
+    def probe_flush(self) -> Chunk:
+        """Probe the accumulated batch; None when nothing was buffered."""
+        out = C.POINTER(GxResult)()
+        self._lib.check(self._lib.lib.gxop_join_probe_flush(self._op,
+                                                            C.byref(out)),
+                        "join_probe_flush")
+        if not out:
+            return None
+        return self._lib.result_to_chunk(out)
+
     def stats(self):
         """Cumulative probe-kernel stats (bench roofline leg)."""
         from .abi import GxJoinStats

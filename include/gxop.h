@@ -189,6 +189,17 @@ int gxop_join_build(gx_op *op);
 /* probe with one chunk; *out receives the joined rows for this chunk
  * (may be empty; may exceed 1000 rows — the shim re-chunks). */
 int gxop_join_probe(gx_op *op, const gx_chunk *probe_chunk, gx_result **out);
+/* Buffered probe — the reference's own chunk-buffering pattern between
+ * pipelines (LocalBufferExec.java:35-55): push stages a probe chunk
+ * device-side (two async copies per column, NO kernel launches, NO sync);
+ * flush probes the whole accumulated batch and returns ONE result
+ * (*out = NULL when nothing is buffered). Rationale, measured: one
+ * gxop_join_probe call costs ~60 us of launches+syncs, so pushing the
+ * CN's CHUNK_SIZE=1000-row chunks through it runs at ~16M rows/s vs
+ * 7.5G/s monolithic; the JNI shim buffers chunks via push and flushes
+ * at >=64K rows (bench.py c2chunk measures both cadences). */
+int gxop_join_probe_push(gx_op *op, const gx_chunk *probe_chunk);
+int gxop_join_probe_flush(gx_op *op, gx_result **out);
 /* drain pass-through / outer-null tail rows after the last probe chunk;
  * returns 0 and *out=NULL when exhausted. */
 int gxop_join_tail(gx_op *op, gx_result **out);

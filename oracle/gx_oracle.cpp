@@ -471,6 +471,52 @@ struct JoinOp : gx_op {
 
     int consume(const gx_chunk *ch) { return build.append(ch); }
 
+    /* buffered probe (gxop.h: the LocalBufferExec pattern) */
+    Store probe_bufs;
+    bool probe_buf_init = false;
+
+    int probe_push(const gx_chunk *ch) {
+        if (!built) { set_err("probe before build"); return -1; }
+        if (!probe_buf_init) {
+            const auto &pt = cfg.build_outer ? inner_types : outer_types;
+            probe_bufs.init((int32_t)pt.size(), pt.data());
+            probe_buf_init = true;
+        }
+        return probe_bufs.append(ch);
+    }
+
+    int probe_flush(gx_result **out) {
+        *out = nullptr;
+        if (!probe_buf_init || probe_bufs.n_rows == 0) return 0;
+        std::vector<gx_block> blocks(probe_bufs.cols.size());
+        for (size_t c = 0; c < probe_bufs.cols.size(); c++) {
+            Column &col = probe_bufs.cols[c];
+            gx_block &b = blocks[c];
+            std::memset(&b, 0, sizeof(b));
+            b.type = col.type;
+            b.mem = GX_MEM_HOST;
+            b.nulls = col.null_.data();
+            switch (col.type) {
+            case GX_I64: b.values = col.i64v.data(); break;
+            case GX_I32: b.values = col.i32v.data(); break;
+            case GX_F64: b.values = col.f64v.data(); break;
+            case GX_DECIMAL: b.values = col.bytes.data(); break;
+            case GX_SLICE:
+                /* store offsets start at 0 = chunk-local for one chunk */
+                b.offsets = col.off.data();
+                b.data = col.bytes.data();
+                break;
+            }
+        }
+        gx_chunk ch{(int32_t)probe_bufs.n_rows, (int32_t)blocks.size(),
+                    blocks.data()};
+        int rc = probe(&ch, out);
+        probe_bufs = Store();
+        probe_buf_init = false;
+        return rc;
+    }
+
+
     int do_build() {
         if (built) return 0; /* first-come barrier */
         const int64_t size = (int64_t)build.n_rows;
@@ -2103,6 +2149,14 @@ int gxop_join_probe(gx_op *op, const gx_chunk *c, gx_result **out) {
 int gxop_join_tail(gx_op *op, gx_result **out) {
     if (!op || op->kind != OP_JOIN) { set_err("not a join op"); return -1; }
     return static_cast<JoinOp *>(op)->tail(out);
+}
+int gxop_join_probe_push(gx_op *op, const gx_chunk *c) {
+    if (!op || op->kind != OP_JOIN) { set_err("not a join op"); return -1; }
+    return static_cast<JoinOp *>(op)->probe_push(c);
+}
+int gxop_join_probe_flush(gx_op *op, gx_result **out) {
+    if (!op || op->kind != OP_JOIN) { set_err("not a join op"); return -1; }
+    return static_cast<JoinOp *>(op)->probe_flush(out);
 }
 int gxop_join_close(gx_op *op) { delete op; return 0; }
 

@@ -37,3 +37,50 @@ def test_c1_join_count():
                   [c for c in out], device=-1)
     rows = [t for c in agg for t in c.rows()]
     assert rows == [(n_probe,)]
+
+
+def test_buffered_probe_equals_per_chunk():
+    """probe_push/probe_flush (the LocalBufferExec-style buffered cadence)
+    must produce the same multiset as per-chunk gxop_join_probe."""
+    import numpy as np
+    from galaxysql_amd import abi
+    from galaxysql_amd.chunk import Block, Chunk, I64, multiset
+    from galaxysql_amd.operators import ParallelHashJoinExec, EquiJoinKey
+
+    lib = abi.load_oracle()
+    rng = np.random.default_rng(17)
+    bk = np.arange(2000, dtype=np.int64)
+    bv = bk * 3
+    pk = rng.integers(0, 4000, 10_000).astype(np.int64)
+    pv = np.arange(10_000, dtype=np.int64)
+
+    def run(buffered):
+        op = ParallelHashJoinExec(lib, abi.INNER, [EquiJoinKey(0, 0, I64)],
+                                  [I64, I64], [I64, I64], device=-1)
+        try:
+            op.consume_chunk(Chunk([Block(I64, values=bk),
+                                    Block(I64, values=bv)]))
+            op.build_consume()
+            rows = []
+            for lo in range(0, 10_000, 1000):
+                ch = Chunk([Block(I64, values=pk[lo:lo + 1000]),
+                            Block(I64, values=pv[lo:lo + 1000])])
+                if buffered:
+                    op.probe_push(ch)
+                    if lo == 4000:  # mid-stream flush
+                        r = op.probe_flush()
+                        if r:
+                            rows.extend(r.rows())
+                else:
+                    r = op.probe_chunk(ch)
+                    if r:
+                        rows.extend(r.rows())
+            if buffered:
+                r = op.probe_flush()
+                if r:
+                    rows.extend(r.rows())
+            return rows
+        finally:
+            op.close()
+
+    assert multiset(run(True)) == multiset(run(False))

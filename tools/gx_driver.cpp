@@ -38,6 +38,8 @@ struct GxApi {
     decltype(&gxop_join_build)    join_build;
     decltype(&gxop_join_probe)    join_probe;
     decltype(&gxop_join_tail)     join_tail;
+    decltype(&gxop_join_probe_push)  join_probe_push;
+    decltype(&gxop_join_probe_flush) join_probe_flush;
     decltype(&gxop_join_close)    join_close;
     decltype(&gxop_agg_create)    agg_create;
     decltype(&gxop_agg_consume)   agg_consume;
@@ -92,6 +94,8 @@ struct GxApi {
         BIND(join_build, gxop_join_build)
         BIND(join_probe, gxop_join_probe)
         BIND(join_tail, gxop_join_tail)
+        BIND(join_probe_push, gxop_join_probe_push)
+        BIND(join_probe_flush, gxop_join_probe_flush)
         BIND(join_close, gxop_join_close)
         BIND(agg_create, gxop_agg_create)
         BIND(agg_consume, gxop_agg_consume)
