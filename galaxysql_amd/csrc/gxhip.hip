@@ -346,6 +346,46 @@ __global__ void k_join_scatter(const int32_t *hashes, const uint8_t *keynull,
     }
 }
 
+/* ---- bloom pre-filter (fast path) --------------------------------------
+ * The reference tests a FastIntBloomFilter before the bucket walk
+ * (AbstractHashJoinExec.java:80-106, FastIntBloomFilter.java:30-61).
+ * Here: a cache-line-BLOCKED bloom — one 64-B line per key, two bits
+ * inside it — sized ~8 bits/build key so it stays Infinity-Cache-resident
+ * while the inline-bucket table does not: a negative probe then costs an
+ * L3 hit instead of a random HBM line. Bit placement is internal (the
+ * filter only gates the bucket read; false positives fall through), so
+ * it uses a 64-bit mix of the key rather than the Java rotate chain. */
+__device__ static inline uint64_t bloom_mix(int64_t key) {
+    uint64_t x = (uint64_t)key * 0x9E3779B97F4A7C15ull;
+    x ^= x >> 29; x *= 0xBF58476D1CE4E5B9ull; x ^= x >> 32;
+    return x;
+}
+
+/* line_mask = (n_lines - 1); lines of 16 u32 words (64 B) */
+__global__ void k_bloom_set(DevColView key0, int64_t n, uint32_t *bloom,
+                            uint32_t line_mask) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        if (col_is_null(key0, i)) continue;
+        uint64_t x = bloom_mix(((const int64_t *)key0.values)[i]);
+        uint32_t *line = bloom + ((uint32_t)x & line_mask) * 16;
+        uint32_t b1 = (uint32_t)(x >> 32) & 511u;
+        uint32_t b2 = (uint32_t)(x >> 41) & 511u;
+        atomicOr(&line[b1 >> 5], 1u << (b1 & 31));
+        atomicOr(&line[b2 >> 5], 1u << (b2 & 31));
+    }
+}
+
+__device__ static inline bool bloom_test(const uint32_t *bloom,
+                                         uint32_t line_mask, int64_t key) {
+    uint64_t x = bloom_mix(key);
+    const uint32_t *line = bloom + ((uint32_t)x & line_mask) * 16;
+    uint32_t b1 = (uint32_t)(x >> 32) & 511u;
+    uint32_t b2 = (uint32_t)(x >> 41) & 511u;
+    if (!(line[b1 >> 5] & (1u << (b1 & 31)))) return false;
+    return (line[b2 >> 5] & (1u << (b2 & 31))) != 0;
+}
+
 /* ---- inline-bucket build (plain joins) -------------------------------- */
 
 /* overflow run length per bucket (entries beyond the 4 inline slots) */
@@ -522,6 +562,8 @@ struct ProbeParams {
     uint32_t *build_matched;    /* bitmap (u32 words), build_outer only */
     int32_t n_conds;            /* residual condition terms (AND) */
     JoinCondDev conds[GX_MAX_CONDS];
+    const uint32_t *bloom;      /* blocked bloom pre-filter; NULL = off */
+    uint32_t bloom_mask;
 };
 
 __device__ static inline bool join_conds_pass(const ProbeParams &P,
@@ -599,21 +641,25 @@ __device__ static inline void probe_one(const ProbeParams &P, EmitStage &E,
         RadixRow r = active ? P.staged[bi] : RadixRow{0x80000000u, 0, 0};
         i = r.rowid & 0x7FFFFFFFu;
         if (active && !(r.rowid & 0x80000000u)) {
-            uint32_t b = (uint32_t)gx_mix(r.hash) & P.mask;
-            ebase = (uint64_t)b * 4;
-            e0 = P.table[ebase];
-            end = e0.pad;
             want_key = r.key;
+            if (!P.bloom || bloom_test(P.bloom, P.bloom_mask, want_key)) {
+                uint32_t b = (uint32_t)gx_mix(r.hash) & P.mask;
+                ebase = (uint64_t)b * 4;
+                e0 = P.table[ebase];
+                end = e0.pad;
+            }
         }
     } else if (P.fast_i64) {
         i = active ? (uint32_t)bi : 0;
         const DevColView &kc = P.probe_keys.col[0];
         if (active && !col_is_null(kc, i)) {
             want_key = ((const int64_t *)kc.values)[i];
-            uint32_t b = (uint32_t)gx_mix(gx_hash_i64(want_key)) & P.mask;
-            ebase = (uint64_t)b * 4;
-            e0 = P.table[ebase];
-            end = e0.pad;
+            if (!P.bloom || bloom_test(P.bloom, P.bloom_mask, want_key)) {
+                uint32_t b = (uint32_t)gx_mix(gx_hash_i64(want_key)) & P.mask;
+                ebase = (uint64_t)b * 4;
+                e0 = P.table[ebase];
+                end = e0.pad;
+            }
         }
     } else {
         i = active ? (uint32_t)bi : 0;
@@ -1092,6 +1138,8 @@ struct JoinOp : gx_op {
     DevBuf d_scan_tmp;
     DevBuf d_pidx, d_bpos, d_meta, d_ph, d_pn; /* reused across probe calls */
     DevBuf d_staged, d_radix_cnt;               /* radix-staged probe rows */
+    DevBuf d_bloom;            /* blocked bloom pre-filter (fast path) */
+    uint32_t bloom_mask = 0;
     uint32_t mask = 0;
     int64_t n_buckets = 0;
     bool fast_i64 = false;
@@ -1142,6 +1190,7 @@ struct JoinOp : gx_op {
         d_pidx.release(); d_bpos.release(); d_meta.release();
         d_ph.release(); d_pn.release();
         d_staged.release(); d_radix_cnt.release();
+        d_bloom.release();
         for (auto &b : d_cond_pats) b.release();
         if (ev0) (void)hipEventDestroy(ev0);
         if (ev1) (void)hipEventDestroy(ev1);
@@ -1319,6 +1368,25 @@ struct JoinOp : gx_op {
                                    (JoinEntry *)d_table.p,
                                    (JoinEntry *)d_entries.p,
                                    build.view(build_key_cols[0]), (int)fast_i64);
+                /* bloom pre-filter: GX_BLOOM=1 forces on, =0 off; default
+                 * on for builds whose table exceeds the Infinity Cache
+                 * while the bloom (8 bits/key) still fits (the regime
+                 * where a negative saves a random HBM line) */
+                const char *be = getenv("GX_BLOOM");
+                bool use_bloom = fast_i64 &&
+                    (be ? be[0] == '1'
+                        : (n >= (1 << 22) && n <= (200 << 20)));
+                if (use_bloom) {
+                    int64_t n_lines = gx_pow2(std::max<int64_t>(n / 64, 1));
+                    bloom_mask = (uint32_t)(n_lines - 1);
+                    if (d_bloom.grow((size_t)n_lines * 64, stream)) return -1;
+                    HIP_OK(hipMemsetAsync(d_bloom.p, 0, (size_t)n_lines * 64,
+                                          stream));
+                    hipLaunchKernelGGL(k_bloom_set, dim3(gx_grid(n)),
+                                       dim3(256), 0, stream,
+                                       build.view(build_key_cols[0]), n,
+                                       (uint32_t *)d_bloom.p, bloom_mask);
+                }
             }
         }
 
@@ -1570,6 +1638,8 @@ struct JoinOp : gx_op {
                 P.staged = nullptr;
                 P.part_starts = nullptr;
                 P.n_parts = 0;
+                P.bloom = (const uint32_t *)d_bloom.p;
+                P.bloom_mask = bloom_mask;
                 P.fast_i64 = (int)fast_i64;
                 P.build_keys = key_views(build, build_key_cols);
                 P.probe_keys = pk;
