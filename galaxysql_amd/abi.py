@@ -89,6 +89,7 @@ class GxJoinCfg(C.Structure):
         ("memory_budget_bytes", C.c_int64),
         ("n_conds", C.c_int32),
         ("conds", C.POINTER(GxJoinCond)),
+        ("enable_bloom", C.c_int32),
     ]
 
 

@@ -1375,7 +1375,8 @@ struct JoinOp : gx_op {
                 const char *be = getenv("GX_BLOOM");
                 bool use_bloom = fast_i64 &&
                     (be ? be[0] == '1'
-                        : (n >= (1 << 22) && n <= (200 << 20)));
+                        : (cfg.enable_bloom == 1 && n >= (1 << 16) &&
+                           n <= INT64_C(200) << 20));
                 if (use_bloom) {
                     int64_t n_lines = gx_pow2(std::max<int64_t>(n / 64, 1));
                     bloom_mask = (uint32_t)(n_lines - 1);

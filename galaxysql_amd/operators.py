@@ -74,7 +74,7 @@ class ParallelHashJoinExec:
     def __init__(self, lib, join_type, join_keys, outer_types, inner_types,
                  max_one_row=False, build_outer=False, anti_null_col=-1,
                  device=-1, stream=0, expected_build_rows=0, out_proj=None,
-                 memory_budget_bytes=0, conds=None):
+                 memory_budget_bytes=0, conds=None, enable_bloom=False):
         self._lib = lib
         self._keep = []
         keys = (GxEquiKey * len(join_keys))()
@@ -91,7 +91,8 @@ class ParallelHashJoinExec:
             anti_null_col=anti_null_col, device=device, stream=stream,
             expected_build_rows=expected_build_rows,
             n_out_proj=len(out_proj or []), out_proj=op_arr,
-            memory_budget_bytes=memory_budget_bytes)
+            memory_budget_bytes=memory_budget_bytes,
+            enable_bloom=int(enable_bloom))
         if conds:
             cfg.n_conds = len(conds)
             cfg.conds = _conds_to_ctypes(conds, self._keep)

@@ -110,7 +110,9 @@ def run_q3(lib, device, cust, orders, lineitem, expected_groups=0,
     j1 = ParallelHashJoinExec(
         lib, abi.SEMI, [EquiJoinKey(0, 0, I64)],
         outer_types=ORDERS_TYPES, inner_types=CUST_TYPES,
-        device=device, expected_build_rows=_src_rows(cust))
+        device=device, expected_build_rows=_src_rows(cust),
+        enable_bloom=True)  # selective probe (~20% hit): the planner's
+                            # ENABLE_HASH_TABLE_BLOOM_FILTER choice
     try:
         _consume_src(lib, j1, cust, CUST_TYPES)
         j1.build_consume()
@@ -126,7 +128,8 @@ def run_q3(lib, device, cust, orders, lineitem, expected_groups=0,
             lib, abi.INNER, [EquiJoinKey(0, 1, I64)],
             outer_types=LINEITEM_TYPES, inner_types=ORDERS_TYPES,
             device=device, expected_build_rows=n_orders_kept,
-            out_proj=[0, 1, 2, 5, 6])
+            out_proj=[0, 1, 2, 5, 6],
+            enable_bloom=True)  # ~49% hit: bloom won the r2 A/B (-8%/step)
         try:
             if reshuffle_by_orderkey:
                 # N>1: join1 ran custkey-sharded; its result re-shards by
@@ -515,7 +518,8 @@ def run_q9(lib, device, part_res, supplier, partsupp, orders, lineitem,
     ja = ParallelHashJoinExec(lib, abi.SEMI, [EquiJoinKey(0, 0, I64)],
                               outer_types=Q9_PARTSUPP_TYPES,
                               inner_types=[I64], device=device,
-                              expected_build_rows=max(1, n_part))
+                              expected_build_rows=max(1, n_part),
+                              enable_bloom=True)  # ~5% hit probe
     try:
         _consume_src(lib, ja, part_src, [I64])
         if not isinstance(part_src, list):

@@ -177,6 +177,13 @@ typedef struct gx_join_cfg {
     /* residual condition: AND of terms; n_conds = 0 -> pure equi-join */
     int32_t n_conds;
     const gx_join_cond *conds;
+    /* bloom pre-filter before the bucket read, mirroring the CN's
+     * ENABLE_HASH_TABLE_BLOOM_FILTER knob (ConnectionParams) /
+     * FastIntBloomFilter.java:30-61. Measured on MI355X: ~8% whole-step
+     * win on Q3 (49%-hit probe), 30% LOSS on a 100%-hit join (every
+     * probe pays an extra L3 line) — so the planner sets it where its
+     * selectivity stats say the probe is selective. 0 = off. */
+    int32_t enable_bloom;
 } gx_join_cfg;
 
 typedef struct gx_op gx_op;  /* opaque operator instance */
