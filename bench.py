@@ -865,7 +865,11 @@ def main():
         "data": "synthetic",
         "config": dict(wl.config(world),
                        agg_groups_per_s=(acc["groups"] / elapsed)
-                       if acc["groups"] else None),
+                       if acc["groups"] else None,
+                       probe_launches=acc["probe_launches"],
+                       probe_event_ms_per_launch=(
+                           acc["probe_kernel_ms"] / acc["probe_launches"])
+                       if acc["probe_launches"] else None),
         "roofline": {
             "bound": "hbm",
             "achieved": achieved / 1e9,
