@@ -221,3 +221,50 @@ def test_fuzz_groupjoin_window(seed):
     want = run_window(ora, [0], aggs, [I64, I64], chunks, reset=reset,
                       device=-1)
     assert rows_of(got) == rows_of(want), f"seed {seed} win"
+
+
+@pytest.mark.parametrize("seed", range(16))
+def test_fuzz_join_conditions(seed):
+    """Random residual (non-equi) conditions on top of random joins:
+    HIP vs oracle (checkJoinCondition semantics, gx_join_cond)."""
+    from galaxysql_amd.operators import JoinCond
+    rng = np.random.default_rng(3000 + seed)
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    key_types = [KEY_TYPES[rng.integers(0, len(KEY_TYPES))]]
+    jt = [abi.INNER, abi.LEFT, abi.SEMI, abi.ANTI][rng.integers(0, 4)]
+    null_frac = float(rng.choice([0.0, 0.2]))
+    card = int(rng.choice([10, 300]))
+    n_build = int(rng.integers(1, 3000))
+    n_probe = int(rng.integers(1, 6000))
+    bp = [PAYLOADS[rng.integers(0, 3)]]  # numeric payloads for conds
+    pp = [PAYLOADS[rng.integers(0, 3)]]
+    btypes, build = _side(rng, key_types, n_build, card, null_frac, bp)
+    ptypes, probe = _side(rng, key_types, n_probe, card, null_frac, pp)
+    keys = [EquiJoinKey(0, 0, key_types[0])]
+
+    # condition row = outer (probe side) cols then inner (build side) cols
+    row_types = ptypes + btypes
+    numeric = [i for i, t in enumerate(row_types) if t in (I64, I32, F64)]
+    conds = []
+    for _ in range(int(rng.integers(1, 3))):
+        a = int(rng.choice(numeric))
+        cmp = int(rng.choice([abi.LT, abi.LE, abi.GT, abi.GE, abi.EQ,
+                              abi.NE, abi.EQ_NULLSAFE, abi.NE_NULLSAFE]))
+        same_t = [i for i in numeric if row_types[i] == row_types[a]
+                  and i != a]
+        if same_t and rng.random() < 0.4:
+            conds.append(JoinCond(a, cmp, int(rng.choice(same_t))))
+        else:
+            v = float(rng.integers(-50, 400)) if row_types[a] == F64 \
+                else int(rng.integers(-50, 400))
+            if rng.random() < 0.1:
+                v = None  # SQL NULL constant
+            conds.append(JoinCond(a, cmp, -1, v))
+
+    kw = dict(conds=conds, enable_bloom=bool(rng.random() < 0.5))
+    got = run_join(hip, jt, keys, build, probe, ptypes, btypes,
+                   device=0, expected_build_rows=n_build, **kw)
+    want = run_join(ora, jt, keys, build, probe, ptypes, btypes,
+                    device=-1, **kw)
+    assert_rows_match(rows_of(got), rows_of(want), f"cond seed {seed}")
