@@ -66,7 +66,9 @@ def probe_bytes_per_row(n_build, matches, probe_rows):
     8*m pair, where s = max(1, m + expected bucket colliders) — e0 (count +
     first entry, one 16-B slot) is always read; matches and colliders come
     from the same 64-B line."""
-    n_buckets = next_pow2(max(2, n_build))  # 4-slot buckets, load <= 1
+    n_buckets = next_pow2(max(2, n_build))  # 4-slot buckets
+    if n_build > n_buckets * 3 // 4:
+        n_buckets <<= 1                     # load cap 0.75 (gxhip do_build)
     m = matches / probe_rows if probe_rows else 0.0
     s = max(1.0, m + n_build / n_buckets)
     return 4 + 1 + 8 + 16 * s + 8 * m

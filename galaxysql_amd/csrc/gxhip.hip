@@ -1322,8 +1322,13 @@ struct JoinOp : gx_op {
             } else {
                 /* inline-bucket build: hist -> overflow-run scan -> bucket
                  * metadata stamp -> scatter (key+pos only; pads carry the
-                 * metadata). */
-                n_buckets = gx_pow2(n); /* 4 slots/bucket, load <= 1 */
+                 * metadata). Load cap 0.75: at load ~0.9 the lazy e1..e3
+                 * slot loads frequently miss L2 (the wave's other random
+                 * fetches churn ~a full XCD L2 per iteration) and C2's
+                 * probe paid ~1.7 HBM lines/row (r2 PMC: 7.3 GB for a
+                 * 60M-row probe) — one extra doubling buys ~1 line/row. */
+                n_buckets = gx_pow2(n); /* 4 slots/bucket */
+                if (n > (n_buckets * 3) / 4) n_buckets <<= 1;
                 mask = (uint32_t)(n_buckets - 1);
                 if (d_counts.grow((size_t)(n_buckets + 1) * 4, stream) ||
                     d_starts.grow((size_t)(n_buckets + 1) * 4, stream) ||
