@@ -27,11 +27,14 @@ GPU operator results can enter the real CN's RemoteExchange buffers:
   accepts that for any decimal payload; simple-mode detection is a reader
   optimization, not a format change.
 
-All integers little-endian (airlift Slice byte order). Compression is NOT
-implemented: frames always carry marker 0, which every reference reader
-accepts (PagesSerde.deserialize takes the UNCOMPRESSED branch,
-PagesSerde.java:99-106); airlift LZ4 is an absent third-party dep
- and is a size optimization only.
+All integers little-endian (airlift Slice byte order). Compression:
+standard LZ4 block format, as airlift's Lz4Compressor/Lz4Decompressor
+(PagesSerde.java:66-118). The reader accepts reference-compressed frames
+(format-exact decompressor); the writer applies the reference's own
+keep-if-saves->=20% policy with a greedy spec-conformant compressor whose
+output any LZ4 decoder accepts — not byte-identical to airlift's match
+choices, so frame BYTE-identity holds for uncompressed frames (and
+value-identity always).
 """
 from __future__ import annotations
 
@@ -42,7 +45,118 @@ import numpy as np
 from .chunk import Block, Chunk, I64, I32, F64, SLICE, DECIMAL
 
 UNCOMPRESSED = 0
-COMPRESSED = 1
+COMPRESSED = 1  # ChunkCompression.java:24-25
+
+
+# ---- LZ4 block codec (standard block format; what airlift's
+# Lz4Compressor/Lz4Decompressor move through PagesSerde.java:66-118).
+# _lz4_compress MIRRORS gx_serde.inc's greedy matcher EXACTLY (same hash,
+# same single-probe table, same traversal) so python and native frames
+# stay byte-identical; _lz4_decompress is format-exact and accepts
+# reference-compressed frames.
+
+def _lz4_decompress(src, unc):
+    d = bytearray(unc)
+    si, di, sl = 0, 0, len(src)
+    while si < sl:
+        tok = src[si]
+        si += 1
+        lit = tok >> 4
+        if lit == 15:
+            while True:
+                b = src[si]
+                si += 1
+                lit += b
+                if b != 255:
+                    break
+        if si + lit > sl or di + lit > unc:
+            raise ValueError("corrupt LZ4 block (literals)")
+        d[di:di + lit] = src[si:si + lit]
+        si += lit
+        di += lit
+        if si >= sl:
+            break
+        off = src[si] | (src[si + 1] << 8)
+        si += 2
+        if off == 0 or off > di:
+            raise ValueError("corrupt LZ4 block (offset)")
+        ml = tok & 15
+        if ml == 15:
+            while True:
+                b = src[si]
+                si += 1
+                ml += b
+                if b != 255:
+                    break
+        ml += 4
+        if di + ml > unc:
+            raise ValueError("corrupt LZ4 block (match)")
+        for _ in range(ml):
+            d[di] = d[di - off]
+            di += 1
+    if di != unc:
+        raise ValueError("LZ4 block inflated to %d, expected %d" % (di, unc))
+    return bytes(d)
+
+
+_LZ4_HBITS = 13
+
+
+def _lz4_compress(s, cap):
+    # greedy matcher identical to gx_serde.inc lz4_compress; returns
+    # compressed bytes or None when they would not fit cap
+    import struct as _st
+    sl = len(s)
+    out = bytearray()
+
+    def put_lit_run(frm, to, ml4):
+        lit = to - frm
+        tok = (15 if lit >= 15 else lit) << 4
+        tok |= 15 if ml4 >= 15 else ml4
+        out.append(tok)
+        if lit >= 15:
+            v = lit - 15
+            while v >= 255:
+                out.append(255)
+                v -= 255
+            out.append(v)
+        out.extend(s[frm:to])
+
+    if sl >= 13:
+        tab = [-1] * (1 << _LZ4_HBITS)
+        si = anchor = 0
+        mflimit = sl - 12
+        matchlim = sl - 5
+        while si < mflimit:
+            (v,) = _st.unpack_from("<I", s, si)
+            h = ((v * 2654435761) & 0xFFFFFFFF) >> (32 - _LZ4_HBITS)
+            cand = tab[h]
+            tab[h] = si
+            if cand >= 0 and si - cand <= 65535 and \
+                    s[cand:cand + 4] == s[si:si + 4]:
+                ml = 4
+                while si + ml < matchlim and s[cand + ml] == s[si + ml]:
+                    ml += 1
+                put_lit_run(anchor, si, ml - 4)
+                off = si - cand
+                out.append(off & 0xFF)
+                out.append(off >> 8)
+                if ml - 4 >= 15:
+                    v2 = ml - 4 - 15
+                    while v2 >= 255:
+                        out.append(255)
+                        v2 -= 255
+                    out.append(v2)
+                si += ml
+                anchor = si
+                if len(out) > cap:
+                    return None
+            else:
+                si += 1
+        put_lit_run(anchor, sl, 0)
+    else:
+        put_lit_run(0, sl, 0)
+    return bytes(out) if len(out) <= cap else None
 
 _DEFAULT_CHARSET = b"UTF8MB4"
 _DEFAULT_COLLATION = b"UTF8MB4_GENERAL_CI"
@@ -143,6 +257,14 @@ def serialize_chunk(chunk: Chunk) -> bytes:
         else:
             raise ValueError(f"unsupported block type {b.type}")
     payload = b"".join(out)
+    # keep the LZ4 payload only when it saves >= 20%
+    # (PagesSerde.serializeForce:70-90, MINIMUM_COMPRESSION_RATIO 0.8)
+    comp = _lz4_compress(payload, len(payload) - 1) \
+        if len(payload) >= 32 else None
+    if comp is not None and len(comp) / len(payload) <= 0.8:
+        frame = struct.pack("<ibii", chunk.n_rows, COMPRESSED,
+                            len(payload), len(comp))
+        return frame + comp
     frame = struct.pack("<ibii", chunk.n_rows, UNCOMPRESSED,
                         len(payload), len(payload))
     return frame + payload
@@ -210,8 +332,14 @@ def deserialize_chunk(buf: bytes, types, pos: int = 0):
     reference's type-list-driven BlockEncodingBuilders.create."""
     n_rows, marker, _unc, size = struct.unpack_from("<ibii", buf, pos)
     pos += 13
+    if marker == COMPRESSED:
+        raw = _lz4_decompress(bytes(buf[pos:pos + size]), _unc)
+        inner, _ = deserialize_chunk(
+            struct.pack("<ibii", n_rows, UNCOMPRESSED, _unc, _unc) + raw,
+            types, 0)
+        return inner, pos + size
     if marker != UNCOMPRESSED:
-        raise ValueError("compressed frames not supported (see module doc)")
+        raise ValueError("unknown compression marker %d" % marker)
     end = pos + size
     (n_blocks,) = struct.unpack_from("<i", buf, pos)
     pos += 4
