@@ -17,8 +17,8 @@ GOLDEN = os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden",
                       "window_vectors.json")
 
 AGG_BY_NAME = {"count_row": abi.COUNT_ROW, "count_col": abi.COUNT_COL,
-               "sum_i64": abi.SUM_I64, "min_i64": abi.MIN_I64,
-               "max_i64": abi.MAX_I64}
+               "sum_i64": abi.SUM_I64, "sum_i64n": abi.SUM_I64N,
+               "min_i64": abi.MIN_I64, "max_i64": abi.MAX_I64}
 
 
 def _load():
@@ -60,3 +60,18 @@ def test_window_golden_oracle():
 @pytest.mark.gpu
 def test_window_golden_hip():
     _run(abi.load_hip(), 0)
+
+
+def test_running_sum_nullinit_over_nullable_col():
+    """Derived from the Sum (null-init) semantics the golden case pins:
+    running SUM_I64N over the NULLABLE column stays NULL until the first
+    non-null input (Sum.java null init), where Sum0 emits 0."""
+    chunks = _input_chunks(_load()["_input"])
+    lib = abi.load_oracle()
+    out = run_window(lib, [0], [(abi.SUM_I64N, 2), (abi.SUM_I64, 2)],
+                     [I32, I32, I32], chunks, reset=[False, False])
+    rows = [r for c in out for r in c.rows()]
+    got_n = [r[3] for r in rows]
+    got_0 = [r[4] for r in rows]
+    assert got_n == [None, 2, 1, 3, 3, 5, 1]
+    assert got_0 == [0, 2, 1, 3, 3, 5, 1]
