@@ -314,6 +314,64 @@ static void t_concurrent(int device) {
     std::printf("  join concurrent (4 threads, shared op): ok\n");
 }
 
+/* INNER join with a residual condition (gx_join_cond): build keys 0..B-1
+ * with payload key*10; probe keys 0..2B-1; condition = inner payload < T.
+ * Expected matches = probe keys k < B with k*10 < T, i.e. k < T/10. */
+static void t_join_condition(int device) {
+    const int B = 4000, P = 16000;
+    const int64_t T = 17770; /* threshold: keys < 1777 pass */
+    std::vector<int64_t> bk(B), bv(B), pk(P), pv(P);
+    for (int i = 0; i < B; i++) { bk[i] = i; bv[i] = (int64_t)i * 10; }
+    int64_t expect = 0;
+    for (int i = 0; i < P; i++) {
+        pk[i] = (int64_t)(rnd() % (2 * B));
+        pv[i] = i;
+        if (pk[i] < B && pk[i] * 10 < T) expect++;
+    }
+    gx_equi_key key{0, 0, GX_I64, 0};
+    int32_t t2[2] = {GX_I64, GX_I64};
+    gx_join_cond cond{};
+    cond.col_a = 3;          /* condition row: outer(2) then inner(2) */
+    cond.cmp = GX_CMP_LT;
+    cond.col_b = -1;
+    cond.v_i64 = T;
+    gx_join_cfg cfg{};
+    cfg.join_type = GX_JOIN_INNER;
+    cfg.n_keys = 1; cfg.keys = &key;
+    cfg.n_outer_cols = 2; cfg.outer_types = t2;
+    cfg.n_inner_cols = 2; cfg.inner_types = t2;
+    cfg.anti_null_col = -1;
+    cfg.device = device;
+    cfg.n_conds = 1; cfg.conds = &cond;
+    gx_op *op = api.join_create(&cfg);
+    CHECK(op, "cond join_create");
+    if (!op) return;
+    gx_block bb[2] = {mk_i64(bk.data()), mk_i64(bv.data())};
+    gx_chunk bc{B, 2, bb};
+    CHECK(api.join_consume(op, &bc) == 0, "cond consume");
+    CHECK(api.join_build(op) == 0, "cond build");
+    gx_block pb[2] = {mk_i64(pk.data()), mk_i64(pv.data())};
+    gx_chunk pc{P, 2, pb};
+    gx_result *res = nullptr;
+    CHECK(api.join_probe(op, &pc, &res) == 0, "cond probe");
+    int64_t got = 0;
+    if (res) {
+        CHECK(api.result_to_host(res) == 0, "cond to_host");
+        got = res->chunk.n_rows;
+        for (int r = 0; r < res->chunk.n_rows; r++)
+            if (cell_i64(res, 3, r) >= T) {
+                CHECK(false, "condition leaked row %d", r);
+                break;
+            }
+        api.result_release(res);
+    }
+    CHECK(got == expect, "cond matches %lld != %lld", (long long)got,
+          (long long)expect);
+    api.join_close(op);
+    std::printf("  join residual condition: %lld matches ok\n",
+                (long long)got);
+}
+
 /* LEFT join: probe keys half-missing -> unmatched probe rows carry NULL
  * inner columns. */
 static void t_join_left(int device) {
@@ -931,6 +989,7 @@ int main(int argc, char **argv) {
 
     if (cmd == "selftest") {
         t_join_inner(device);
+        t_join_condition(device);
         t_concurrent(device);
         t_join_left(device);
         t_agg(device);
