@@ -527,7 +527,8 @@ class C4:
             cust = shuffle_columns(lib, cust, Q18_CUST_TYPES, [0],
                                    device=local_rank)
         _, info = run_q18(lib, local_rank, cust, orders, lineitem,
-                          expected_groups=self.expected_groups)
+                          expected_groups=self.expected_groups,
+                          reshuffle_by_custkey=distributed)
         self.last_info = info
         ast = info["agg_stats"]
         return {"probe_kernel_ms": ast["kernel_ms"], "probe_launches":

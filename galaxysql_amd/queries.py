@@ -298,13 +298,12 @@ Q18_CUST_TYPES = [I64, I64]              # c_custkey, c_payload
 
 
 def run_q18(lib, device, cust, orders, lineitem, having=300,
-            expected_groups=0):
+            expected_groups=0, reshuffle_by_custkey=False):
     """Q18: lineitem GROUP BY l_orderkey SUM(l_quantity)
     (the dominant operator: ~150M groups at SF100), HAVING sum > 300
-    (~hundreds of survivors), then the tiny survivors join orders and
-    customer. The HAVING filter runs in torch on the device result — the
-    reference's FilterExec is host glue around the operators (SURVEY.md §2
-    out-of-scope rows), the operator work is the aggregate + joins.
+    (~hundreds of survivors) via the library's own scan (GT predicate on
+    each emitted device batch — the reference's FilterExec above the agg),
+    then the tiny survivors join orders and customer.
 
     Returns (n_final_rows, info)."""
     import os
@@ -391,6 +390,17 @@ def run_q18(lib, device, cust, orders, lineitem, having=300,
         ja.close()
 
     # ⋈ customer on custkey (build = the tiny result, probe = customer)
+    if reshuffle_by_custkey and cols1 is not None:
+        # N>1: the survivors⋈orders output is orderkey-sharded but the
+        # customer table is custkey-sharded — re-shard the tiny result by
+        # o_custkey (col 1) so the final join is colocated (the
+        # reference's FIXED shuffle between plan fragments, SURVEY §8e;
+        # without this, cross-rank customer matches are LOST — caught by
+        # tests/test_q18_distributed.py).
+        from .exchange import shuffle_columns
+        cols1 = shuffle_columns(lib, cols1, Q18_ORDERS_TYPES + [I64, I64],
+                                [1], device=device)
+        n_r1 = cols1[0].numel()
     jb = ParallelHashJoinExec(lib, abi.INNER, [EquiJoinKey(0, 1, I64)],
                               outer_types=Q18_CUST_TYPES,
                               inner_types=Q18_ORDERS_TYPES + [I64, I64],
