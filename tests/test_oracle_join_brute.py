@@ -1,0 +1,178 @@
+"""Independent brute-force cross-check of the ORACLE's join semantics.
+
+The oracle is the parity anchor for the GPU path, pinned by transcribed
+golden vectors at fixed points; this fuzz checks it against a third,
+dictionary-based restatement of the reference's probe loop
+(AbstractBufferedJoinExec.nextRows:185-266 + AbstractJoinExec output
+schemas :102-227 + checkJoinCondition:227-250) on random inputs — all
+join types, null keys (build skip / probe never-match), buildOuter tails,
+ANTI NOT-IN null rules, residual conditions, single-join errors.
+"""
+import numpy as np
+import pytest
+
+from galaxysql_amd import abi
+from galaxysql_amd.chunk import Block, Chunk, I64, multiset, rows_of
+from galaxysql_amd.operators import EquiJoinKey, JoinCond, run_join
+
+
+def _col(rng, n, card, null_frac):
+    vals = rng.integers(0, card, n).astype(np.int64)
+    nulls = (rng.random(n) < null_frac).astype(np.uint8)
+    return vals, nulls
+
+
+def _brute(jt, bk, bn, bv, bvn, pk, pn, pv, pvn, build_outer, anti_null,
+           cond):
+    """cond: None or (side, col_is_payload?, cmp, const) simplified to a
+    predicate over (probe_payload, build_payload) pairs."""
+    # table: key -> list of build positions (NULL keys never inserted,
+    # ExecUtils.buildOneChunk:933-941)
+    tab = {}
+    for i in range(len(bk)):
+        if bn[i]:
+            continue
+        tab.setdefault(int(bk[i]), []).append(i)
+    rows = []
+    matched_build = set()
+    for r in range(len(pk)):
+        matches = []
+        if not pn[r]:
+            for m in tab.get(int(pk[r]), []):
+                if cond is not None and not cond(r, m):
+                    continue  # failing candidates never count as matches
+                matches.append(m)
+        if jt in (abi.INNER, abi.LEFT):
+            for m in matches:
+                rows.append(_row(jt, r, m, pk, pn, pv, pvn, bk, bn, bv, bvn,
+                                 build_outer))
+                matched_build.add(m)
+            if jt == abi.LEFT and not matches and not build_outer:
+                rows.append(_row(jt, r, None, pk, pn, pv, pvn, bk, bn, bv,
+                                 bvn, build_outer))
+            if build_outer:
+                pass  # null rows drain via the tail
+        elif jt == abi.RIGHT:
+            for m in matches:
+                rows.append(_row(jt, r, m, pk, pn, pv, pvn, bk, bn, bv, bvn,
+                                 build_outer))
+                matched_build.add(m)
+            if not matches:
+                rows.append(_row(jt, r, None, pk, pn, pv, pvn, bk, bn, bv,
+                                 bvn, build_outer))
+        elif jt == abi.SEMI:
+            if matches:
+                rows.append((_v(pk, pn, r), _v(pv, pvn, r)))
+        else:  # ANTI
+            if not matches:
+                if anti_null and pn[r]:
+                    continue  # NOT-IN: NULL operand suppresses the row
+                rows.append((_v(pk, pn, r), _v(pv, pvn, r)))
+    tail = []
+    if build_outer:
+        for m in range(len(bk)):
+            if m not in matched_build:
+                # outer (build) cols then nulls (nextJoinNullRows:372-401)
+                tail.append((_v(bk, bn, m), _v(bv, bvn, m), None, None))
+    return rows, tail
+
+
+def _v(vals, nulls, i):
+    return None if nulls[i] else int(vals[i])
+
+
+def _row(jt, r, m, pk, pn, pv, pvn, bk, bn, bv, bvn, build_outer):
+    probe = (_v(pk, pn, r), _v(pv, pvn, r))
+    build = (None, None) if m is None else (_v(bk, bn, m), _v(bv, bvn, m))
+    if build_outer:
+        # build side is the OUTER side: outer cols then inner (probe) cols
+        return build + probe
+    if jt == abi.RIGHT:
+        return build + probe  # inner (build) first for RIGHT
+    return probe + build
+
+
+@pytest.mark.parametrize("seed", range(30))
+def test_oracle_join_vs_brute(seed):
+    rng = np.random.default_rng(7000 + seed)
+    lib = abi.load_oracle()
+    jt = [abi.INNER, abi.LEFT, abi.RIGHT, abi.SEMI, abi.ANTI][
+        rng.integers(0, 5)]
+    n_build = int(rng.integers(0, 400))
+    n_probe = int(rng.integers(1, 900))
+    card = int(rng.choice([3, 40, 1000]))
+    nf = float(rng.choice([0.0, 0.15, 0.5]))
+    bk, bn = _col(rng, n_build, card, nf)
+    bv, bvn = _col(rng, n_build, 50, nf / 2)
+    pk, pn = _col(rng, n_probe, card, nf)
+    pv, pvn = _col(rng, n_probe, 50, nf / 2)
+
+    kw = {}
+    anti_null = False
+    build_outer = False
+    cond = None
+    conds_kw = None
+    if jt == abi.ANTI and rng.random() < 0.5:
+        kw["anti_null_col"] = 0
+        anti_null = True
+        # ANTI NOT-IN with a build NULL -> empty result handled by the
+        # oracle's doSpecialCheckForSemiJoin; mirror by skipping such seeds
+        if n_build and bn.any():
+            bn[:] = 0
+    if jt in (abi.INNER, abi.LEFT) and rng.random() < 0.3:
+        build_outer = True
+        kw["build_outer"] = True
+    if rng.random() < 0.5:
+        # residual condition over the payloads: probe_payload < build_payload
+        # or payload vs constant (SQL semantics: NULL fails)
+        if rng.random() < 0.5:
+            t = int(rng.integers(0, 50))
+            # condition row: outer cols then inner cols
+            if build_outer:
+                conds_kw = [JoinCond(3, abi.LT, -1, t)]   # inner=probe pv
+            elif jt == abi.RIGHT:
+                conds_kw = [JoinCond(1, abi.LT, -1, t)]   # build pv first
+            else:
+                conds_kw = [JoinCond(3, abi.LT, -1, t)]   # build pv last
+            # the chosen condition column per layout:
+            #  non-RIGHT non-BO: row = probe(2) + build(2) -> col3 = bv
+            #  RIGHT:            row = build(2) + probe(2) -> col1 = bv
+            #  build_outer:      row = build(2) + probe(2) -> col3 = PROBE pv
+            if build_outer:
+                cond = (lambda r, m: (not pvn[r]) and int(pv[r]) < t)
+            else:
+                cond = (lambda r, m: (not bvn[m]) and int(bv[m]) < t)
+        else:
+            # col-vs-col: probe payload < build payload (both non-null)
+            if build_outer:
+                conds_kw = [JoinCond(1, abi.GT, 3)]  # build pv > probe pv
+                cond = (lambda r, m: (not bvn[m]) and (not pvn[r])
+                        and int(bv[m]) > int(pv[r]))
+            elif jt == abi.RIGHT:
+                conds_kw = [JoinCond(3, abi.LT, 1)]  # probe pv < build pv
+                cond = (lambda r, m: (not bvn[m]) and (not pvn[r])
+                        and int(pv[r]) < int(bv[m]))
+            else:
+                conds_kw = [JoinCond(1, abi.LT, 3)]  # probe pv < build pv
+                cond = (lambda r, m: (not bvn[m]) and (not pvn[r])
+                        and int(pv[r]) < int(bv[m]))
+    if conds_kw:
+        kw["conds"] = conds_kw
+
+    def chunks(k, kn, v, vn):
+        return [Chunk([Block(I64, values=k, nulls=kn if kn.any() else None),
+                       Block(I64, values=v, nulls=vn if vn.any() else None)])]
+
+    build_chunks = chunks(bk, bn, bv, bvn)
+    probe_chunks = chunks(pk, pn, pv, pvn)
+    if build_outer:
+        # outer types describe the BUILD side
+        out = run_join(lib, jt, [EquiJoinKey(0, 0, I64)], build_chunks,
+                       probe_chunks, [I64, I64], [I64, I64], **kw)
+    else:
+        out = run_join(lib, jt, [EquiJoinKey(0, 0, I64)], build_chunks,
+                       probe_chunks, [I64, I64], [I64, I64], **kw)
+    got = rows_of(out)
+    want, tail = _brute(jt, bk, bn, bv, bvn, pk, pn, pv, pvn, build_outer,
+                        anti_null, cond)
+    assert multiset(got) == multiset(want + tail), f"seed {seed} jt {jt}"
