@@ -344,3 +344,48 @@ def test_agg_sum_i64n_null_group_oracle():
 def test_agg_sum_i64n_null_group_gpu():
     rows = _sum_i64n_agg(abi.load_hip(), 0)
     assert rows == [(1, 12, 12), (2, None, 0), (3, None, 0)]
+
+
+@pytest.mark.gpu
+def test_range_frames_gpu_at_scale():
+    """500K rows, ~200 partitions: RANGE sliding SUM + MIN (dynamic
+    sparse-table levels from the measured max span) vs the oracle."""
+    rng = np.random.default_rng(77)
+    n = 500_000
+    parts = np.sort(rng.integers(0, 200, n)).astype(np.int64)
+    order = np.empty(n, dtype=np.int64)
+    i = 0
+    while i < n:
+        j = i
+        while j < n and parts[j] == parts[i]:
+            j += 1
+        order[i:j] = np.sort(rng.integers(0, 10_000, j - i))
+        i = j
+    vals = rng.integers(-1000, 1000, n).astype(np.int64)
+    vnulls = (rng.random(n) < 0.05).astype(np.uint8)
+    chunk = Chunk([Block(I64, values=parts), Block(I64, values=order),
+                   Block(I64, values=vals, nulls=vnulls)])
+    frames = [(abi.SUM_I64N, 2, abi.FRAME_RANGE_SLIDING, 50, 50, 1, 1),
+              (abi.MIN_I64, 2, abi.FRAME_RANGE_SLIDING, 100, 0, 1, 1),
+              (abi.SUM_I64, 2, abi.FRAME_RANGE_UNBOUNDED_PRECEDING,
+               0, 25, 1, 1)]
+
+    def run(lib, device):
+        op = OverWindowFramesExec(lib, part_cols=[0], frames=frames,
+                                  input_types=[I64, I64, I64], device=device)
+        try:
+            op.consume_chunk(chunk)
+            op.finish()
+            out = op.result_chunks()
+        finally:
+            op.close()
+        rows = []
+        for c in out:
+            rows.extend(c.rows())
+        return rows
+
+    hip = run(abi.load_hip(), 0)
+    ora = run(abi.load_oracle(), -1)
+    assert len(hip) == len(ora)
+    for i, (h, o) in enumerate(zip(hip, ora)):
+        assert h[3:] == o[3:], (i, h, o)
