@@ -1,0 +1,68 @@
+"""Independent brute-force cross-check of the ORACLE's fused group-join
+(HashGroupJoinExec restatement): one group PER BUILD ROW POSITION,
+null-safe key matching (Chunk.equals — unlike the plain join's null-skip,
+HashGroupJoinExec.buildOneChunk:296-311), INNER emits matched groups,
+LEFT emits every group with one null-row accumulation for unmatched
+(COUNT(*)=1, COUNT(col)=0, SUM0=0, null-init aggs stay NULL)."""
+import numpy as np
+import pytest
+
+from galaxysql_amd import abi
+from galaxysql_amd.chunk import Block, Chunk, I64, multiset, rows_of
+from galaxysql_amd.operators import EquiJoinKey, run_groupjoin
+
+
+def _v(vals, nulls, i):
+    return None if nulls[i] else int(vals[i])
+
+
+@pytest.mark.parametrize("seed", range(20))
+def test_oracle_groupjoin_vs_brute(seed):
+    rng = np.random.default_rng(9000 + seed)
+    from galaxysql_amd.abi import load_oracle
+    lib = load_oracle()
+    jt = abi.INNER if rng.random() < 0.5 else abi.LEFT
+    n_build = int(rng.integers(1, 300))
+    n_probe = int(rng.integers(0, 900))
+    card = int(rng.choice([3, 30]))
+    nf = float(rng.choice([0.0, 0.25]))
+    bk = rng.integers(0, card, n_build).astype(np.int64)
+    bn = (rng.random(n_build) < nf).astype(np.uint8)
+    bg = rng.integers(0, 1000, n_build).astype(np.int64)  # group payload
+    pk = rng.integers(0, card, n_probe).astype(np.int64)
+    pn = (rng.random(n_probe) < nf).astype(np.uint8)
+    pv = rng.integers(-50, 50, n_probe).astype(np.int64)
+    pvn = (rng.random(n_probe) < 0.15).astype(np.uint8)
+
+    build = [Chunk([Block(I64, values=bk, nulls=bn if bn.any() else None),
+                    Block(I64, values=bg)])]
+    probe = [Chunk([Block(I64, values=pk, nulls=pn if pn.any() else None),
+                    Block(I64, values=pv,
+                          nulls=pvn if pvn.any() else None)])]
+    out = run_groupjoin(lib, jt, [EquiJoinKey(0, 0, I64)], build, probe,
+                        [I64, I64], [I64, I64], group_cols=[0, 1],
+                        aggs=[(abi.COUNT_ROW, -1), (abi.COUNT_COL, 1),
+                              (abi.SUM_I64, 1)])
+    got = rows_of(out)
+
+    # brute: group = build position; null-safe key equality
+    want = []
+    for m in range(n_build):
+        cnt = ccnt = ssum = 0
+        matched = False
+        for r in range(n_probe):
+            eq = (bn[m] and pn[r]) or \
+                 (not bn[m] and not pn[r] and bk[m] == pk[r])
+            if not eq:
+                continue
+            matched = True
+            cnt += 1
+            if not pvn[r]:
+                ccnt += 1
+                ssum += int(pv[r])
+        if jt == abi.INNER and not matched:
+            continue
+        if jt == abi.LEFT and not matched:
+            cnt = 1  # buildNullRow: one all-null probe row accumulates
+        want.append((_v(bk, bn, m), int(bg[m]), cnt, ccnt, ssum))
+    assert multiset(got) == multiset(want), f"seed {seed} jt {jt}"
