@@ -112,3 +112,48 @@ def test_hybrid_single_join_error():
         run_join(hip, abi.INNER, keys, build, probe, [I64], [I64, I64],
                  device=0, max_one_row=True, expected_build_rows=10**6,
                  memory_budget_bytes=TINY)
+
+
+def test_hybrid_buffered_probe_matches_plain():
+    """gxop_join_probe_push on the HYBRID variant spills like probe;
+    flush is a no-op and results drain via tail — same multiset as the
+    plain per-chunk probe path."""
+    import numpy as np
+    from galaxysql_amd import abi
+    from galaxysql_amd.chunk import Block, Chunk, I64, multiset, rows_of
+    from galaxysql_amd.operators import (ParallelHashJoinExec, EquiJoinKey)
+    lib = abi.load_hip()
+    rng = np.random.default_rng(42)
+    bk = rng.integers(0, 5000, 20000).astype(np.int64)
+    bv = rng.integers(0, 100, 20000).astype(np.int64)
+    pk = rng.integers(0, 10000, 50000).astype(np.int64)
+    pv = np.arange(50000, dtype=np.int64)
+
+    def run(buffered):
+        op = ParallelHashJoinExec(
+            lib, abi.INNER, [EquiJoinKey(0, 0, I64)], [I64, I64], [I64, I64],
+            device=0, expected_build_rows=20000,
+            memory_budget_bytes=65536)  # forces the hybrid variant
+        try:
+            op.consume_chunk(Chunk([Block(I64, values=bk),
+                                    Block(I64, values=bv)]))
+            op.build_consume()
+            rows = []
+            for lo in range(0, 50000, 10000):
+                ch = Chunk([Block(I64, values=pk[lo:lo + 10000]),
+                            Block(I64, values=pv[lo:lo + 10000])])
+                if buffered:
+                    op.probe_push(ch)
+                else:
+                    r = op.probe_chunk(ch)
+                    if r:
+                        rows.extend(r.rows())
+            if buffered:
+                r = op.probe_flush()
+                assert r is None  # hybrid defers everything to the tail
+            rows.extend(r2 for c in op.tail_chunks() for r2 in c.rows())
+            return rows
+        finally:
+            op.close()
+
+    assert multiset(run(True)) == multiset(run(False))

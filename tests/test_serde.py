@@ -143,7 +143,9 @@ def test_serde_random_roundtrip_fuzz():
                     for i in range(n)]))
             else:
                 dt = {I64: np.int64, I32: np.int32, F64: np.float64}[t]
-                v = rng.integers(-1000, 1000, n).astype(dt)
+                # low-cardinality half the time: exercises the LZ4 path
+                hi = 4 if rng.random() < 0.5 else 1000
+                v = rng.integers(-hi, hi, n).astype(dt)
                 blocks.append(Block(t, values=v,
                                     nulls=nulls.astype(np.uint8)
                                     if nulls.any() else None))
