@@ -961,6 +961,53 @@ static void bench_join(int device, int64_t build_rows, int64_t probe_rows,
                 st.probe_kernel_ms, (long long)matches);
 }
 
+/* agg bench (the C4 shape, no Python): GROUP BY key SUM(val) over rows
+ * with rows/groups duplication factor; each step = create+consume+build
+ * so the whole fused insert is timed */
+static void bench_agg(int device, int64_t rows, int64_t groups, int steps) {
+    std::vector<int64_t> k(rows), v(rows);
+    for (int64_t i = 0; i < rows; i++) {
+        k[i] = (int64_t)(rnd() % (uint64_t)groups);
+        v[i] = (int64_t)(rnd() % 100);
+    }
+    int32_t t2[2] = {GX_I64, GX_I64};
+    int32_t gcols[1] = {0};
+    gx_agg_spec sp[1] = {{GX_AGG_SUM_I64, 1}};
+    double total_s = 0;
+    int64_t got_groups = 0;
+    for (int s = -1; s < steps; s++) { /* s = -1: warmup */
+        gx_agg_cfg cfg{};
+        cfg.n_group_cols = 1; cfg.group_cols = gcols;
+        cfg.n_aggs = 1; cfg.aggs = sp;
+        cfg.n_input_cols = 2; cfg.input_types = t2;
+        cfg.expected_groups = groups;
+        cfg.device = device;
+        auto t0 = std::chrono::steady_clock::now();
+        gx_op *op = api.agg_create(&cfg);
+        if (!op) { std::fprintf(stderr, "agg: %s\n", api.last_error()); return; }
+        gx_block b[2] = {mk_i64(k.data()), mk_i64(v.data())};
+        gx_chunk c{(int32_t)rows, 2, b};
+        api.agg_consume(op, &c);
+        api.agg_build(op);
+        got_groups = 0;
+        gx_result *res = nullptr;
+        while (api.agg_next(op, &res) == 0 && res) {
+            got_groups += res->chunk.n_rows;
+            api.result_release(res);
+            res = nullptr;
+        }
+        api.agg_close(op);
+        auto t1 = std::chrono::steady_clock::now();
+        if (s >= 0)
+            total_s += std::chrono::duration<double>(t1 - t0).count();
+    }
+    std::printf("{\"driver\": \"gx_driver\", \"op\": \"hash_agg_sum\", "
+                "\"rows\": %lld, \"groups\": %lld, \"steps\": %d, "
+                "\"wall_s\": %.4f, \"agg_rows_per_s\": %.3e}\n",
+                (long long)rows, (long long)got_groups, steps, total_s,
+                (double)rows * steps / total_s);
+}
+
 int main(int argc, char **argv) {
     const char *lib = nullptr;
     int device = 0;
@@ -981,6 +1028,8 @@ int main(int argc, char **argv) {
             "usage: gx_driver --lib <gxop .so> [--device N] selftest\n"
             "       gx_driver --lib <gxop .so> [--device N] bench "
             "[--build-rows B] [--probe-rows P] [--steps K]\n"
+            "       gx_driver --lib <gxop .so> [--device N] bench-agg "
+            "[--build-rows GROUPS] [--probe-rows ROWS] [--steps K]\n"
             "(--device -1 selects the CPU oracle build, test use only)\n");
         return 2;
     }
@@ -1009,6 +1058,10 @@ int main(int argc, char **argv) {
     }
     if (cmd == "bench") {
         bench_join(device, build_rows, probe_rows, steps);
+        return 0;
+    }
+    if (cmd == "bench-agg") {
+        bench_agg(device, probe_rows, build_rows, steps);
         return 0;
     }
     std::fprintf(stderr, "unknown command %s\n", cmd.c_str());
