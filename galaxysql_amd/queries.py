@@ -9,11 +9,11 @@ no output columns in Q3), the filtered orders build the second join,
 lineitem probes it, and a 3-key hash aggregate sums revenue per
 (l_orderkey, o_orderdate, o_shippriority).
 
-Inputs arrive PRE-FILTERED and revenue PRE-PROJECTED (extendedprice *
-(1-discount)) — the vectorized filter/project stage is §8(f) row 1, not yet
-built; DESIGN.md states this caveat. Revenue is carried twice: DOUBLE
-(rel-tol 1e-9) and scaled-int64 cents (DECIMAL(15,2)-sum semantics,
-bit-exact).
+run_q3 takes PRE-FILTERED inputs with revenue PRE-PROJECTED; the HONEST
+benched path is run_q3_honest, which runs the device-side vectorized
+filter/projection scans (gxop_scan, §8(f) row 1) over UNFILTERED tables
+inside the timed step. Revenue is carried twice: DOUBLE (rel-tol 1e-9)
+and scaled-int64 cents (DECIMAL(15,2)-sum semantics, bit-exact).
 """
 from __future__ import annotations
 
