@@ -6,13 +6,15 @@ synthetic workloads, whole-job over all ranks, %HBM roofline for the
 dominant kernel.
 
 Workloads (BASELINE.md):
-  c3 (default) — SF100 Q3: customer SEMI orders INNER lineitem + 3-key
-       GROUP BY SUM(revenue) as DOUBLE and as scaled-int64 cents
-       (DECIMAL(15,2)-sum semantics). The config the metric is quoted on
-       ("TPC-H SF100 Q3 ... 1xMI355X"); fits one GPU. Inputs pre-filtered /
-       revenue pre-projected (vectorized filter stage is SURVEY §8f row 1 —
-       DESIGN.md states the caveat).
+  c3 (default) — HONEST SF100 Q3: UNFILTERED customer/orders/lineitem in
+       HBM; the timed step runs the device-side vectorized filter +
+       projection scans, the SEMI+INNER join chain and the 3-key GROUP BY
+       SUM(revenue) as DOUBLE and as scaled-int64 cents (DECIMAL(15,2)-sum
+       semantics). The config the metric is quoted on ("TPC-H SF100 Q3 ...
+       1xMI355X"); fits one GPU.
   c2 — SF10 orders⋈lineitem full materialization (configs[1]).
+  c2chunk — the CN's CHUNK_SIZE push cadence (per-chunk and buffered).
+  c4 / c5 — SF100 Q18 and the per-GPU SF300/8 Q9 shard.
 
 One "step" = one complete operator pass over the batch through the gxop
 C-ABI (HIP/gfx950), inputs already resident in HBM. N>1: weak scaling with
@@ -22,8 +24,9 @@ keys are orderkey-led, so rows group-colocate after the orderkey shuffle
 and the local aggregate is final (the reference optimizer's own colocation
 reasoning).
 
-CPU baseline (rank 0, N=1 only): the C++ oracle ("port"), bounded sample,
-single thread.
+CPU baseline (rank 0, N=1 only): the C++ oracle ("port") on ALL host
+cores, one thread per core (BASELINE.md protocol), bounded sample, core
+count stated in the JSON.
 """
 import argparse
 import json
