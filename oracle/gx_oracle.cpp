@@ -46,6 +46,25 @@ static inline int32_t jadd(int32_t a, int32_t b) {
     return (int32_t)((uint32_t)a + (uint32_t)b);
 }
 
+
+/* Java Math.min/max semantics for doubles (NaN propagates, -0.0 < +0.0) —
+ * what Double2DoubleMin/Max accumulate with (Math.min,
+ * Double2DoubleMin.java:40-44); std::min/max differ on NaN and -0.0. */
+static inline double jmin_f64o(double a, double b) {
+    if (a != a) return a;
+    if (b != b) return b;
+    if (a == 0.0 && b == 0.0)
+        return std::signbit(a) ? a : b;
+    return a < b ? a : b;
+}
+static inline double jmax_f64o(double a, double b) {
+    if (a != a) return a;
+    if (b != b) return b;
+    if (a == 0.0 && b == 0.0)
+        return std::signbit(a) ? b : a;
+    return a > b ? a : b;
+}
+
 /* fastutil HashCommon.mix (int phi 0x9E3779B9, h ^= h >>> 16) —
  * ConcurrentRawHashTable.java:93,114 uses it for bucket placement. */
 static inline int32_t hc_mix(int32_t x) {
@@ -1025,8 +1044,8 @@ struct AggOp : gx_op {
             if (!c->is_null(row)) {
                 double v = c->f64v[row];
                 if (s.isnull[gid]) { s.f64[gid] = v; s.isnull[gid] = 0; }
-                else s.f64[gid] = sp.func == GX_AGG_MIN_F64 ? std::min(s.f64[gid], v)
-                                                            : std::max(s.f64[gid], v);
+                else s.f64[gid] = sp.func == GX_AGG_MIN_F64
+                        ? jmin_f64o(s.f64[gid], v) : jmax_f64o(s.f64[gid], v);
             }
             break;
         }
@@ -1231,7 +1250,7 @@ struct GroupJoinOp : gx_op {
                 double v = c->f64v[row];
                 if (s.isnull[gid]) { s.f64[gid] = v; s.isnull[gid] = 0; }
                 else s.f64[gid] = sp.func == GX_AGG_MIN_F64
-                        ? std::min(s.f64[gid], v) : std::max(s.f64[gid], v);
+                        ? jmin_f64o(s.f64[gid], v) : jmax_f64o(s.f64[gid], v);
             }
             break;
         }
@@ -1400,8 +1419,8 @@ struct WindowOp : gx_op {
             if (!c->is_null(r)) {
                 double v = c->f64v[r];
                 if (s.isnull) { s.f64 = v; s.isnull = 0; }
-                else s.f64 = sp.func == GX_AGG_MIN_F64 ? std::min(s.f64, v)
-                                                       : std::max(s.f64, v);
+                else s.f64 = sp.func == GX_AGG_MIN_F64 ? jmin_f64o(s.f64, v)
+                                                       : jmax_f64o(s.f64, v);
             }
             break;
         case GX_AGG_AVG_F64:
@@ -1718,8 +1737,8 @@ struct FWindowOp : gx_op {
                 if (!c->is_null(r)) {
                     double v = c->f64v[r];
                     if (isnull) { dv = v; isnull = false; }
-                    else dv = f.func == GX_AGG_MIN_F64 ? std::min(dv, v)
-                                                       : std::max(dv, v);
+                    else dv = f.func == GX_AGG_MIN_F64 ? jmin_f64o(dv, v)
+                                                       : jmax_f64o(dv, v);
                 }
                 break;
             case GX_AGG_AVG_F64:

@@ -130,3 +130,46 @@ def test_oracle_agg_vs_brute(seed):
     # float compare with rounding (accumulation order differs)
     assert multiset(got, f64_round=6) == multiset(want, f64_round=6), \
         f"seed {seed}"
+
+
+def _nan_case(lib, device):
+    from galaxysql_amd.operators import run_agg
+    nan = float("nan")
+    keys = np.array([1, 1, 1, 2, 2, 3, 3, 4], dtype=np.int64)
+    vals = np.array([1.5, nan, 0.5, -0.0, 0.0, 0.0, -0.0, 2.0])
+    chunk = Chunk([Block(I64, values=keys), Block(F64, values=vals)])
+    out = run_agg(lib, [0], [(abi.MIN_F64, 1), (abi.MAX_F64, 1)],
+                  [I64, F64], [chunk], device=device)
+    rows = {}
+    for c in out:
+        for r in c.rows():
+            rows[r[0]] = (r[1], r[2])
+    return rows
+
+
+def test_agg_minmax_f64_java_nan_semantics_oracle():
+    """Math.min/max semantics (Double2DoubleMin.java:40-44): NaN
+    propagates; -0.0 < +0.0."""
+    import math
+    import struct as st
+    rows = _nan_case(abi.load_oracle(), -1)
+    assert math.isnan(rows[1][0]) and math.isnan(rows[1][1])
+    # group 2: min = -0.0 (signbit set), max = +0.0
+    assert rows[2][0] == 0.0 and st.pack("<d", rows[2][0])[7] == 0x80
+    assert rows[2][1] == 0.0 and st.pack("<d", rows[2][1])[7] == 0x00
+    # group 3 (order flipped): same answers
+    assert st.pack("<d", rows[3][0])[7] == 0x80
+    assert st.pack("<d", rows[3][1])[7] == 0x00
+    assert rows[4] == (2.0, 2.0)
+
+
+@pytest.mark.gpu
+def test_agg_minmax_f64_java_nan_semantics_gpu():
+    import math
+    import struct as st
+    rows = _nan_case(abi.load_hip(), 0)
+    assert math.isnan(rows[1][0]) and math.isnan(rows[1][1])
+    assert st.pack("<d", rows[2][0])[7] == 0x80
+    assert st.pack("<d", rows[2][1])[7] == 0x00
+    assert st.pack("<d", rows[3][0])[7] == 0x80
+    assert st.pack("<d", rows[3][1])[7] == 0x00
