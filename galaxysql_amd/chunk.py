@@ -194,15 +194,23 @@ def dec40_decode(p, scale: int) -> int:
     return -v if isneg else v
 
 
-def multiset(rows, f64_round=None):
+def multiset(rows, f64_round=None, f64_sign_zero=False):
     """Canonical multiset of rows for order-insensitive parity compares
     (mirrors BaseExecTest.assertExecResultByRow, BaseExecTest.java:78-103).
-    f64_round: decimal places to round floats (tolerance compare)."""
+    f64_round: decimal places to round floats (tolerance compare).
+    NaN canonicalizes to a sentinel (all Java NaNs are equal); with
+    f64_sign_zero, -0.0 and +0.0 are kept DISTINCT (Math.min ordering)."""
+    import math
     from collections import Counter
 
     def canon(v):
-        if isinstance(v, float) and f64_round is not None:
-            return round(v, f64_round)
+        if isinstance(v, float):
+            if v != v:
+                return "NaN"
+            if f64_sign_zero and v == 0.0 and math.copysign(1.0, v) < 0:
+                return "-0.0"
+            if f64_round is not None:
+                return round(v, f64_round)
         return v
 
     return Counter(tuple(canon(v) for v in r) for r in rows)

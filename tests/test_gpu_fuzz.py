@@ -271,3 +271,67 @@ def test_fuzz_join_conditions(seed):
     want = run_join(ora, jt, keys, build, probe, ptypes, btypes,
                     device=-1, **kw)
     assert_rows_match(rows_of(got), rows_of(want), f"cond seed {seed}")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(6))
+def test_fuzz_f64_minmax_nan_signed_zero(seed):
+    """MIN/MAX(F64) over payloads laced with NaN and -0.0: HIP vs oracle,
+    both implementing Java Math.min/max (Double2DoubleMin.java:40-44 —
+    NaN propagates, -0.0 < +0.0). multiset keeps -0.0 distinct from
+    +0.0 and canonicalizes NaN; only the quiet NaN is injected so the
+    propagated bits are order-independent."""
+    from galaxysql_amd.operators import run_fwindow
+    rng = np.random.default_rng(12000 + seed)
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    n = int(rng.integers(16, 20000))
+    card = int(rng.choice([3, 50]))
+    keys = rng.integers(0, card, n).astype(np.int64)
+    pool = np.array([np.nan, -0.0, 0.0, 1.5, -2.25, 7.0, -0.0, np.nan])
+    vals = pool[rng.integers(0, len(pool), n)]
+    nulls = (rng.random(n) < 0.15).astype(np.uint8)
+    types = [I64, F64]
+    cols = [(keys, None), (vals, nulls if nulls.any() else None)]
+    chunks = chunks_from_columns(types, cols,
+                                 chunk_size=int(rng.integers(64, 4096)))
+    kw = dict(group_cols=[0], aggs=[(abi.MIN_F64, 1), (abi.MAX_F64, 1)])
+    got = run_agg(hip, input_types=types, input_chunks=chunks, device=0,
+                  **kw)
+    want = run_agg(ora, input_types=types, input_chunks=chunks, device=-1,
+                   **kw)
+    assert multiset(rows_of(got), f64_sign_zero=True) == \
+        multiset(rows_of(want), f64_sign_zero=True), f"agg seed {seed}"
+
+    # NaN-free run so every group exercises the -0.0 < +0.0 ordering
+    # (with NaN in the pool nearly all groups collapse to NaN)
+    zpool = np.array([-0.0, 0.0, -0.0, 0.0, 3.5])
+    zvals = zpool[rng.integers(0, len(zpool), n)]
+    zchunks = chunks_from_columns(types, [(keys, None), (zvals, None)],
+                                  chunk_size=2048)
+    gz = run_agg(hip, input_types=types, input_chunks=zchunks, device=0,
+                 **kw)
+    wz = run_agg(ora, input_types=types, input_chunks=zchunks, device=-1,
+                 **kw)
+    assert multiset(rows_of(gz), f64_sign_zero=True) == \
+        multiset(rows_of(wz), f64_sign_zero=True), f"agg-z seed {seed}"
+
+    # running window over sorted partitions: output is positional, so tag
+    # each row with its index before the multiset compare
+    sp = np.sort(keys)
+    wchunks = chunks_from_columns(types, [(sp, None), cols[1]],
+                                  chunk_size=int(rng.integers(64, 4096)))
+    waggs = [(abi.MIN_F64, 1), (abi.MAX_F64, 1)]
+    gw = run_window(hip, [0], waggs, types, wchunks, device=0)
+    ww = run_window(ora, [0], waggs, types, wchunks, device=-1)
+    tag = lambda rows: [(i,) + r for i, r in enumerate(rows)]
+    assert multiset(tag(rows_of(gw)), f64_sign_zero=True) == \
+        multiset(tag(rows_of(ww)), f64_sign_zero=True), f"win seed {seed}"
+
+    # sliding frame MIN/MAX via the sparse table (jmin over tree combine)
+    fr = [(abi.MIN_F64, 1, abi.FRAME_ROWS_SLIDING, 3, 2),
+          (abi.MAX_F64, 1, abi.FRAME_ROWS_SLIDING, 2, 4)]
+    gf = run_fwindow(hip, [0], fr, types, wchunks, device=0)
+    wf = run_fwindow(ora, [0], fr, types, wchunks, device=-1)
+    assert multiset(tag(rows_of(gf)), f64_sign_zero=True) == \
+        multiset(tag(rows_of(wf)), f64_sign_zero=True), f"fw seed {seed}"
