@@ -335,3 +335,47 @@ def test_fuzz_f64_minmax_nan_signed_zero(seed):
     wf = run_fwindow(ora, [0], fr, types, wchunks, device=-1)
     assert multiset(tag(rows_of(gf)), f64_sign_zero=True) == \
         multiset(tag(rows_of(wf)), f64_sign_zero=True), f"fw seed {seed}"
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(4))
+def test_fuzz_f64_join_conditions_nan(seed):
+    """Residual f64 conditions (col-vs-const and col-vs-col) with NaN and
+    -0.0 in the compared payloads: HIP vs oracle. Both use raw IEEE
+    compares (Java primitive <, AbstractJoinExec.checkJoinCondition
+    :227-250): NaN fails <,<=,>,>=,=; NaN != NaN is true; -0.0 == 0.0."""
+    from galaxysql_amd.operators import JoinCond
+    rng = np.random.default_rng(13000 + seed)
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    jt = [abi.INNER, abi.LEFT, abi.SEMI, abi.ANTI][rng.integers(0, 4)]
+    n_b, n_p = int(rng.integers(1, 2000)), int(rng.integers(1, 5000))
+    card = int(rng.choice([5, 100]))
+    pool = np.array([np.nan, -0.0, 0.0, 1.5, -3.75, 10.0])
+
+    def side(n):
+        k = rng.integers(0, card, n).astype(np.int64)
+        f = pool[rng.integers(0, len(pool), n)]
+        fn = (rng.random(n) < 0.1).astype(np.uint8)
+        return chunks_from_columns(
+            [I64, F64], [(k, None), (f, fn if fn.any() else None)],
+            chunk_size=int(rng.integers(100, 3000)))
+
+    build, probe = side(n_b), side(n_p)
+    # condition row: probe cols (0,1) then build cols (2,3)
+    pick = rng.random()
+    if pick < 0.4:
+        conds = [JoinCond(1, abi.LT, -1, float(rng.choice([0.0, -0.0, 2.0])))]
+    elif pick < 0.7:
+        conds = [JoinCond(1, abi.NE, 3)]
+    else:
+        conds = [JoinCond(1, abi.GE, 3),
+                 JoinCond(3, abi.LE, -1, 5.0)]
+    kw = dict(conds=conds)
+    keys = [EquiJoinKey(0, 0, I64)]
+    got = run_join(hip, jt, keys, build, probe, [I64, F64], [I64, F64],
+                   device=0, **kw)
+    want = run_join(ora, jt, keys, build, probe, [I64, F64], [I64, F64],
+                    device=-1, **kw)
+    assert multiset(rows_of(got), f64_sign_zero=True) == \
+        multiset(rows_of(want), f64_sign_zero=True), f"seed {seed} jt {jt}"
