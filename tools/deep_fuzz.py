@@ -15,7 +15,9 @@ import test_gpu_fuzz as F  # noqa: E402
 N = int(sys.argv[1]) if len(sys.argv) > 1 else 200
 fails = 0
 for seed in range(100000, 100000 + N):
-    for fn in (F.test_fuzz_join, F.test_fuzz_agg, F.test_fuzz_join_conditions):
+    for fn in (F.test_fuzz_join, F.test_fuzz_agg, F.test_fuzz_join_conditions,
+               F.test_fuzz_f64_minmax_nan_signed_zero,
+               F.test_fuzz_f64_join_conditions_nan):
         try:
             fn.__wrapped__(seed) if hasattr(fn, "__wrapped__") else fn(seed)
         except AssertionError:
@@ -30,5 +32,5 @@ for seed in range(100000, 100000 + N):
             sys.exit(2)
     if (seed - 100000) % 25 == 0:
         print(f"... {seed - 100000}/{N}", flush=True)
-print("DEEP FUZZ:", "FAILED" if fails else "CLEAN", N, "seeds x 3 suites")
+print("DEEP FUZZ:", "FAILED" if fails else "CLEAN", N, "seeds x 5 suites")
 sys.exit(1 if fails else 0)
