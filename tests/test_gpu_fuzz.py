@@ -379,3 +379,43 @@ def test_fuzz_f64_join_conditions_nan(seed):
                     device=-1, **kw)
     assert multiset(rows_of(got), f64_sign_zero=True) == \
         multiset(rows_of(want), f64_sign_zero=True), f"seed {seed} jt {jt}"
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(4))
+def test_fuzz_f64_sum_frames_nonfinite(seed):
+    """Sliding SUM/AVG(F64) frames with NaN and +-Inf in the column: the
+    HIP path computes frames by segmented-prefix DIFFERENCE, which a
+    non-finite value poisons for every later frame in the partition
+    (NaN sticks, inf-inf=NaN) even when the frame excludes it; the
+    reference rescans each frame. k_fw_diff_f64 falls back to a direct
+    rescan whenever the diff is non-finite — a finite diff implies an
+    all-finite frame. HIP vs oracle, ROWS and RANGE bounds."""
+    from galaxysql_amd.operators import run_fwindow
+    rng = np.random.default_rng(14000 + seed)
+    hip = abi.load_hip()
+    ora = abi.load_oracle()
+    n = int(rng.integers(64, 12000))
+    parts = np.sort(rng.integers(0, max(n // 40, 1), n)).astype(np.int64)
+    pool = np.array([np.nan, np.inf, -np.inf, 1.5, -2.25, 3.0, 0.5, 4.0])
+    # mostly-finite: non-finite ~6% so many frames are clean (fast path)
+    # and partitions still get poisoned tails (rescue path)
+    pick = rng.random(n)
+    vals = pool[3 + rng.integers(0, 5, n)]
+    vals[pick < 0.06] = pool[rng.integers(0, 3, (pick < 0.06).sum())]
+    nulls = (rng.random(n) < 0.1).astype(np.uint8)
+    order = np.arange(n, dtype=np.int64)  # strictly increasing per part
+    types = [I64, I64, F64]
+    chunks = chunks_from_columns(
+        types, [(parts, None), (order, None),
+                (vals, nulls if nulls.any() else None)],
+        chunk_size=int(rng.integers(100, 4000)))
+    fr = [(abi.SUM_F64, 2, abi.FRAME_ROWS_SLIDING, int(rng.integers(0, 6)),
+           int(rng.integers(0, 6))),
+          (abi.AVG_F64, 2, abi.FRAME_ROWS_SLIDING, n, 0),
+          (abi.SUM_F64, 2, abi.FRAME_RANGE_SLIDING, 7, 7, 1, 1)]
+    got = run_fwindow(hip, [0], fr, types, chunks, device=0)
+    want = run_fwindow(ora, [0], fr, types, chunks, device=-1)
+    tag = lambda rows: [(i,) + r for i, r in enumerate(rows)]
+    assert multiset(tag(rows_of(got)), f64_round=6) == \
+        multiset(tag(rows_of(want)), f64_round=6), f"seed {seed}"
