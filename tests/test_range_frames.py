@@ -18,7 +18,8 @@ import numpy as np
 import pytest
 
 from galaxysql_amd import abi
-from galaxysql_amd.chunk import Block, Chunk, I32, I64, F64
+from galaxysql_amd.chunk import Block, Chunk, I32, I64, F64, \
+    chunks_from_columns, rows_of
 from galaxysql_amd.operators import OverWindowFramesExec
 
 # the shared @Before input (OverWindowFramesExecTest.java:46-60):
@@ -389,3 +390,61 @@ def test_range_frames_gpu_at_scale():
     assert len(hip) == len(ora)
     for i, (h, o) in enumerate(zip(hip, ora)):
         assert h[3:] == o[3:], (i, h, o)
+
+
+@pytest.mark.parametrize("seed", range(4))
+def test_oracle_frame_sums_nonfinite_vs_brute(seed):
+    """The oracle computes frame SUM/AVG(F64) by per-frame rescan
+    (RowsSlidingOverFrame accumulate semantics) — pin it against a
+    literal Python restatement on NaN/Inf-laced input so the GPU's
+    prefix-diff + poisoned-prefix rescue has a trustworthy anchor."""
+    import math
+    from galaxysql_amd.operators import run_fwindow
+    rng = np.random.default_rng(15000 + seed)
+    lib = abi.load_oracle()
+    n = int(rng.integers(30, 600))
+    parts = np.sort(rng.integers(0, max(n // 25, 1), n)).astype(np.int64)
+    pool = np.array([np.nan, np.inf, -np.inf, 1.5, -2.25, 3.0, 0.5])
+    vals = pool[3 + rng.integers(0, 4, n)]
+    mask = rng.random(n) < 0.08
+    vals[mask] = pool[rng.integers(0, 3, mask.sum())]
+    nulls = (rng.random(n) < 0.1).astype(np.uint8)
+    prec, foll = int(rng.integers(0, 5)), int(rng.integers(0, 5))
+    types = [I64, F64]
+    ch = chunks_from_columns(
+        types, [(parts, None), (vals, nulls if nulls.any() else None)],
+        chunk_size=int(rng.integers(16, 200)))
+    out = rows_of(run_fwindow(
+        lib, [0], [(abi.SUM_F64, 1, abi.FRAME_ROWS_SLIDING, prec, foll),
+                   (abi.AVG_F64, 1, abi.FRAME_ROWS_SLIDING, prec, foll)],
+        types, ch, device=-1))
+
+    starts = {}
+    for i, p in enumerate(parts):
+        starts.setdefault(int(p), i)
+    for i in range(n):
+        p = int(parts[i])
+        s = starts[p]
+        e = s
+        while e < n and parts[e] == p:
+            e += 1
+        lo, hi = max(s, i - prec), min(e - 1, i + foll)
+        acc, cnt = None, 0
+        for r in range(lo, hi + 1):
+            if nulls[r]:
+                continue
+            acc = float(vals[r]) if acc is None else acc + float(vals[r])
+            cnt += 1
+        gs, ga = out[i][2], out[i][3]
+        if acc is None:
+            assert gs is None and ga is None, (seed, i)
+        else:
+            ea = acc / cnt
+            for got, want in ((gs, acc), (ga, ea)):
+                if math.isnan(want):
+                    assert math.isnan(got), (seed, i, got, want)
+                elif math.isinf(want):
+                    assert got == want, (seed, i, got, want)
+                else:
+                    assert abs(got - want) <= 1e-9 * max(1.0, abs(want)), \
+                        (seed, i, got, want)
