@@ -336,6 +336,18 @@ def test_fuzz_f64_minmax_nan_signed_zero(seed):
     assert multiset(tag(rows_of(gf)), f64_sign_zero=True) == \
         multiset(tag(rows_of(wf)), f64_sign_zero=True), f"fw seed {seed}"
 
+    # fused group-join MIN/MAX(F64) accumulates per build position
+    nb = int(rng.integers(1, 800))
+    bkeys = rng.integers(0, card, nb).astype(np.int64)
+    bchunks = chunks_from_columns([I64], [(bkeys, None)], chunk_size=512)
+    gjkw = dict(group_cols=[0], aggs=[(abi.MIN_F64, 1), (abi.MAX_F64, 1)])
+    gj = run_groupjoin(hip, abi.LEFT, [EquiJoinKey(0, 0, I64)], bchunks,
+                       chunks, [I64], types, device=0, **gjkw)
+    wj = run_groupjoin(ora, abi.LEFT, [EquiJoinKey(0, 0, I64)], bchunks,
+                       chunks, [I64], types, device=-1, **gjkw)
+    assert multiset(rows_of(gj), f64_sign_zero=True) == \
+        multiset(rows_of(wj), f64_sign_zero=True), f"gj seed {seed}"
+
 
 @pytest.mark.gpu
 @pytest.mark.parametrize("seed", range(4))
