@@ -89,11 +89,16 @@ class Block:
 
 
 class Chunk:
+    """The reference's Chunk carries positionCount independently of its
+    blocks (chunk/Chunk.java:55-89) — a block-less chunk can still have
+    rows (COUNT(*)-only plans). n_rows overrides the derived count for
+    that case."""
     __slots__ = ("blocks", "n_rows")
 
-    def __init__(self, blocks):
+    def __init__(self, blocks, n_rows=None):
         self.blocks = list(blocks)
-        self.n_rows = self.blocks[0].n_rows if self.blocks else 0
+        self.n_rows = (n_rows if n_rows is not None
+                       else self.blocks[0].n_rows if self.blocks else 0)
         for b in self.blocks:
             assert b.n_rows == self.n_rows, "ragged chunk"
 

@@ -106,7 +106,8 @@ def _side(rng, key_types, n, card, null_frac, extra_payloads):
                 sub.append(Block(b.type, values=b.values[start:end],
                                  nulls=None if b.nulls is None
                                  else b.nulls[start:end]))
-        chunks.append(Chunk(sub))
+        chunks.append(Chunk(sub, n_rows=min(start + chunk_size, n) - start)
+                      if not sub else Chunk(sub))
     return types, chunks
 
 

@@ -173,3 +173,33 @@ def test_agg_minmax_f64_java_nan_semantics_gpu():
     assert st.pack("<d", rows[2][1])[7] == 0x00
     assert st.pack("<d", rows[3][0])[7] == 0x80
     assert st.pack("<d", rows[3][1])[7] == 0x00
+
+
+def _global_agg_cases(lib, device):
+    from galaxysql_amd.chunk import Chunk
+    # (a) zero input chunks  (b) block-less chunks carrying only a row count
+    a = run_agg(lib, group_cols=[], aggs=[(abi.COUNT_ROW, -1),
+                                          (abi.SUM_I64N, 0)],
+                input_types=[I64], input_chunks=[], device=device)
+    b = run_agg(lib, group_cols=[], aggs=[(abi.COUNT_ROW, -1)],
+                input_types=[], input_chunks=[Chunk([], n_rows=123),
+                                              Chunk([], n_rows=77)],
+                device=device)
+    return rows_of(a), rows_of(b)
+
+
+def test_global_agg_empty_input_oracle():
+    """SQL global aggregate over empty input emits ONE row: COUNT(*)=0,
+    null-init SUM stays NULL (HashAggExec no-group-by contract); a
+    block-less chunk still carries positionCount (Chunk.java:55-89).
+    Deep-fuzz seed 100229 caught the HIP side emitting nothing."""
+    a, b = _global_agg_cases(abi.load_oracle(), -1)
+    assert a == [(0, None)]
+    assert b == [(200,)]
+
+
+@pytest.mark.gpu
+def test_global_agg_empty_input_gpu():
+    a, b = _global_agg_cases(abi.load_hip(), 0)
+    assert a == [(0, None)]
+    assert b == [(200,)]
