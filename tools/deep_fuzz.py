@@ -13,8 +13,9 @@ import numpy as np  # noqa: E402
 import test_gpu_fuzz as F  # noqa: E402
 
 N = int(sys.argv[1]) if len(sys.argv) > 1 else 200
+BASE = int(sys.argv[2]) if len(sys.argv) > 2 else 100000
 fails = 0
-for seed in range(100000, 100000 + N):
+for seed in range(BASE, BASE + N):
     for fn in (F.test_fuzz_join, F.test_fuzz_agg, F.test_fuzz_join_conditions,
                F.test_fuzz_f64_minmax_nan_signed_zero,
                F.test_fuzz_f64_join_conditions_nan):
@@ -30,7 +31,7 @@ for seed in range(100000, 100000 + N):
             print(f"ERROR {fn.__name__} seed {seed}")
             traceback.print_exc()
             sys.exit(2)
-    if (seed - 100000) % 25 == 0:
-        print(f"... {seed - 100000}/{N}", flush=True)
+    if (seed - BASE) % 25 == 0:
+        print(f"... {seed - BASE}/{N}", flush=True)
 print("DEEP FUZZ:", "FAILED" if fails else "CLEAN", N, "seeds x 5 suites")
 sys.exit(1 if fails else 0)
