@@ -69,8 +69,11 @@ def _col(rng, t, n, card, null_frac):
     nulls = (rng.random(n) < null_frac).astype(np.uint8) \
         if null_frac > 0 else None
     if t == SLICE:
+        # value 0 maps to the EMPTY string: "" is a real key, equal to
+        # itself and distinct from NULL (Block.equals byte semantics)
         vals = [None if nulls is not None and nulls[i]
-                else f"v{rng.integers(0, card)}" for i in range(n)]
+                else ("" if (x := int(rng.integers(0, card))) == 0
+                      else f"v{x}") for i in range(n)]
         return Block.of(SLICE, vals)
     if t == I32:
         v = rng.integers(-card, card, n).astype(np.int32)
