@@ -18,7 +18,8 @@ fails = 0
 for seed in range(BASE, BASE + N):
     for fn in (F.test_fuzz_join, F.test_fuzz_agg, F.test_fuzz_join_conditions,
                F.test_fuzz_f64_minmax_nan_signed_zero,
-               F.test_fuzz_f64_join_conditions_nan):
+               F.test_fuzz_f64_join_conditions_nan,
+               F.test_fuzz_groupjoin_window):
         try:
             fn.__wrapped__(seed) if hasattr(fn, "__wrapped__") else fn(seed)
         except AssertionError:
@@ -33,5 +34,5 @@ for seed in range(BASE, BASE + N):
             sys.exit(2)
     if (seed - BASE) % 25 == 0:
         print(f"... {seed - BASE}/{N}", flush=True)
-print("DEEP FUZZ:", "FAILED" if fails else "CLEAN", N, "seeds x 5 suites")
+print("DEEP FUZZ:", "FAILED" if fails else "CLEAN", N, "seeds x 6 suites")
 sys.exit(1 if fails else 0)
