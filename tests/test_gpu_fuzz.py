@@ -90,7 +90,7 @@ def _side(rng, key_types, n, card, null_frac, extra_payloads):
     for t in extra_payloads:
         types.append(t)
         blocks.append(_col(rng, t, n, 1000, null_frac / 2))
-    chunk_size = int(rng.integers(100, 2000))
+    chunk_size = int(rng.integers(8, 2000))
     chunks = []
     for start in range(0, n, chunk_size):
         sub = []
@@ -217,7 +217,7 @@ def test_fuzz_groupjoin_window(seed):
     vals = rng.integers(-100, 100, n)
     nulls = (rng.random(n) < 0.15).astype(np.uint8)
     chunks = chunks_from_columns([I64, I64], [(parts, None), (vals, nulls)],
-                                 chunk_size=int(rng.integers(50, 3000)))
+                                 chunk_size=int(rng.integers(2, 3000)))
     aggs = [(abi.COUNT_ROW, -1), (abi.SUM_I64, 1), (abi.MIN_I64, 1)]
     reset = [bool(rng.random() < 0.3) for _ in aggs]
     got = run_window(hip, [0], aggs, [I64, I64], chunks, reset=reset,
