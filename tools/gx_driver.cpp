@@ -467,6 +467,36 @@ static void t_agg(int device) {
     std::printf("  agg: %lld groups, totals ok\n", (long long)groups);
 }
 
+/* SQL global aggregate over EMPTY input emits exactly one row:
+ * COUNT(*)=0, null-init SUM NULL (HashAggExec no-group-by contract;
+ * regression for the consume()-seeding bug deep-fuzz seed 100229 found). */
+static void t_agg_empty_global(int device) {
+    gx_agg_spec specs[2] = {{GX_AGG_COUNT_ROW, -1}, {GX_AGG_SUM_I64N, 0}};
+    int32_t itypes[1] = {GX_I64};
+    gx_agg_cfg cfg{};
+    cfg.n_group_cols = 0; cfg.group_cols = nullptr;
+    cfg.n_aggs = 2; cfg.aggs = specs;
+    cfg.n_input_cols = 1; cfg.input_types = itypes;
+    cfg.device = device;
+    gx_op *op = api.agg_create(&cfg);
+    CHECK(op, "agg_create(empty-global)");
+    if (!op) return;
+    CHECK(api.agg_build(op) == 0, "agg_build(empty)");
+    gx_result *res = nullptr;
+    CHECK(api.agg_next(op, &res) == 0, "agg_next(empty)");
+    CHECK(res && res->chunk.n_rows == 1, "empty global agg must emit 1 row");
+    if (res) {
+        api.result_to_host(res);
+        CHECK(cell_i64(res, 0, 0) == 0, "COUNT(*) over empty != 0");
+        CHECK(res->chunk.blocks[1].nulls &&
+                  ((const uint8_t *)res->chunk.blocks[1].nulls)[0] == 1,
+              "null-init SUM over empty must be NULL");
+        api.result_release(res);
+    }
+    api.agg_close(op);
+    std::printf("  agg empty-global: one row, COUNT=0, SUM NULL\n");
+}
+
 /* partition: row conservation + consume/consume_concat agreement. */
 static void t_part(int device) {
     const int N = 50000, PARTS = 8;
@@ -1042,6 +1072,7 @@ int main(int argc, char **argv) {
         t_concurrent(device);
         t_join_left(device);
         t_agg(device);
+        t_agg_empty_global(device);
         t_part(device);
         t_scan(device);
         t_groupjoin(device);
