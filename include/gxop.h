@@ -291,7 +291,11 @@ gx_op *gxop_agg_create(const gx_agg_cfg *cfg);
 int gxop_agg_consume(gx_op *op, const gx_chunk *input_chunk);
 int gxop_agg_build(gx_op *op);
 /* emit result batches: group-key columns then one column per aggregator
- * (HashAggResultIterator). Returns 0 and *out=NULL when exhausted. */
+ * (HashAggResultIterator). Returns 0 and *out=NULL when exhausted.
+ * With n_group_cols == 0 the aggregate is GLOBAL and emits exactly one
+ * row even when no input rows were consumed (SQL: COUNT(*)=0, SUM0=0,
+ * null-init aggregators NULL); with group columns, empty input emits
+ * nothing. */
 int gxop_agg_next(gx_op *op, gx_result **out);
 int gxop_agg_close(gx_op *op);
 
