@@ -247,3 +247,34 @@ def test_compressed_frame_roundtrip_native_and_python():
         shape=(4000,)).copy()
     lib.lib.gxop_chunk_free(ch_out)
     assert np.array_equal(got, vals)
+
+
+def test_mixed_stream_with_blockless_frame():
+    from galaxysql_amd import serde
+    """A frame stream mixing an incompressible chunk, a highly
+    compressible one (LZ4 marker), and a BLOCK-LESS chunk that carries
+    only positionCount (COUNT(*)-only pages; Chunk.java:55-89) must
+    round-trip value-identically in order."""
+    import numpy as np
+    from galaxysql_amd.chunk import Block, Chunk, I64
+    rng = np.random.default_rng(11)
+    a = Chunk([Block(I64, values=rng.integers(-2**62, 2**62, 500))])
+    b = Chunk([Block(I64, values=np.zeros(4000, dtype=np.int64))])
+    c = Chunk([], n_rows=77)
+    buf = serde.serialize_chunks([a, b, c])
+    out = serde.deserialize_chunks(buf, [I64])
+    # deserialize_chunks takes one type list; blockless frame reports 0
+    # blocks regardless of the declared types
+    assert len(out) == 3
+    assert out[0].rows() == a.rows()
+    assert out[1].rows() == b.rows()
+    assert out[2].n_rows == 77 and len(out[2].blocks) == 0
+    # the all-zeros frame must actually have used the COMPRESSED marker
+    import struct
+    pos = 0
+    markers = []
+    for _ in range(3):
+        n_rows, marker, _unc, size = struct.unpack_from("<ibii", buf, pos)
+        markers.append(marker)
+        pos += 13 + size
+    assert markers[1] == serde.COMPRESSED
