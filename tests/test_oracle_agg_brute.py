@@ -88,7 +88,7 @@ def _brute(groups, ivals, inulls, fvals, fnulls, aggs):
     return rows
 
 
-@pytest.mark.parametrize("seed", range(20))
+@pytest.mark.parametrize("seed", range(32))
 def test_oracle_agg_vs_brute(seed):
     rng = np.random.default_rng(8000 + seed)
     lib = abi.load_oracle()

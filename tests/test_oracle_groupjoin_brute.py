@@ -16,7 +16,7 @@ def _v(vals, nulls, i):
     return None if nulls[i] else int(vals[i])
 
 
-@pytest.mark.parametrize("seed", range(20))
+@pytest.mark.parametrize("seed", range(32))
 def test_oracle_groupjoin_vs_brute(seed):
     rng = np.random.default_rng(9000 + seed)
     from galaxysql_amd.abi import load_oracle

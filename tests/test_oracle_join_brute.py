@@ -92,7 +92,7 @@ def _row(jt, r, m, pk, pn, pv, pvn, bk, bn, bv, bvn, build_outer):
     return probe + build
 
 
-@pytest.mark.parametrize("seed", range(30))
+@pytest.mark.parametrize("seed", range(45))
 def test_oracle_join_vs_brute(seed):
     rng = np.random.default_rng(7000 + seed)
     lib = abi.load_oracle()

@@ -392,7 +392,7 @@ def test_range_frames_gpu_at_scale():
         assert h[3:] == o[3:], (i, h, o)
 
 
-@pytest.mark.parametrize("seed", range(4))
+@pytest.mark.parametrize("seed", range(8))
 def test_oracle_frame_sums_nonfinite_vs_brute(seed):
     """The oracle computes frame SUM/AVG(F64) by per-frame rescan
     (RowsSlidingOverFrame accumulate semantics) — pin it against a
