@@ -218,3 +218,44 @@ def test_ntile_cume_percent_vs_numpy():
         assert r[2] == exp_tile, i
         assert abs(r[3] - exp_cume) < 1e-12, i
         assert abs(r[4] - exp_pct) < 1e-12, i
+
+
+@pytest.mark.parametrize("seed", range(12))
+def test_navigation_oracle_vs_brute(seed):
+    """FIRST/LAST/NTH_VALUE, LAG/LEAD over random partitions with NULLs
+    and string payloads (index-gather semantics, FirstValue/Lag.java):
+    NTH's k and LAG/LEAD offsets are random; out-of-partition indexes
+    and NULL source values both emit NULL."""
+    from galaxysql_amd.chunk import SLICE, chunks_from_columns
+    rng = np.random.default_rng(17000 + seed)
+    lib = abi.load_oracle()
+    n = int(rng.integers(1, 2500))
+    parts = np.sort(rng.integers(0, max(n // 31, 1), n)).astype(np.int64)
+    raw = [None if rng.random() < 0.15 else f"s{rng.integers(0, 50)}"
+           for _ in range(n)]
+    k_nth = int(rng.integers(1, 6))
+    k_lag = int(rng.integers(1, 4))
+    k_lead = int(rng.integers(1, 4))
+    W = abi.FRAME_WHOLE_PARTITION
+    chunks = chunks_from_columns(
+        [I64, SLICE], [(parts, None), Block.of(SLICE, raw)],
+        chunk_size=int(rng.integers(2, 900)))
+    rows = rows_of(run_fwindow(
+        lib, [0],
+        [(abi.FIRST_VALUE, 1, W), (abi.LAST_VALUE, 1, W),
+         (abi.NTH_VALUE, 1, W, k_nth), (abi.LAG, 1, W, k_lag),
+         (abi.LEAD, 1, W, k_lead)],
+        [I64, SLICE], chunks))
+    enc = [None if v is None else v.encode() for v in raw]
+    for i in range(n):
+        seg = np.nonzero(parts == parts[i])[0]
+        s, e = int(seg[0]), int(seg[-1])
+        got = rows[i][2:]
+        want = (
+            enc[s],
+            enc[e],
+            enc[s + k_nth - 1] if s + k_nth - 1 <= e else None,
+            enc[i - k_lag] if i - k_lag >= s else None,
+            enc[i + k_lead] if i + k_lead <= e else None,
+        )
+        assert got == want, (seed, i, got, want)
