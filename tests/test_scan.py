@@ -144,3 +144,86 @@ def test_like_scan_hip_vs_oracle():
     ref = run_like_scan(oracle, chunks, -1)
     got = run_like_scan(hip, chunks, 0)
     assert multiset(got) == multiset(ref)
+
+
+@pytest.mark.parametrize("seed", range(14))
+def test_scan_oracle_vs_brute_random(seed):
+    """Random predicate sets (i32/i64/f64 compares + SLICE CONTAINS) and
+    random projection mixes vs a literal per-row restatement (SQL: NULL
+    fails every predicate; AND of all predicates)."""
+    import subprocess as sp
+    sp.run(["make", "-C", os.path.join(REPO, "oracle")], check=True,
+           capture_output=True)
+    lib = abi.load_oracle()
+    rng = np.random.default_rng(18000 + seed)
+    n = int(rng.integers(1, 4000))
+    dates = rng.integers(0, 40, n).astype(np.int32)
+    dn = (rng.random(n) < 0.1).astype(np.uint8)
+    qty = rng.integers(0, 30, n).astype(np.int64)
+    f = np.round(rng.standard_normal(n) * 5, 3)
+    fn = (rng.random(n) < 0.1).astype(np.uint8)
+    tags = [None if rng.random() < 0.1 else
+            ("" if rng.random() < 0.1 else
+             "".join(rng.choice(list("abc"), rng.integers(1, 6))))
+            for _ in range(n)]
+    types = [I32, I64, F64, SLICE]
+    chunks = chunks_from_columns(
+        types, [(dates, dn if dn.any() else None), (qty, None),
+                (f, fn if fn.any() else None), Block.of(SLICE, tags)],
+        chunk_size=int(rng.integers(2, 1500)))
+
+    cmps = [abi.LT, abi.LE, abi.GT, abi.GE, abi.EQ, abi.NE]
+    preds = []
+    n_preds = int(rng.integers(0, 4))
+    for _ in range(n_preds):
+        pick = rng.random()
+        if pick < 0.35:
+            preds.append((0, cmps[rng.integers(0, 6)], int(rng.integers(0, 40))))
+        elif pick < 0.6:
+            preds.append((1, cmps[rng.integers(0, 6)], int(rng.integers(0, 30))))
+        elif pick < 0.8:
+            preds.append((2, cmps[rng.integers(0, 6)],
+                          float(np.round(rng.standard_normal() * 5, 2))))
+        else:
+            preds.append((3, abi.CONTAINS,
+                          "".join(rng.choice(list("abc"),
+                                             rng.integers(1, 3)))))
+    projs = [(abi.PROJ_COPY, 1, -1), (abi.PROJ_COPY, 3, -1),
+             (abi.PROJ_COPY, 2, -1)]
+
+    op = ScanExec(lib, preds, projs, types, device=-1)
+    got = []
+    try:
+        for ch in chunks:
+            out = op.consume_chunk(ch)
+            if out is not None:
+                got.extend(out.rows())
+    finally:
+        op.close()
+
+    def cmp_ok(v, c, const):
+        if v is None:
+            return False
+        if c == abi.LT: return v < const
+        if c == abi.LE: return v <= const
+        if c == abi.GT: return v > const
+        if c == abi.GE: return v >= const
+        if c == abi.EQ: return v == const
+        return v != const
+
+    want = []
+    for i in range(n):
+        vals = (None if dn[i] else int(dates[i]), int(qty[i]),
+                None if fn[i] else float(f[i]), tags[i])
+        ok = True
+        for col, c, const in preds:
+            if c == abi.CONTAINS:
+                ok = vals[3] is not None and const in vals[3]
+            else:
+                ok = cmp_ok(vals[col], c, const)
+            if not ok:
+                break
+        if ok:
+            t = None if vals[3] is None else vals[3].encode()
+            want.append((vals[1], t, vals[2]))
+    assert got == want, f"seed {seed}: {len(got)} vs {len(want)}"
