@@ -187,7 +187,7 @@ def test_scan_oracle_vs_brute_random(seed):
         else:
             preds.append((3, abi.CONTAINS,
                           "".join(rng.choice(list("abc"),
-                                             rng.integers(1, 3)))))
+                                             rng.integers(0, 3)))))
     projs = [(abi.PROJ_COPY, 1, -1), (abi.PROJ_COPY, 3, -1),
              (abi.PROJ_COPY, 2, -1)]
 
@@ -227,3 +227,17 @@ def test_scan_oracle_vs_brute_random(seed):
             t = None if vals[3] is None else vals[3].encode()
             want.append((vals[1], t, vals[2]))
     assert got == want, f"seed {seed}: {len(got)} vs {len(want)}"
+
+
+def test_contains_empty_pattern_oracle():
+    """LIKE '%%' (empty CONTAINS pattern) keeps every non-NULL string,
+    including the empty string; NULL still fails (SQL LIKE on NULL)."""
+    from galaxysql_amd.chunk import SLICE
+    lib = abi.load_oracle()
+    ch = Chunk([Block.of(I64, [1, 2, 3, 4]),
+                Block.of(SLICE, ["abc", "", None, "bcd"])])
+    sc = ScanExec(lib, [(1, abi.CONTAINS, "")], [(abi.PROJ_COPY, 0, -1)],
+                  [I64, SLICE], device=-1)
+    out = sc.consume_chunk(ch)
+    sc.close()
+    assert [r[0] for r in out.rows()] == [1, 2, 4]
