@@ -1855,6 +1855,10 @@ gx_op *gxop_join_create(const gx_join_cfg *cfg) {
         gx_set_err("bad join cfg");
         return nullptr;
     }
+    if (cfg->join_type < GX_JOIN_INNER || cfg->join_type > GX_JOIN_ANTI) {
+        gx_set_err("unknown join type");
+        return nullptr;
+    }
     if (cfg->n_conds < 0 || cfg->n_conds > GX_MAX_CONDS ||
         (cfg->n_conds > 0 && !cfg->conds)) {
         gx_set_err("bad join condition list (max 4 AND terms)");

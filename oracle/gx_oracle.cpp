@@ -2149,6 +2149,10 @@ extern "C" {
 
 gx_op *gxop_join_create(const gx_join_cfg *cfg) {
     if (!cfg || cfg->n_keys <= 0) { set_err("bad join cfg"); return nullptr; }
+    if (cfg->join_type < GX_JOIN_INNER || cfg->join_type > GX_JOIN_ANTI) {
+        set_err("unknown join type");
+        return nullptr;
+    }
     return new JoinOp(cfg);
 }
 int gxop_join_consume(gx_op *op, const gx_chunk *c) {
@@ -2180,15 +2184,21 @@ int gxop_join_probe_flush(gx_op *op, gx_result **out) {
 int gxop_join_close(gx_op *op) { delete op; return 0; }
 
 static int agg_funcs_ok(const gx_agg_spec *aggs, int32_t n) {
-    for (int32_t i = 0; i < n; i++)
-        if (aggs[i].func == GX_AGG_RANK || aggs[i].func == GX_AGG_DENSE_RANK)
-            return 0;  /* rank family is window-only (needs ORDER BY runs) */
+    for (int32_t i = 0; i < n; i++) {
+        int32_t f = aggs[i].func;
+        /* group-by supports the accumulator set only: COUNT..BIT_XOR and
+         * the null-init Sum; rank/navigation families are window-only and
+         * anything else is unknown */
+        if ((f < GX_AGG_COUNT_ROW || f > GX_AGG_BIT_XOR) &&
+            f != GX_AGG_SUM_I64N)
+            return 0;
+    }
     return 1;
 }
 
 gx_op *gxop_agg_create(const gx_agg_cfg *cfg) {
     if (cfg && !agg_funcs_ok(cfg->aggs, cfg->n_aggs)) {
-        set_err("RANK/DENSE_RANK are window-only");
+        set_err("unknown or window-only agg func in group-by");
         return nullptr;
     }
     if (!cfg) { set_err("bad agg cfg"); return nullptr; }
@@ -2212,7 +2222,7 @@ int gxop_agg_close(gx_op *op) { delete op; return 0; }
 
 gx_op *gxop_groupjoin_create(const gx_groupjoin_cfg *cfg) {
     if (cfg && !agg_funcs_ok(cfg->aggs, cfg->n_aggs)) {
-        set_err("RANK/DENSE_RANK are window-only");
+        set_err("unknown or window-only agg func in group-by");
         return nullptr;
     }
     if (!cfg || cfg->n_keys <= 0 ||
@@ -2243,6 +2253,16 @@ int gxop_groupjoin_close(gx_op *op) { delete op; return 0; }
 
 gx_op *gxop_window_create(const gx_window_cfg *cfg) {
     if (!cfg || cfg->n_aggs <= 0) { set_err("bad window cfg"); return nullptr; }
+    for (int32_t i = 0; i < cfg->n_aggs; i++) {
+        int32_t f = cfg->aggs[i].func;
+        /* running window: accumulators + rank family + null-init Sum;
+         * navigation/distribution funcs are frame-window only */
+        if ((f < GX_AGG_COUNT_ROW || f > GX_AGG_DENSE_RANK) &&
+            f != GX_AGG_SUM_I64N) {
+            set_err("unknown or frame-only func in running window");
+            return nullptr;
+        }
+    }
     return new WindowOp(cfg);
 }
 int gxop_window_consume(gx_op *op, const gx_chunk *c, gx_result **out) {
@@ -2302,7 +2322,7 @@ gx_op *gxop_fwindow_create(const gx_fwindow_cfg *cfg) {
         for (int32_t i = 0; i < cfg->n_frames; i++)
             if (cfg->frames[i].func == GX_AGG_RANK ||
                 cfg->frames[i].func == GX_AGG_DENSE_RANK) {
-                set_err("RANK/DENSE_RANK are window-only");
+                set_err("unknown or window-only agg func in group-by");
                 return nullptr;
             }
     if (!fwindow_cfg_ok(cfg)) { set_err("bad fwindow cfg (unsupported frame/func)"); return nullptr; }
@@ -2367,6 +2387,25 @@ int gxop_agg_get_stats(gx_op *op, gx_agg_stats *out) {
 
 gx_op *gxop_scan_create(const gx_scan_cfg *cfg) {
     if (!cfg || cfg->n_projs <= 0) { set_err("bad scan cfg"); return nullptr; }
+    for (int32_t i = 0; i < cfg->n_preds; i++) {
+        const gx_pred &p = cfg->preds[i];
+        if (p.col < 0 || p.col >= cfg->n_input_cols ||
+            p.cmp < GX_CMP_LT || p.cmp > GX_CMP_CONTAINS) {
+            set_err("bad scan predicate (column or comparison)");
+            return nullptr;
+        }
+    }
+    for (int32_t i = 0; i < cfg->n_projs; i++) {
+        const gx_proj &p = cfg->projs[i];
+        if (p.op < GX_PROJ_COPY || p.op > GX_PROJ_SCALED_TO_DEC ||
+            p.a < 0 || p.a >= cfg->n_input_cols ||
+            ((p.op == GX_PROJ_REV_F64 || p.op == GX_PROJ_REV_SCALED4 ||
+              p.op == GX_PROJ_Q9_AMOUNT4) &&
+             (p.b < 0 || p.b >= cfg->n_input_cols))) {
+            set_err("bad scan projection (op or column)");
+            return nullptr;
+        }
+    }
     return new ScanOp(cfg);
 }
 int gxop_scan_consume(gx_op *op, const gx_chunk *c, gx_result **out) {
