@@ -3,7 +3,9 @@
 re-agg mapping in galaxysql_amd.exchange.final_agg_specs: COUNT->SUM,
 SUM->SUM, MIN/MAX->self). Simulated with 3 'ranks' on the oracle — the
 RCCL exchange mechanics themselves are covered by test_exchange_cpu."""
+import os
 import numpy as np
+import pytest
 
 from galaxysql_amd import abi
 from galaxysql_amd.chunk import I64, I32, F64, chunks_from_columns, \
@@ -113,3 +115,37 @@ def test_two_phase_distributed_gloo(tmp_path):
                 assert np.isnan(a) and np.isnan(b)
             else:
                 assert abs(a - b) < 1e-6, (g, e)
+
+
+@pytest.mark.timeout(300)
+def test_global_agg_distributed_empty_rank(tmp_path):
+    """GLOBAL aggregate at world 2 where rank 1 consumed nothing: its
+    SQL partial row (0, NULL) must flow through the exchange and leave
+    the final totals equal to rank 0's data alone."""
+    import subprocess
+    import sys as _sys
+    world = 2
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29733",
+               WORLD_SIZE=str(world), PYTHONPATH=REPO, OMP_NUM_THREADS="1")
+    worker = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                          "_global_agg_worker.py")
+    procs = []
+    for rank in range(world):
+        e = dict(env, RANK=str(rank))
+        procs.append(subprocess.Popen([_sys.executable, worker,
+                                       str(tmp_path)], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=240)[0].decode() for p in procs]
+    for rank, p in enumerate(procs):
+        assert p.returncode == 0, f"rank {rank} failed:\n{outs[rank]}"
+    import numpy as np
+    vals = np.arange(1, 101)
+    expect_cnt = 100
+    expect_sum = int(vals[vals % 7 != 0].sum())
+    for rank in range(world):
+        got = np.load(tmp_path / f"ga_{rank}.npy")
+        assert int(got[0]) == expect_cnt
+        assert int(got[1]) == expect_sum
